@@ -1,0 +1,174 @@
+/* thrill_amd.h — C-ABI of libt9.so, the MI355X-native implementation of
+ * Thrill's Sort / ReduceByKey DOp hot path (hand-written HIP/CDNA4 kernels,
+ * gfx950). This is the drop-in seam of SURVEY.md §8b: the reference's
+ * SortNode (thrill/api/sort.hpp:64-787) and ReduceNode
+ * (thrill/api/reduce_by_key.hpp:64-239) become thin C++ hosts that marshal
+ * item Files <-> device buffers and call these entry points; anything
+ * (including a patched reference build) can call the shim directly. The
+ * reference-side binding a maintainer would add is shown in INTEGRATION.md.
+ *
+ * Conventions: all functions are synchronous launches on the passed HIP
+ * stream (void* == hipStream_t; NULL = default stream); they return 0 on
+ * success or a negative errno-style value. The caller owns every buffer;
+ * d_* pointers are device memory. No torch types anywhere.
+ *
+ * The GPU library contains the PRODUCT path only: it never falls back to
+ * CPU. Without a GPU, t9_create fails; nothing here routes through the
+ * oracle.
+ */
+#ifndef THRILL_AMD_H
+#define THRILL_AMD_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* Opaque context: device id, rank, world size, optional RCCL communicator.
+ * Mirrors the per-worker api::Context the reference nodes consume
+ * (thrill/api/context.hpp:243-245: my_rank / num_workers). */
+typedef struct t9_context t9_context;
+
+/* comm: an existing ncclComm_t (RCCL) or NULL for single-GPU use. */
+int t9_create(t9_context** out, int device, int rank, int world, void* comm);
+int t9_destroy(t9_context* ctx);
+const char* t9_version(void);
+
+/* ------------------------------------------------------------------ *
+ * Synthetic input generation (device-side, seeded; bit-identical to the
+ * oracle's t9o_gen_* so CPU/GPU parity runs on identical bytes).
+ * Record layout restates examples/terasort/terasort.cpp:31-118.
+ * ------------------------------------------------------------------ */
+int t9_gen_u64(t9_context* ctx, uint64_t* d_out, uint64_t index0, uint64_t n,
+               uint64_t seed, void* stream);
+int t9_gen_records(t9_context* ctx, uint8_t* d_out, uint64_t index0,
+                   uint64_t n, uint64_t seed, void* stream);
+
+/* ------------------------------------------------------------------ *
+ * Local sort — replaces SortAndWriteToFile's std::sort run formation and
+ * the loser-tree merge (thrill/api/sort.hpp:665-786,
+ * thrill/core/multiway_merge.hpp:30-116): the whole per-GPU partition is
+ * sorted in one LSD radix pipeline, so the merge stage vanishes
+ * (SURVEY.md §8a row a5/a6).
+ * ------------------------------------------------------------------ */
+
+/* Workspace bytes for t9_sort_u64 on n keys. */
+uint64_t t9_sort_u64_workspace(uint64_t n);
+/* In-place ascending sort of n u64 keys (n < 2^32). */
+int t9_sort_u64(t9_context* ctx, uint64_t* d_keys, uint64_t n,
+                void* d_workspace, void* stream);
+
+/* Workspace bytes for t9_sort_pairs_u64_u32 on n pairs. */
+uint64_t t9_sort_pairs_workspace(uint64_t n);
+/* In-place stable ascending sort of (key, payload) pairs by key. */
+int t9_sort_pairs_u64_u32(t9_context* ctx, uint64_t* d_keys,
+                          uint32_t* d_vals, uint64_t n, void* d_workspace,
+                          void* stream);
+
+/* Extract the big-endian u64 prefix of each record's key (bytes
+ * key_off .. key_off+7) and the record index iota. rec_size % 4 == 0. */
+int t9_extract_key64(t9_context* ctx, const uint8_t* d_recs, uint64_t n,
+                     uint32_t rec_size, uint32_t key_off, uint64_t* d_keys,
+                     uint32_t* d_idx, void* stream);
+
+/* out[i] = recs[idx[i]] for fixed-size records (rec_size % 4 == 0). */
+int t9_gather_records(t9_context* ctx, const uint8_t* d_recs,
+                      const uint32_t* d_idx, uint64_t n, uint32_t rec_size,
+                      uint8_t* d_out, void* stream);
+
+/* Workspace bytes for t9_sort_records. */
+uint64_t t9_sort_records_workspace(uint64_t n, uint32_t rec_size);
+/* Sort n fixed-size records by the acceptance total order: lexicographic
+ * over the whole record (key prefix first — terasort.cpp:35-37 compares the
+ * 10-byte key; ties beyond the radix-sorted u64 key prefix are resolved by
+ * comparing the remaining bytes). d_in is preserved; d_out receives the
+ * sorted sequence. key_len <= rec_size; key bytes start at offset 0.
+ * Synchronizes the stream internally (tie fixing may need a host pass). */
+int t9_sort_records(t9_context* ctx, const uint8_t* d_in, uint8_t* d_out,
+                    uint64_t n, uint32_t rec_size, uint32_t key_len,
+                    void* d_workspace, void* stream);
+
+/* ------------------------------------------------------------------ *
+ * Classification + partition — replaces TransmitItems' tree-descent loop
+ * (thrill/api/sort.hpp:434-535). bucket(item i) = #{ j : (splitter_key[j],
+ * splitter_idx[j]) < (key_i, gidx0+i) lexicographically }, which is exactly
+ * the tree descent + EqualSampleGreaterIndex walk (sort.hpp:424-426,
+ * 487-501) in closed form (proof: oracle/t9_oracle.cpp, cross-checked
+ * against the literal tree restatement in tests).
+ * ------------------------------------------------------------------ */
+
+/* d_bucket[i] = bucket of key i; d_counts[p] (u64, zeroed by the call)
+ * accumulates per-bucket totals. p <= 256. */
+int t9_classify_u64(t9_context* ctx, const uint64_t* d_keys, uint64_t n,
+                    uint64_t gidx0, const uint64_t* d_spl_keys,
+                    const uint64_t* d_spl_idx, uint32_t p,
+                    uint32_t* d_bucket, uint64_t* d_counts, void* stream);
+
+/* Workspace bytes for t9_partition_idx. */
+uint64_t t9_partition_idx_workspace(uint64_t n);
+/* Stable counting-sort of the identity permutation by bucket id:
+ * d_perm receives record indices grouped by bucket (bucket-major, original
+ * order within a bucket — the scatter of SURVEY.md §8b t9_scatter).
+ * d_offsets[p+1] (u64) receives the exclusive bucket start offsets. */
+int t9_partition_idx(t9_context* ctx, const uint32_t* d_bucket, uint64_t n,
+                     uint32_t p, uint32_t* d_perm, uint64_t* d_offsets,
+                     void* d_workspace, void* stream);
+
+/* ------------------------------------------------------------------ *
+ * Shuffle — replaces the CatStream/MixStream + Multiplexer TCP exchange
+ * (thrill/data/stream_sink.cpp:97-226, multiplexer.cpp:282-463) with an
+ * RCCL all-to-all-v over xGMI (grouped ncclSend/ncclRecv), one rank per
+ * GPU. counts are element counts per destination rank (host memory);
+ * displacements are their exclusive prefix sums (caller-computed).
+ * ------------------------------------------------------------------ */
+int t9_alltoall(t9_context* ctx, const void* d_send,
+                const uint64_t* send_counts, const uint64_t* send_displs,
+                void* d_recv, const uint64_t* recv_counts,
+                const uint64_t* recv_displs, uint64_t elem_size,
+                void* stream);
+
+/* ------------------------------------------------------------------ *
+ * Reduce — replaces ReduceProbingHashTable::Insert
+ * (thrill/core/reduce_probing_hash_table.hpp:190-268) with a device
+ * open-addressing table: linear probing on Hash128to64(salt, key)
+ * (thrill/common/hash.hpp:64-72; index mapping
+ * core/reduce_functional.hpp:60-72), wave-level pre-combination of equal
+ * keys (ballot match + shuffle reduce) before one atomic CAS/ADD per
+ * distinct key per wave — the skew control for Zipf keys (SURVEY.md §7
+ * step 6). The empty-slot sentinel key 0xFFFF..F is reduced in a dedicated
+ * extra slot, mirroring reduce_probing_hash_table.hpp:195-217.
+ * Table arrays have capacity+1 entries; capacity must be a power of two
+ * >= 2x the number of distinct keys (no grow/spill: 288 GB HBM holds the
+ * table — reference grow/spill machinery is subsumed by sizing).
+ * ------------------------------------------------------------------ */
+int t9_reduce_init(t9_context* ctx, uint64_t* d_table_keys,
+                   uint64_t* d_table_vals, uint64_t capacity, void* stream);
+/* Accumulate n (key, value) pairs into the table; value reduce = u64 add.
+ * d_error (device u32, zeroed by the call) is set nonzero if the table
+ * overflows. */
+int t9_reduce_build(t9_context* ctx, const uint64_t* d_keys,
+                    const uint64_t* d_vals, uint64_t n,
+                    uint64_t* d_table_keys, uint64_t* d_table_vals,
+                    uint64_t capacity, uint64_t salt, uint32_t* d_error,
+                    void* stream);
+/* Compact occupied slots to (key, value) arrays (unordered);
+ * d_out_n (device u64) receives the count. */
+int t9_reduce_drain(t9_context* ctx, const uint64_t* d_table_keys,
+                    const uint64_t* d_table_vals, uint64_t capacity,
+                    uint64_t* d_out_keys, uint64_t* d_out_vals,
+                    uint64_t* d_out_n, void* stream);
+
+/* Zipf(s, q, N) token sampling by inverse CDF (bit-identical to the
+ * oracle's t9o_zipf_tokens given the same d_cdf table — the CDF itself is
+ * computed once by the oracle/host and copied to the device). Restates
+ * thrill/common/zipf_distribution.hpp:55-120's mass function. */
+int t9_zipf_tokens(t9_context* ctx, uint64_t* d_out, const double* d_cdf,
+                   uint64_t N, uint64_t index0, uint64_t n, uint64_t seed,
+                   void* stream);
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* THRILL_AMD_H */

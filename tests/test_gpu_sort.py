@@ -1,0 +1,260 @@
+"""GPU parity tests: the HIP path (libt9 C ABI) vs the CPU oracle on
+identical seeded inputs. Bit-exact everywhere (integer/byte work).
+
+Parity definition: Sort — byte-identical output under the
+acceptance total order; classification — element-wise equal bucket ids.
+"""
+import ctypes
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from tests import _gpu as G
+    from thrill_amd import Native
+
+
+@pytest.fixture(scope="module")
+def nat():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    n = Native(device=0)
+    yield n
+    n.close()
+
+
+# ------------------------------------------------------------- generators
+
+def test_gen_u64_matches_oracle(nat, oracle):
+    n = 1 << 20
+    d = G.empty(n, np.uint64)
+    nat.gen_u64(G.ptr(d), 0, n, 0x7421, G.stream())
+    got = G.host(d, np.uint64)
+    assert np.array_equal(got, oracle.gen_u64(n, seed=0x7421))
+
+
+def test_gen_u64_offset_matches_oracle(nat, oracle):
+    n = 4096
+    d = G.empty(n, np.uint64)
+    nat.gen_u64(G.ptr(d), 1000, n, 5, G.stream())
+    assert np.array_equal(G.host(d, np.uint64),
+                          oracle.gen_u64(n, seed=5, index0=1000))
+
+
+def test_gen_records_matches_oracle(nat, oracle):
+    n = 100_000
+    d = G.empty(n * 100, np.uint8)
+    nat.gen_records(G.ptr(d), 0, n, 42, G.stream())
+    got = G.host(d, np.uint8).reshape(n, 100)
+    assert np.array_equal(got, oracle.gen_records(n, seed=42))
+
+
+# ------------------------------------------------------------- u64 sort
+
+@pytest.mark.parametrize("n", [0, 1, 2, 63, 64, 255, 4096, 4097,
+                               1_000_003, 1 << 22])
+def test_sort_u64_parity(nat, oracle, n):
+    keys = oracle.gen_u64(n, seed=n + 1) if n else np.empty(0, np.uint64)
+    d = G.dev(keys) if n else G.empty(0, np.uint64)
+    w = G.ws(nat.ws("sort_u64", n))
+    nat.sort_u64(G.ptr(d), n, G.ptr(w), G.stream())
+    got = G.host(d, np.uint64)
+    assert np.array_equal(got, np.sort(keys))
+
+
+def test_sort_u64_known_integer_identity(nat):
+    # sort_node_test.cpp:25-53 restated: reversed known integers -> identity
+    n = 1 << 21
+    keys = np.arange(n - 1, -1, -1, dtype=np.uint64)
+    d = G.dev(keys)
+    w = G.ws(nat.ws("sort_u64", n))
+    nat.sort_u64(G.ptr(d), n, G.ptr(w), G.stream())
+    assert np.array_equal(G.host(d, np.uint64),
+                          np.arange(n, dtype=np.uint64))
+
+
+@pytest.mark.parametrize("case", ["all_equal", "few_values", "presorted",
+                                  "reverse_u32_range", "high_bits_only"])
+def test_sort_u64_distributions(nat, case):
+    n = 1 << 18
+    rng = np.random.default_rng(7)
+    if case == "all_equal":
+        keys = np.full(n, 0xDEADBEEF, dtype=np.uint64)
+    elif case == "few_values":
+        keys = rng.integers(0, 4, n).astype(np.uint64)
+    elif case == "presorted":
+        keys = np.sort(rng.integers(0, 1 << 63, n).astype(np.uint64))
+    elif case == "reverse_u32_range":
+        keys = np.arange(n, dtype=np.uint64)[::-1].copy()
+    else:
+        keys = (rng.integers(0, 1 << 16, n).astype(np.uint64)) << 48
+    d = G.dev(keys)
+    w = G.ws(nat.ws("sort_u64", n))
+    nat.sort_u64(G.ptr(d), n, G.ptr(w), G.stream())
+    assert np.array_equal(G.host(d, np.uint64), np.sort(keys))
+
+
+# ------------------------------------------------------------- pair sort
+
+def test_sort_pairs_stability(nat):
+    # stable by construction: equal keys keep ascending payload order
+    n = 1 << 18
+    rng = np.random.default_rng(3)
+    keys = rng.integers(0, 64, n).astype(np.uint64)
+    vals = np.arange(n, dtype=np.uint32)
+    dk, dv = G.dev(keys), G.dev(vals)
+    w = G.ws(nat.ws("sort_pairs", n))
+    nat.sort_pairs_u64_u32(G.ptr(dk), G.ptr(dv), n, G.ptr(w), G.stream())
+    gk, gv = G.host(dk, np.uint64), G.host(dv, np.uint32)
+    assert np.array_equal(gk, np.sort(keys))
+    # numpy stable argsort is the stability oracle
+    order = np.argsort(keys, kind="stable").astype(np.uint32)
+    assert np.array_equal(gv, order)
+
+
+# ------------------------------------------------------------- record sort
+
+@pytest.mark.parametrize("n", [0, 1, 2, 1000, 100_000])
+def test_sort_records_parity(nat, oracle, n):
+    recs = oracle.gen_records(n, seed=n + 9) if n \
+        else np.empty((0, 100), np.uint8)
+    din = G.dev(recs.reshape(-1)) if n else G.empty(0, np.uint8)
+    dout = G.empty(max(n, 1) * 100, np.uint8)
+    w = G.ws(nat.ws("sort_records", n, 100))
+    nat.sort_records(G.ptr(din), G.ptr(dout), n, 100, 10, G.ptr(w),
+                     G.stream())
+    if n:
+        got = G.host(dout, np.uint8)[:n * 100].reshape(n, 100)
+        assert np.array_equal(got, oracle.sort_records(recs))
+
+
+def test_sort_records_equal_key_ties(nat, oracle):
+    # adversarial: many records share the full 10-byte key; acceptance
+    # order = full-record lexicographic (ties resolved by value bytes).
+    n = 4096
+    recs = oracle.gen_records(n, seed=1)
+    recs[:, :10] = 0x55            # all keys identical
+    rng = np.random.default_rng(2)
+    recs[:, 10:] = rng.integers(0, 256, (n, 90)).astype(np.uint8)
+    din = G.dev(recs.reshape(-1))
+    dout = G.empty(n * 100, np.uint8)
+    w = G.ws(nat.ws("sort_records", n, 100))
+    nat.sort_records(G.ptr(din), G.ptr(dout), n, 100, 10, G.ptr(w),
+                     G.stream())
+    got = G.host(dout, np.uint8).reshape(n, 100)
+    assert np.array_equal(got, oracle.sort_records(recs))
+
+
+def test_sort_records_prefix_collisions(nat, oracle):
+    # records with equal u64 prefixes but different key bytes 8..9:
+    # exercises the tie-fix path's key-tail ordering.
+    n = 2048
+    recs = oracle.gen_records(n, seed=4)
+    recs[:, :8] = np.tile(np.arange(16, dtype=np.uint8), (n, 1))[:, :8]
+    din = G.dev(recs.reshape(-1))
+    dout = G.empty(n * 100, np.uint8)
+    w = G.ws(nat.ws("sort_records", n, 100))
+    nat.sort_records(G.ptr(din), G.ptr(dout), n, 100, 10, G.ptr(w),
+                     G.stream())
+    got = G.host(dout, np.uint8).reshape(n, 100)
+    assert np.array_equal(got, oracle.sort_records(recs))
+
+
+def test_extract_key64_bigendian(nat, oracle):
+    n = 10_000
+    recs = oracle.gen_records(n, seed=6)
+    din = G.dev(recs.reshape(-1))
+    dk, di = G.empty(n, np.uint64), G.empty(n, np.uint32)
+    nat.extract_key64(G.ptr(din), n, 100, 0, G.ptr(dk), G.ptr(di),
+                      G.stream())
+    gk = G.host(dk, np.uint64)
+    expect = np.array([int.from_bytes(r[:8].tobytes(), "big")
+                       for r in recs], dtype=np.uint64)
+    assert np.array_equal(gk, expect)
+    assert np.array_equal(G.host(di, np.uint32),
+                          np.arange(n, dtype=np.uint32))
+
+
+# ------------------------------------------------------------- classify
+
+@pytest.mark.parametrize("p", [2, 3, 5, 8])
+def test_classify_parity(nat, oracle, p):
+    n = 200_000
+    rng = np.random.default_rng(p)
+    # duplicate-heavy keys to exercise the (key, gidx) tiebreak
+    keys = rng.integers(0, 97, n).astype(np.uint64)
+    pos = rng.choice(n, 128, replace=False).astype(np.uint64)
+    sk = keys[pos.astype(np.int64)]
+    spl_k, spl_i = oracle.select_splitters_u64(sk, pos, p)
+    gidx0 = 10_000
+    expect = oracle.classify_u64(keys, gidx0, spl_k, spl_i, p)
+
+    dk = G.dev(keys)
+    dsk, dsi = G.dev(spl_k), G.dev(spl_i)
+    db = G.empty(n, np.uint32)
+    dc = G.empty(p, np.uint64)
+    nat.classify_u64(G.ptr(dk), n, gidx0, G.ptr(dsk), G.ptr(dsi), p,
+                     G.ptr(db), G.ptr(dc), G.stream())
+    got = G.host(db, np.uint32)
+    assert np.array_equal(got, expect)
+    counts = G.host(dc, np.uint64)
+    assert np.array_equal(counts, np.bincount(expect, minlength=p)
+                          .astype(np.uint64))
+
+
+@pytest.mark.parametrize("p", [2, 8])
+def test_partition_idx_stable(nat, oracle, p):
+    n = 100_000
+    rng = np.random.default_rng(p + 50)
+    bucket = rng.integers(0, p, n).astype(np.uint32)
+    db = G.dev(bucket)
+    dperm = G.empty(n, np.uint32)
+    doffs = G.empty(p + 1, np.uint64)
+    w = G.ws(nat.ws("partition_idx", n))
+    nat.partition_idx(G.ptr(db), n, p, G.ptr(dperm), G.ptr(doffs),
+                      G.ptr(w), G.stream())
+    perm = G.host(dperm, np.uint32)
+    offs = G.host(doffs, np.uint64)
+    expect_offs = np.concatenate(
+        [[0], np.cumsum(np.bincount(bucket, minlength=p))]).astype(np.uint64)
+    assert np.array_equal(offs, expect_offs)
+    # stable grouping: within each bucket original order is preserved
+    expect_perm = np.argsort(bucket, kind="stable").astype(np.uint32)
+    assert np.array_equal(perm, expect_perm)
+
+
+def test_end_to_end_single_gpu_sample_sort(nat, oracle):
+    """Full single-GPU pipeline at p=4 virtual buckets: classify ->
+    partition -> per-bucket sort -> concat == total sort (the reference
+    MainOp semantics, api/sort.hpp:537-663)."""
+    n = 500_000
+    keys = oracle.gen_u64(n, seed=99)
+    # worker-style samples: every k-th key, global indices = positions
+    pos = np.arange(0, n, n // 512, dtype=np.uint64)[:512]
+    p = 4
+    spl_k, spl_i = oracle.select_splitters_u64(
+        keys[pos.astype(np.int64)], pos, p)
+    dk = G.dev(keys)
+    dsk, dsi = G.dev(spl_k), G.dev(spl_i)  # keep alive: kernels are queued
+    db = G.empty(n, np.uint32)
+    dc = G.empty(p, np.uint64)
+    nat.classify_u64(G.ptr(dk), n, 0, G.ptr(dsk), G.ptr(dsi), p,
+                     G.ptr(db), G.ptr(dc), G.stream())
+    dperm = G.empty(n, np.uint32)
+    doffs = G.empty(p + 1, np.uint64)
+    w = G.ws(max(nat.ws("partition_idx", n), nat.ws("sort_u64", n)))
+    nat.partition_idx(G.ptr(db), n, p, G.ptr(dperm), G.ptr(doffs),
+                      G.ptr(w), G.stream())
+    perm = G.host(dperm, np.uint32)
+    offs = G.host(doffs, np.uint64)
+    grouped = keys[perm.astype(np.int64)]
+    out = []
+    for b in range(p):
+        part = grouped[int(offs[b]):int(offs[b + 1])]
+        dpart = G.dev(part)
+        nat.sort_u64(G.ptr(dpart), len(part), G.ptr(w), G.stream())
+        out.append(G.host(dpart, np.uint64))
+    assert np.array_equal(np.concatenate(out), np.sort(keys))

@@ -1,0 +1,283 @@
+/* t9_records.hip — 100-byte record path: device-side generation (layout
+ * restated from examples/terasort/terasort.cpp:31-118, key bytes from a
+ * seeded counter-based splitmix64 identical to oracle/t9_oracle.cpp),
+ * key-prefix extraction, payload gather and the end-to-end record sort
+ * (SURVEY.md §7 step 4). Zipf token sampling for the word_count input
+ * (thrill/common/zipf_distribution.hpp:55-120 mass function by inverse CDF).
+ */
+
+#include "t9_common.h"
+
+#include <cstdlib>
+#include <cstring>
+#include <vector>
+#include <algorithm>
+
+extern "C" int t9i_count_tied(const u64* d_keys, u64 n, u32* d_ntied,
+                              hipStream_t s);
+extern "C" u64 t9_sort_pairs_workspace(u64 n);
+extern "C" int t9_sort_pairs_u64_u32(t9_context*, u64*, u32*, u64, void*,
+                                     void*);
+
+/* ------------------------------------------------------------------ */
+
+__global__ __launch_bounds__(256) void k_gen_u64(u64* __restrict__ out,
+                                                 u64 index0, u64 n,
+                                                 u64 seed) {
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 i = (u64)blockIdx.x * 256 + threadIdx.x; i < n; i += stride)
+        out[i] = t9_splitmix64_at(seed, index0 + i);
+}
+
+/* One record per thread, staged through LDS so global writes are
+ * word-coalesced. Value layout: terasort.cpp:73-111. */
+__global__ __launch_bounds__(256) void k_gen_records(u8* __restrict__ out,
+                                                     u64 index0, u64 n,
+                                                     u64 seed) {
+    __shared__ u8 buf[256 * 100];
+    const u32 tid = threadIdx.x;
+    const u64 blk0 = (u64)blockIdx.x * 256;
+    const u64 r = blk0 + tid;
+    const u32 nrec = (u32)((n - blk0 < 256) ? (n - blk0) : 256);
+    if (r < n) {
+        const u64 rec = index0 + r;
+        u8* o = buf + tid * 100;
+        u64 k0 = t9_splitmix64_at(seed, 2 * rec);
+        u64 k1 = t9_splitmix64_at(seed, 2 * rec + 1);
+        for (int j = 0; j < 8; ++j) o[j] = (u8)(k0 >> (56 - 8 * j));
+        o[8] = (u8)(k1 >> 56);
+        o[9] = (u8)(k1 >> 48);
+        u8* v = o + 10;
+        const char* hexd = "0123456789ABCDEF";
+        *v++ = 0x00;
+        *v++ = 0x11;
+        for (int j = 0; j != 16; ++j) *v++ = hexd[(rec >> (4 * j)) & 0x0F];
+        for (int j = 0; j != 16; ++j) *v++ = '0';
+        *v++ = 0x88; *v++ = 0x99; *v++ = 0xAA; *v++ = 0xBB;
+        for (int j = 0; j < 12; ++j) {
+            u8 f = hexd[((20 + rec) >> (4 * j)) & 0x0F];
+            *v++ = f; *v++ = f; *v++ = f; *v++ = f;
+        }
+        *v++ = 0xCC; *v++ = 0xDD; *v++ = 0xEE; *v++ = 0xFF;
+    }
+    __syncthreads();
+    const u32 nb = nrec * 100;
+    u32* gout = (u32*)(out + blk0 * 100);
+    const u32* gin = (const u32*)buf;
+    for (u32 w = tid; w * 4 < nb; w += 256) gout[w] = gin[w];
+}
+
+/* big-endian u64 key prefix at key_off + record index iota */
+__global__ __launch_bounds__(256) void k_extract_key64(
+    const u8* __restrict__ recs, u64 n, u32 rec_words, u32 key_off,
+    u64* __restrict__ keys, u32* __restrict__ idx) {
+    const u64 stride = (u64)gridDim.x * 256;
+    const u32* r32 = (const u32*)recs;
+    for (u64 i = (u64)blockIdx.x * 256 + threadIdx.x; i < n; i += stride) {
+        u64 w = (u64)i * rec_words + key_off / 4;
+        u32 w0 = r32[w], w1 = r32[w + 1];
+        keys[i] = ((u64)__builtin_bswap32(w0) << 32) | __builtin_bswap32(w1);
+        idx[i] = (u32)i;
+    }
+}
+
+/* out[i] = recs[idx[i]], word-wise: consecutive threads write consecutive
+ * output words; reads touch whole source records (contiguous lines). */
+__global__ __launch_bounds__(256) void k_gather_records(
+    const u8* __restrict__ recs, const u32* __restrict__ idx, u64 n,
+    u32 rec_words, u8* __restrict__ out) {
+    const u64 total_words = n * rec_words;
+    const u64 stride = (u64)gridDim.x * 256;
+    const u32* rin = (const u32*)recs;
+    u32* rout = (u32*)out;
+    for (u64 g = (u64)blockIdx.x * 256 + threadIdx.x; g < total_words;
+         g += stride) {
+        u64 rec = g / rec_words;
+        u32 w = (u32)(g - rec * rec_words);
+        rout[g] = rin[(u64)idx[rec] * rec_words + w];
+    }
+}
+
+__global__ __launch_bounds__(256) void k_zipf_tokens(
+    u64* __restrict__ out, const double* __restrict__ cdf, u64 N, u64 index0,
+    u64 n, u64 seed) {
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 i = (u64)blockIdx.x * 256 + threadIdx.x; i < n; i += stride) {
+        double u = (double)(t9_splitmix64_at(seed, index0 + i) >> 11) *
+                   (1.0 / 9007199254740992.0);
+        u64 lo = 0, hi = N - 1;
+        while (lo < hi) {
+            u64 mid = (lo + hi) / 2;
+            if (cdf[mid] > u) hi = mid; else lo = mid + 1;
+        }
+        out[i] = lo + 1;
+    }
+}
+
+/* ------------------------------------------------------------------ *
+ * host orchestration
+ * ------------------------------------------------------------------ */
+
+namespace {
+u32 grid_for(u64 work) {
+    u64 want = t9_ceil_div(work, 256);
+    return (u32)((want < 4096) ? (want ? want : 1) : 4096);
+}
+} // namespace
+
+extern "C" {
+
+int t9_gen_u64(t9_context* ctx, u64* d_out, u64 index0, u64 n, u64 seed,
+               void* stream) {
+    (void)ctx;
+    if (!d_out) return T9_EINVAL;
+    if (n == 0) return T9_OK;
+    hipLaunchKernelGGL(k_gen_u64, dim3(grid_for(n)), dim3(256), 0,
+                       (hipStream_t)stream, d_out, index0, n, seed);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+int t9_gen_records(t9_context* ctx, u8* d_out, u64 index0, u64 n, u64 seed,
+                   void* stream) {
+    (void)ctx;
+    if (!d_out) return T9_EINVAL;
+    if (n == 0) return T9_OK;
+    u64 blocks = t9_ceil_div(n, 256);
+    if (blocks >= (1ull << 31)) return T9_EINVAL;
+    hipLaunchKernelGGL(k_gen_records, dim3((u32)blocks), dim3(256), 0,
+                       (hipStream_t)stream, d_out, index0, n, seed);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+int t9_extract_key64(t9_context* ctx, const u8* d_recs, u64 n, u32 rec_size,
+                     u32 key_off, u64* d_keys, u32* d_idx, void* stream) {
+    (void)ctx;
+    if (!d_recs || !d_keys || !d_idx) return T9_EINVAL;
+    if (rec_size % 4 || key_off % 4 || key_off + 8 > rec_size)
+        return T9_EINVAL;
+    if (n == 0) return T9_OK;
+    hipLaunchKernelGGL(k_extract_key64, dim3(grid_for(n)), dim3(256), 0,
+                       (hipStream_t)stream, d_recs, n, rec_size / 4, key_off,
+                       d_keys, d_idx);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+int t9_gather_records(t9_context* ctx, const u8* d_recs, const u32* d_idx,
+                      u64 n, u32 rec_size, u8* d_out, void* stream) {
+    (void)ctx;
+    if (!d_recs || !d_idx || !d_out || rec_size % 4) return T9_EINVAL;
+    if (n == 0) return T9_OK;
+    hipLaunchKernelGGL(k_gather_records,
+                       dim3(grid_for(n * (rec_size / 4))), dim3(256), 0,
+                       (hipStream_t)stream, d_recs, d_idx, n, rec_size / 4,
+                       d_out);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+int t9_zipf_tokens(t9_context* ctx, u64* d_out, const double* d_cdf, u64 N,
+                   u64 index0, u64 n, u64 seed, void* stream) {
+    (void)ctx;
+    if (!d_out || !d_cdf || N == 0) return T9_EINVAL;
+    if (n == 0) return T9_OK;
+    hipLaunchKernelGGL(k_zipf_tokens, dim3(grid_for(n)), dim3(256), 0,
+                       (hipStream_t)stream, d_out, d_cdf, N, index0, n,
+                       seed);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+u64 t9_sort_records_workspace(u64 n, u32 rec_size) {
+    (void)rec_size;
+    if (n < 2) return 256;
+    return t9_align256(n * 8) + t9_align256(n * 4) +
+           t9_sort_pairs_workspace(n) + 256 /* ntied */;
+}
+
+/* Sort records by the acceptance total order (full-record lexicographic;
+ * the u64 big-endian prefix of the key is the radix sort key, stability of
+ * the LSD pipeline keeps equal-prefix records in input order, and the rare
+ * equal-prefix runs are re-ordered by comparing the remaining bytes — a
+ * host pass, since at the benchmark's uniform-key sizes the expected number
+ * of colliding 8-byte prefixes is < 1 (SURVEY.md §7 hard part (a));
+ * adversarial all-equal inputs take the slow path but stay correct). */
+int t9_sort_records(t9_context* ctx, const u8* d_in, u8* d_out, u64 n,
+                    u32 rec_size, u32 key_len, void* d_workspace,
+                    void* stream) {
+    if (!d_in || !d_out || !d_workspace) return T9_EINVAL;
+    if (rec_size % 4 || key_len > rec_size || n >= (1ull << 32))
+        return T9_EINVAL;
+    hipStream_t s = (hipStream_t)stream;
+    if (n == 0) return T9_OK;
+    if (n == 1) {
+        HIP_TRY(hipMemcpyAsync(d_out, d_in, rec_size, hipMemcpyDeviceToDevice,
+                               s));
+        return T9_OK;
+    }
+    char* p = (char*)d_workspace;
+    u64* d_keys = (u64*)p;
+    p += t9_align256(n * 8);
+    u32* d_idx = (u32*)p;
+    p += t9_align256(n * 4);
+    void* pair_ws = p;
+    p += t9_sort_pairs_workspace(n);
+    u32* d_ntied = (u32*)p;
+
+    int rc = t9_extract_key64(ctx, d_in, n, rec_size, 0, d_keys, d_idx,
+                              stream);
+    if (rc) return rc;
+    rc = t9_sort_pairs_u64_u32(ctx, d_keys, d_idx, n, pair_ws, stream);
+    if (rc) return rc;
+    rc = t9i_count_tied(d_keys, n, d_ntied, s);
+    if (rc) return rc;
+    u32 ntied = 0;
+    HIP_TRY(hipMemcpyAsync(&ntied, d_ntied, 4, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+
+    if (ntied) {
+        /* slow path: re-order equal-prefix runs on the host by the bytes
+         * beyond the u64 prefix */
+        std::vector<u64> hk(n);
+        std::vector<u32> hi(n);
+        HIP_TRY(hipMemcpy(hk.data(), d_keys, n * 8, hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(hi.data(), d_idx, n * 4, hipMemcpyDeviceToHost));
+        const u32 rest = rec_size - 8;
+        std::vector<u8> recbuf;
+        bool changed = false;
+        for (u64 a = 0; a < n;) {
+            u64 b = a + 1;
+            while (b < n && hk[b] == hk[a]) ++b;
+            if (b - a > 1) {
+                u64 len = b - a;
+                recbuf.resize(len * rest);
+                for (u64 t = 0; t < len; ++t)
+                    HIP_TRY(hipMemcpy(recbuf.data() + t * rest,
+                                      d_in + (u64)hi[a + t] * rec_size + 8,
+                                      rest, hipMemcpyDeviceToHost));
+                std::vector<u32> ord(len);
+                for (u64 t = 0; t < len; ++t) ord[t] = (u32)t;
+                std::stable_sort(ord.begin(), ord.end(),
+                                 [&](u32 x, u32 y) {
+                                     return memcmp(recbuf.data() + x * rest,
+                                                   recbuf.data() + y * rest,
+                                                   rest) < 0;
+                                 });
+                std::vector<u32> fixed(len);
+                for (u64 t = 0; t < len; ++t) fixed[t] = hi[a + ord[t]];
+                memcpy(&hi[a], fixed.data(), len * 4);
+                changed = true;
+            }
+            a = b;
+        }
+        if (changed)
+            HIP_TRY(hipMemcpy(d_idx, hi.data(), n * 4,
+                              hipMemcpyHostToDevice));
+    }
+
+    return t9_gather_records(ctx, d_in, d_idx, n, rec_size, d_out, stream);
+}
+
+} /* extern "C" */

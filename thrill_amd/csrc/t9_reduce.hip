@@ -1,0 +1,161 @@
+/* t9_reduce.hip — device open-addressing reduce table, gfx950.
+ *
+ * Replaces ReduceProbingHashTable::Insert
+ * (thrill/core/reduce_probing_hash_table.hpp:190-268): linear probing on
+ * Hash128to64(salt, key) (thrill/common/hash.hpp:64-72; index mapping per
+ * core/reduce_functional.hpp:60-72 — the (h/p) % size local index reduces
+ * to a mask of the low hash bits here since the table is power-of-two and
+ * single-partition per GPU; the partition split across ranks uses the same
+ * h % p as the reference). The reference's grow/spill machinery
+ * (:293-409) is subsumed by sizing: 288 GB HBM holds the table.
+ *
+ * Skew control: equal keys within a wavefront are
+ * pre-combined by ballot-match + shuffle reduction, so a Zipf(1.1) head
+ * key costs one atomicAdd per wave instead of 64 — the same role the
+ * reference's in-table in-place reduce (:233) plays for its cache.
+ * The empty-slot sentinel key 0xFFFF..F is reduced in a dedicated extra
+ * slot, mirroring reduce_probing_hash_table.hpp:195-217.
+ */
+
+#include "t9_common.h"
+
+#define T9_EMPTY 0xFFFFFFFFFFFFFFFFull
+
+__global__ __launch_bounds__(256) void k_reduce_init(u64* __restrict__ tk,
+                                                     u64* __restrict__ tv,
+                                                     u64 cap) {
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 i = (u64)blockIdx.x * 256 + threadIdx.x; i <= cap; i += stride) {
+        tk[i] = (i == cap) ? 0 : T9_EMPTY;  /* aux slot counts sentinel-key
+                                               occurrences */
+        tv[i] = 0;
+    }
+}
+
+__global__ __launch_bounds__(256) void k_reduce_build(
+    const u64* __restrict__ keys, const u64* __restrict__ vals, u64 n,
+    u64* __restrict__ tk, u64* __restrict__ tv, u64 cap, u64 salt,
+    u32* __restrict__ err) {
+    const u64 gsz = (u64)gridDim.x * 256;
+    const u32 lane = threadIdx.x & 63;
+    for (u64 base = (u64)blockIdx.x * 256; base < n; base += gsz) {
+        const u64 i = base + threadIdx.x;
+        const bool valid = i < n;
+        const u64 k = valid ? keys[i] : 0;
+        const u64 v = valid ? vals[i] : 0;
+
+        /* wave-level combine: iterate distinct keys present in the wave */
+        u64 pending = __ballot(valid);
+        while (pending) {
+            const u32 leader = (u32)__ffsll((unsigned long long)pending) - 1;
+            const u64 lk = __shfl(k, (int)leader);
+            const u64 grp = __ballot(valid && k == lk) & pending;
+            u64 gsum = 0;
+            u64 g = grp;
+            while (g) {
+                const u32 src = (u32)__ffsll((unsigned long long)g) - 1;
+                const u64 sv = __shfl(v, (int)src);
+                if (lane == leader) gsum += sv;
+                g &= g - 1;
+            }
+            if (lane == leader) {
+                if (lk == T9_EMPTY) {
+                    atomicAdd((unsigned long long*)&tk[cap],
+                              (unsigned long long)__popcll(grp));
+                    atomicAdd((unsigned long long*)&tv[cap],
+                              (unsigned long long)gsum);
+                }
+                else {
+                    u64 slot = t9_hash128to64(salt, lk) & (cap - 1);
+                    u64 probes = 0;
+                    for (;;) {
+                        u64 prev = atomicCAS((unsigned long long*)&tk[slot],
+                                             (unsigned long long)T9_EMPTY,
+                                             (unsigned long long)lk);
+                        if (prev == T9_EMPTY || prev == lk) {
+                            atomicAdd((unsigned long long*)&tv[slot],
+                                      (unsigned long long)gsum);
+                            break;
+                        }
+                        slot = (slot + 1) & (cap - 1);
+                        if (++probes > cap) {
+                            atomicExch(err, 1u);
+                            break;
+                        }
+                    }
+                }
+            }
+            pending &= ~grp;
+        }
+    }
+}
+
+__global__ __launch_bounds__(256) void k_reduce_drain(
+    const u64* __restrict__ tk, const u64* __restrict__ tv, u64 cap,
+    u64* __restrict__ ok, u64* __restrict__ ov, u64* __restrict__ out_n) {
+    const u64 stride = (u64)gridDim.x * 256;
+    const u64 gid = (u64)blockIdx.x * 256 + threadIdx.x;
+    for (u64 i = gid; i < cap; i += stride) {
+        if (tk[i] != T9_EMPTY) {
+            u64 pos = atomicAdd((unsigned long long*)out_n, 1ull);
+            ok[pos] = tk[i];
+            ov[pos] = tv[i];
+        }
+    }
+    if (gid == 0 && tk[cap] > 0) {
+        u64 pos = atomicAdd((unsigned long long*)out_n, 1ull);
+        ok[pos] = T9_EMPTY;
+        ov[pos] = tv[cap];
+    }
+}
+
+namespace {
+u32 grid_for(u64 work) {
+    u64 want = (work + 255) / 256;
+    return (u32)((want < 4096) ? (want ? want : 1) : 4096);
+}
+bool is_pow2(u64 x) { return x && !(x & (x - 1)); }
+} // namespace
+
+extern "C" {
+
+int t9_reduce_init(t9_context* ctx, u64* d_tk, u64* d_tv, u64 cap,
+                   void* stream) {
+    (void)ctx;
+    if (!d_tk || !d_tv || !is_pow2(cap)) return T9_EINVAL;
+    hipLaunchKernelGGL(k_reduce_init, dim3(grid_for(cap + 1)), dim3(256), 0,
+                       (hipStream_t)stream, d_tk, d_tv, cap);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+int t9_reduce_build(t9_context* ctx, const u64* d_keys, const u64* d_vals,
+                    u64 n, u64* d_tk, u64* d_tv, u64 cap, u64 salt,
+                    u32* d_error, void* stream) {
+    (void)ctx;
+    if (!d_keys || !d_vals || !d_tk || !d_tv || !d_error || !is_pow2(cap))
+        return T9_EINVAL;
+    hipStream_t s = (hipStream_t)stream;
+    HIP_TRY(hipMemsetAsync(d_error, 0, 4, s));
+    if (n == 0) return T9_OK;
+    hipLaunchKernelGGL(k_reduce_build, dim3(grid_for(n)), dim3(256), 0, s,
+                       d_keys, d_vals, n, d_tk, d_tv, cap, salt, d_error);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+int t9_reduce_drain(t9_context* ctx, const u64* d_tk, const u64* d_tv,
+                    u64 cap, u64* d_ok, u64* d_ov, u64* d_out_n,
+                    void* stream) {
+    (void)ctx;
+    if (!d_tk || !d_tv || !d_ok || !d_ov || !d_out_n || !is_pow2(cap))
+        return T9_EINVAL;
+    hipStream_t s = (hipStream_t)stream;
+    HIP_TRY(hipMemsetAsync(d_out_n, 0, 8, s));
+    hipLaunchKernelGGL(k_reduce_drain, dim3(grid_for(cap)), dim3(256), 0, s,
+                       d_tk, d_tv, cap, d_ok, d_ov, d_out_n);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+} /* extern "C" */

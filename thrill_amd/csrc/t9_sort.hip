@@ -1,0 +1,455 @@
+/* t9_sort.hip — LSD radix sort + splitter classification kernels, gfx950.
+ *
+ * MI355X-native replacement for the reference's local sort path
+ * (thrill/api/sort.hpp:665-786 SortAndWriteToFile/std::sort + PushData
+ * loser-tree merge via core/multiway_merge.hpp:30-116): the whole per-GPU
+ * partition is sorted in one LSD radix pipeline (8-bit digits, stable
+ * within-tile ranking by wavefront ballot/popcount, LDS-staged reorder so
+ * global writes are digit-run coalesced), so the run/merge split of the
+ * reference — an artifact of bounded RAM — vanishes in 288 GB HBM3E.
+ *
+ * Classification (k_classify) replaces TransmitItems' tree-descent +
+ * EqualSampleGreaterIndex walk (thrill/api/sort.hpp:434-535, :424-426) by
+ * the equivalent closed form: bucket = #{ j : (splitter_key[j],
+ * splitter_idx[j]) < (key, gidx) lexicographically }. Equivalence is
+ * cross-checked against the literal tree restatement in the oracle tests.
+ *
+ * Roofline: integer/byte work, HBM-bound; no MFMA (BASELINE.json
+ * north_star). Per pass the pipeline moves: hist read 8 B + scatter read
+ * 8 B (L2-resident re-read inside the same block) + write 8 B per key,
+ * plus ~3% hist/scan traffic. 8 passes for a full u64.
+ */
+
+#include "t9_common.h"
+
+#include <algorithm>
+#include <cstring>
+
+/* ------------------------------------------------------------------ *
+ * kernels
+ * ------------------------------------------------------------------ */
+
+template <int TILE, bool EXT_DIGIT>
+__global__ __launch_bounds__(256) void k_hist(
+    const u64* __restrict__ in_keys, const u32* __restrict__ ext_digit,
+    u64 n, u32 shift, u32* __restrict__ hist) {
+    __shared__ u32 s_cnt[T9_RADIX];
+    const u32 tid = threadIdx.x;
+    const u64 base = (u64)blockIdx.x * TILE;
+    const u32 tn = (u32)((n - base < (u64)TILE) ? (n - base) : (u64)TILE);
+    s_cnt[tid] = 0;
+    __syncthreads();
+    for (u32 i = tid; i < tn; i += 256) {
+        u32 d = EXT_DIGIT ? ext_digit[base + i]
+                          : ((u32)(in_keys[base + i] >> shift) & 255u);
+        atomicAdd(&s_cnt[d], 1u);
+    }
+    __syncthreads();
+    hist[(u64)blockIdx.x * T9_RADIX + tid] = s_cnt[tid];
+}
+
+/* column (per-digit) partial sums over T9_SCAN_CHUNK hist rows */
+__global__ __launch_bounds__(256) void k_colsum(
+    const u32* __restrict__ hist, u64 B, u32* __restrict__ chunkpart) {
+    const u32 tid = threadIdx.x;
+    const u64 r0 = (u64)blockIdx.x * T9_SCAN_CHUNK;
+    const u64 r1 = (r0 + T9_SCAN_CHUNK < B) ? r0 + T9_SCAN_CHUNK : B;
+    u32 sum = 0;
+    for (u64 r = r0; r < r1; ++r) sum += hist[r * T9_RADIX + tid];
+    chunkpart[(u64)blockIdx.x * T9_RADIX + tid] = sum;
+}
+
+/* single block: exclusive scan of chunk partials per digit (in place) and
+ * exclusive digit base offsets from the column totals */
+__global__ __launch_bounds__(256) void k_chunkscan(
+    u32* __restrict__ chunkpart, u64 Bc, u32* __restrict__ digit_base) {
+    const u32 tid = threadIdx.x;
+    u32 running = 0;
+    for (u64 c = 0; c < Bc; ++c) {
+        u32 v = chunkpart[c * T9_RADIX + tid];
+        chunkpart[c * T9_RADIX + tid] = running;
+        running += v;
+    }
+    __shared__ u32 s[T9_RADIX];
+    s[tid] = running;
+    __syncthreads();
+    for (int off = 1; off < T9_RADIX; off <<= 1) {
+        u32 y = (tid >= (u32)off) ? s[tid - off] : 0;
+        __syncthreads();
+        s[tid] += y;
+        __syncthreads();
+    }
+    digit_base[tid] = s[tid] - running;   /* exclusive */
+}
+
+/* rewrite hist rows into final global exclusive offsets per (block, digit) */
+__global__ __launch_bounds__(256) void k_finaloffs(
+    u32* __restrict__ hist, u64 B, const u32* __restrict__ chunkpart,
+    const u32* __restrict__ digit_base) {
+    const u32 tid = threadIdx.x;
+    const u64 r0 = (u64)blockIdx.x * T9_SCAN_CHUNK;
+    const u64 r1 = (r0 + T9_SCAN_CHUNK < B) ? r0 + T9_SCAN_CHUNK : B;
+    u32 run = chunkpart[(u64)blockIdx.x * T9_RADIX + tid] + digit_base[tid];
+    for (u64 r = r0; r < r1; ++r) {
+        u32 v = hist[r * T9_RADIX + tid];
+        hist[r * T9_RADIX + tid] = run;
+        run += v;
+    }
+}
+
+/* Stable scatter of one radix pass. Each 256-thread block owns a TILE:
+ * (1) load + LDS digit cache + per-digit counts, (2) LDS exclusive scan,
+ * (3) stable intra-tile ranking — per wave a ballot over the 8 digit bits
+ * yields the same-digit lane mask; rank = running count + earlier-wave
+ * counts + popcount of lower same-digit lanes — and reorder into an LDS
+ * staging tile, (4) digit-run coalesced global writes at the scanned
+ * offsets. */
+template <int TILE, bool HAS_KEY, bool HAS_VAL, bool EXT_DIGIT, bool IOTA_VAL>
+__global__ __launch_bounds__(256, 2) void k_scatter(
+    const u64* __restrict__ in_keys, const u32* __restrict__ in_vals,
+    const u32* __restrict__ ext_digit, u64* __restrict__ out_keys,
+    u32* __restrict__ out_vals, const u32* __restrict__ offs, u64 n,
+    u32 shift) {
+    constexpr int CHUNKS = TILE / 256;
+    __shared__ u64 s_keys[HAS_KEY ? TILE : 1];
+    __shared__ u32 s_vals[HAS_VAL ? TILE : 1];
+    __shared__ u64 s_okeys[HAS_KEY ? TILE : 1];
+    __shared__ u32 s_ovals[HAS_VAL ? TILE : 1];
+    __shared__ u8 s_dig[TILE];
+    __shared__ u8 s_digof[TILE];
+    __shared__ u32 s_cnt[T9_RADIX];
+    __shared__ u32 s_start[T9_RADIX];
+    __shared__ u32 s_run[T9_RADIX];
+    __shared__ u32 s_wavecnt[4 * T9_RADIX];
+    __shared__ u32 s_goff[T9_RADIX];
+
+    const u32 tid = threadIdx.x;
+    const u64 base = (u64)blockIdx.x * TILE;
+    const u32 tn = (u32)((n - base < (u64)TILE) ? (n - base) : (u64)TILE);
+    const u32 wave = tid >> 6, lane = tid & 63;
+
+    s_cnt[tid] = 0;
+    s_run[tid] = 0;
+    s_goff[tid] = offs[(u64)blockIdx.x * T9_RADIX + tid];
+    __syncthreads();
+
+    for (int c = 0; c < CHUNKS; ++c) {
+        u32 i = c * 256 + tid;
+        if (i < tn) {
+            u64 k = 0;
+            if (HAS_KEY) {
+                k = in_keys[base + i];
+                s_keys[i] = k;
+            }
+            u32 d = EXT_DIGIT ? ext_digit[base + i]
+                              : ((u32)(k >> shift) & 255u);
+            if (HAS_VAL)
+                s_vals[i] = IOTA_VAL ? (u32)(base + i) : in_vals[base + i];
+            s_dig[i] = (u8)d;
+            atomicAdd(&s_cnt[d], 1u);
+        }
+    }
+    __syncthreads();
+
+    /* exclusive scan of s_cnt into s_start (Hillis-Steele, in place) */
+    {
+        u32 x = s_cnt[tid];
+        s_start[tid] = x;
+        __syncthreads();
+        for (int off = 1; off < T9_RADIX; off <<= 1) {
+            u32 y = (tid >= (u32)off) ? s_start[tid - off] : 0;
+            __syncthreads();
+            s_start[tid] += y;
+            __syncthreads();
+        }
+        u32 incl = s_start[tid];
+        __syncthreads();
+        s_start[tid] = incl - x;
+    }
+    __syncthreads();
+
+    for (int c = 0; c < CHUNKS; ++c) {
+        u32 i = c * 256 + tid;
+        bool valid = i < tn;
+        u32 d = valid ? (u32)s_dig[i] : 0u;
+        for (int w = 0; w < 4; ++w) s_wavecnt[w * T9_RADIX + tid] = 0;
+        __syncthreads();
+        u64 vmask = __ballot(valid);
+        u64 m = vmask;
+        for (int bit = 0; bit < 8; ++bit) {
+            u64 bb = __ballot((d >> bit) & 1u);
+            m &= ((d >> bit) & 1u) ? bb : ~bb;
+        }
+        u32 wave_rank = (u32)__popcll(m & ((1ull << lane) - 1ull));
+        if (valid && wave_rank == 0)
+            s_wavecnt[wave * T9_RADIX + d] = (u32)__popcll(m);
+        __syncthreads();
+        if (valid) {
+            u32 before = 0;
+            for (u32 w = 0; w < wave; ++w)
+                before += s_wavecnt[w * T9_RADIX + d];
+            u32 pos = s_start[d] + s_run[d] + before + wave_rank;
+            if (HAS_KEY) s_okeys[pos] = s_keys[i];
+            if (HAS_VAL) s_ovals[pos] = s_vals[i];
+            s_digof[pos] = (u8)d;
+        }
+        __syncthreads();
+        u32 tot = 0;
+        for (int w = 0; w < 4; ++w) tot += s_wavecnt[w * T9_RADIX + tid];
+        __syncthreads();
+        s_run[tid] += tot;
+        __syncthreads();
+    }
+
+    for (int c = 0; c < CHUNKS; ++c) {
+        u32 j = c * 256 + tid;
+        if (j < tn) {
+            u32 d = s_digof[j];
+            u64 gpos = (u64)s_goff[d] + (j - s_start[d]);
+            if (HAS_KEY) out_keys[gpos] = s_okeys[j];
+            if (HAS_VAL) out_vals[gpos] = s_ovals[j];
+        }
+    }
+}
+
+/* bucket offsets (u64, p+1 entries) from the digit base array */
+__global__ __launch_bounds__(512) void k_bucket_offsets(
+    const u32* __restrict__ digit_base, u32 p, u64 n,
+    u64* __restrict__ offsets) {
+    u32 i = threadIdx.x;
+    if (i < p) offsets[i] = digit_base[i];
+    if (i == p) offsets[p] = n;
+}
+
+/* classification: closed form of TransmitItems (api/sort.hpp:434-535) */
+__global__ __launch_bounds__(256) void k_classify(
+    const u64* __restrict__ keys, u64 n, u64 gidx0,
+    const u64* __restrict__ spl_k, const u64* __restrict__ spl_i, u32 p,
+    u32* __restrict__ bucket, u64* __restrict__ counts) {
+    __shared__ u64 sk[T9_RADIX], si[T9_RADIX];
+    __shared__ u32 scnt[T9_RADIX];
+    const u32 tid = threadIdx.x;
+    if (tid < p - 1) {
+        sk[tid] = spl_k[tid];
+        si[tid] = spl_i[tid];
+    }
+    scnt[tid] = 0;
+    __syncthreads();
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 i = (u64)blockIdx.x * 256 + tid; i < n; i += stride) {
+        u64 k = keys[i], g = gidx0 + i;
+        u32 b = 0;
+        for (u32 j = 0; j < p - 1; ++j)
+            if (sk[j] < k || (sk[j] == k && si[j] < g)) b = j + 1;
+        bucket[i] = b;
+        atomicAdd(&scnt[b], 1u);
+    }
+    __syncthreads();
+    if (tid < p && scnt[tid])
+        atomicAdd((unsigned long long*)&counts[tid],
+                  (unsigned long long)scnt[tid]);
+}
+
+/* count elements whose u64 key equals their left neighbour's (tie probe) */
+__global__ __launch_bounds__(256) void k_count_tied(
+    const u64* __restrict__ keys, u64 n, u32* __restrict__ ntied) {
+    __shared__ u32 s;
+    if (threadIdx.x == 0) s = 0;
+    __syncthreads();
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 i = (u64)blockIdx.x * 256 + threadIdx.x; i < n; i += stride)
+        if (i > 0 && keys[i] == keys[i - 1]) atomicAdd(&s, 1u);
+    __syncthreads();
+    if (threadIdx.x == 0 && s) atomicAdd(ntied, s);
+}
+
+/* ------------------------------------------------------------------ *
+ * host orchestration
+ * ------------------------------------------------------------------ */
+
+namespace {
+
+struct ScanWs {
+    u32* hist;
+    u32* chunkpart;
+    u32* digit_base;
+    u64 B, Bc;
+};
+
+u64 scan_ws_bytes(u64 B) {
+    u64 Bc = t9_ceil_div(B, T9_SCAN_CHUNK);
+    return t9_align256(B * T9_RADIX * 4) + t9_align256(Bc * T9_RADIX * 4) +
+           t9_align256(T9_RADIX * 4);
+}
+
+ScanWs carve_scan_ws(char*& p, u64 B) {
+    ScanWs w;
+    w.B = B;
+    w.Bc = t9_ceil_div(B, T9_SCAN_CHUNK);
+    w.hist = (u32*)p;
+    p += t9_align256(B * T9_RADIX * 4);
+    w.chunkpart = (u32*)p;
+    p += t9_align256(w.Bc * T9_RADIX * 4);
+    w.digit_base = (u32*)p;
+    p += t9_align256(T9_RADIX * 4);
+    return w;
+}
+
+int run_scan(const ScanWs& w, hipStream_t s) {
+    hipLaunchKernelGGL(k_colsum, dim3((u32)w.Bc), dim3(256), 0, s, w.hist,
+                       w.B, w.chunkpart);
+    hipLaunchKernelGGL(k_chunkscan, dim3(1), dim3(256), 0, s, w.chunkpart,
+                       w.Bc, w.digit_base);
+    hipLaunchKernelGGL(k_finaloffs, dim3((u32)w.Bc), dim3(256), 0, s,
+                       w.hist, w.B, w.chunkpart, w.digit_base);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+} // namespace
+
+extern "C" {
+
+u64 t9_sort_u64_workspace(u64 n) {
+    if (n < 2) return 256;
+    u64 B = t9_ceil_div(n, T9_KEYS_TILE);
+    return t9_align256(n * 8) + scan_ws_bytes(B);
+}
+
+int t9_sort_u64(t9_context* ctx, u64* d_keys, u64 n, void* d_workspace,
+                void* stream) {
+    (void)ctx;
+    if (n < 2) return T9_OK;
+    if (!d_keys || !d_workspace || n >= (1ull << 32)) return T9_EINVAL;
+    hipStream_t s = (hipStream_t)stream;
+    char* p = (char*)d_workspace;
+    u64* alt = (u64*)p;
+    p += t9_align256(n * 8);
+    const u64 B = t9_ceil_div(n, T9_KEYS_TILE);
+    ScanWs w = carve_scan_ws(p, B);
+
+    u64* bufA = d_keys;
+    u64* bufB = alt;
+    for (int pass = 0; pass < 8; ++pass) {
+        u32 shift = pass * 8;
+        hipLaunchKernelGGL((k_hist<T9_KEYS_TILE, false>), dim3((u32)B),
+                           dim3(256), 0, s, bufA, nullptr, n, shift, w.hist);
+        int rc = run_scan(w, s);
+        if (rc) return rc;
+        hipLaunchKernelGGL(
+            (k_scatter<T9_KEYS_TILE, true, false, false, false>),
+            dim3((u32)B), dim3(256), 0, s, bufA, nullptr, nullptr, bufB,
+            nullptr, w.hist, n, shift);
+        T9_LAUNCH_CHECK();
+        std::swap(bufA, bufB);
+    }
+    /* 8 passes: result is back in d_keys */
+    return T9_OK;
+}
+
+u64 t9_sort_pairs_workspace(u64 n) {
+    if (n < 2) return 256;
+    u64 B = t9_ceil_div(n, T9_PAIRS_TILE);
+    return t9_align256(n * 8) + t9_align256(n * 4) + scan_ws_bytes(B);
+}
+
+int t9_sort_pairs_u64_u32(t9_context* ctx, u64* d_keys, u32* d_vals, u64 n,
+                          void* d_workspace, void* stream) {
+    (void)ctx;
+    if (n < 2) return T9_OK;
+    if (!d_keys || !d_vals || !d_workspace || n >= (1ull << 32))
+        return T9_EINVAL;
+    hipStream_t s = (hipStream_t)stream;
+    char* p = (char*)d_workspace;
+    u64* alt_k = (u64*)p;
+    p += t9_align256(n * 8);
+    u32* alt_v = (u32*)p;
+    p += t9_align256(n * 4);
+    const u64 B = t9_ceil_div(n, T9_PAIRS_TILE);
+    ScanWs w = carve_scan_ws(p, B);
+
+    u64* kA = d_keys;
+    u64* kB = alt_k;
+    u32* vA = d_vals;
+    u32* vB = alt_v;
+    for (int pass = 0; pass < 8; ++pass) {
+        u32 shift = pass * 8;
+        hipLaunchKernelGGL((k_hist<T9_PAIRS_TILE, false>), dim3((u32)B),
+                           dim3(256), 0, s, kA, nullptr, n, shift, w.hist);
+        int rc = run_scan(w, s);
+        if (rc) return rc;
+        hipLaunchKernelGGL(
+            (k_scatter<T9_PAIRS_TILE, true, true, false, false>),
+            dim3((u32)B), dim3(256), 0, s, kA, vA, nullptr, kB, vB, w.hist,
+            n, shift);
+        T9_LAUNCH_CHECK();
+        std::swap(kA, kB);
+        std::swap(vA, vB);
+    }
+    return T9_OK;
+}
+
+int t9_classify_u64(t9_context* ctx, const u64* d_keys, u64 n, u64 gidx0,
+                    const u64* d_spl_keys, const u64* d_spl_idx, u32 p,
+                    u32* d_bucket, u64* d_counts, void* stream) {
+    (void)ctx;
+    if (!d_keys || !d_spl_keys || !d_spl_idx || !d_bucket || !d_counts)
+        return T9_EINVAL;
+    if (p < 1 || p > 256) return T9_EINVAL;
+    hipStream_t s = (hipStream_t)stream;
+    HIP_TRY(hipMemsetAsync(d_counts, 0, p * 8, s));
+    if (n == 0) return T9_OK;
+    u64 want = t9_ceil_div(n, 256);
+    u32 grid = (u32)((want < 2048) ? want : 2048);
+    hipLaunchKernelGGL(k_classify, dim3(grid), dim3(256), 0, s, d_keys, n,
+                       gidx0, d_spl_keys, d_spl_idx, p, d_bucket, d_counts);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+u64 t9_partition_idx_workspace(u64 n) {
+    if (n == 0) return 256;
+    u64 B = t9_ceil_div(n, T9_PAIRS_TILE);
+    return scan_ws_bytes(B);
+}
+
+int t9_partition_idx(t9_context* ctx, const u32* d_bucket, u64 n, u32 p,
+                     u32* d_perm, u64* d_offsets, void* d_workspace,
+                     void* stream) {
+    (void)ctx;
+    if (!d_bucket || !d_perm || !d_offsets || !d_workspace) return T9_EINVAL;
+    if (p < 1 || p > 256 || n >= (1ull << 32)) return T9_EINVAL;
+    hipStream_t s = (hipStream_t)stream;
+    if (n == 0) {
+        HIP_TRY(hipMemsetAsync(d_offsets, 0, (p + 1) * 8, s));
+        return T9_OK;
+    }
+    char* cp = (char*)d_workspace;
+    const u64 B = t9_ceil_div(n, T9_PAIRS_TILE);
+    ScanWs w = carve_scan_ws(cp, B);
+    hipLaunchKernelGGL((k_hist<T9_PAIRS_TILE, true>), dim3((u32)B),
+                       dim3(256), 0, s, nullptr, d_bucket, n, 0, w.hist);
+    int rc = run_scan(w, s);
+    if (rc) return rc;
+    hipLaunchKernelGGL((k_scatter<T9_PAIRS_TILE, false, true, true, true>),
+                       dim3((u32)B), dim3(256), 0, s, nullptr, nullptr,
+                       d_bucket, nullptr, d_perm, w.hist, n, 0);
+    hipLaunchKernelGGL(k_bucket_offsets, dim3(1), dim3(512), 0, s,
+                       w.digit_base, p, n, d_offsets);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+/* internal: used by t9_sort_records (t9_records.hip) */
+int t9i_count_tied(const u64* d_keys, u64 n, u32* d_ntied, hipStream_t s) {
+    HIP_TRY(hipMemsetAsync(d_ntied, 0, 4, s));
+    if (n < 2) return T9_OK;
+    u64 want = t9_ceil_div(n, 256);
+    u32 grid = (u32)((want < 2048) ? want : 2048);
+    hipLaunchKernelGGL(k_count_tied, dim3(grid), dim3(256), 0, s, d_keys, n,
+                       d_ntied);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+} /* extern "C" */

@@ -1,0 +1,100 @@
+"""ctypes binding of libt9.so (include/thrill_amd.h).
+
+Pointers are raw device addresses (torch tensor .data_ptr()); streams are
+hipStream_t handles (torch.cuda.Stream.cuda_stream). No CPU fallback: a
+missing library or GPU raises T9Error.
+"""
+import ctypes
+import os
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+SO = os.path.join(HERE, "libt9.so")
+
+u64 = ctypes.c_uint64
+u32 = ctypes.c_uint32
+i32 = ctypes.c_int
+vp = ctypes.c_void_p
+
+
+class T9Error(RuntimeError):
+    pass
+
+
+def lib_path():
+    return SO
+
+
+def load_lib(path=None):
+    path = path or SO
+    if not os.path.exists(path):
+        raise T9Error(
+            f"libt9.so not found at {path}; build it with "
+            "`make -C thrill_amd` (hipcc --offload-arch=gfx950). "
+            "There is no CPU fallback.")
+    return ctypes.CDLL(path)
+
+
+_SIGS = {
+    "t9_version": (ctypes.c_char_p, []),
+    "t9_create": (i32, [ctypes.POINTER(vp), i32, i32, i32, vp]),
+    "t9_destroy": (i32, [vp]),
+    "t9_gen_u64": (i32, [vp, vp, u64, u64, u64, vp]),
+    "t9_gen_records": (i32, [vp, vp, u64, u64, u64, vp]),
+    "t9_sort_u64_workspace": (u64, [u64]),
+    "t9_sort_u64": (i32, [vp, vp, u64, vp, vp]),
+    "t9_sort_pairs_workspace": (u64, [u64]),
+    "t9_sort_pairs_u64_u32": (i32, [vp, vp, vp, u64, vp, vp]),
+    "t9_extract_key64": (i32, [vp, vp, u64, u32, u32, vp, vp, vp]),
+    "t9_gather_records": (i32, [vp, vp, vp, u64, u32, vp, vp]),
+    "t9_sort_records_workspace": (u64, [u64, u32]),
+    "t9_sort_records": (i32, [vp, vp, vp, u64, u32, u32, vp, vp]),
+    "t9_classify_u64": (i32, [vp, vp, u64, u64, vp, vp, u32, vp, vp, vp]),
+    "t9_partition_idx_workspace": (u64, [u64]),
+    "t9_partition_idx": (i32, [vp, vp, u64, u32, vp, vp, vp, vp]),
+    "t9_alltoall": (i32, [vp, vp, vp, vp, vp, vp, vp, u64, vp]),
+    "t9_reduce_init": (i32, [vp, vp, vp, u64, vp]),
+    "t9_reduce_build": (i32, [vp, vp, vp, u64, vp, vp, u64, u64, vp, vp]),
+    "t9_reduce_drain": (i32, [vp, vp, vp, u64, vp, vp, vp, vp]),
+    "t9_zipf_tokens": (i32, [vp, vp, vp, u64, u64, u64, u64, vp]),
+}
+
+
+class Native:
+    """Loaded libt9 with a live t9_context. Requires a GPU."""
+
+    def __init__(self, device=0, rank=0, world=1, comm=None):
+        self._lib = load_lib()
+        for name, (res, args) in _SIGS.items():
+            fn = getattr(self._lib, name)
+            fn.restype = res
+            fn.argtypes = args
+        ctx = vp()
+        rc = self._lib.t9_create(ctypes.byref(ctx), device, rank, world,
+                                 comm if comm else None)
+        if rc != 0:
+            raise T9Error(f"t9_create failed rc={rc} (no usable GPU?)")
+        self.ctx = ctx
+
+    def __getattr__(self, name):
+        fn = getattr(self._lib, "t9_" + name, None)
+        if fn is None:
+            raise AttributeError(name)
+
+        def call(*args):
+            rc = fn(self.ctx, *args)
+            if rc != 0:
+                raise T9Error(f"t9_{name} failed rc={rc}")
+            return rc
+        return call
+
+    def ws(self, name, *args):
+        """workspace byte queries (no ctx argument)."""
+        return getattr(self._lib, f"t9_{name}_workspace")(*args)
+
+    def version(self):
+        return self._lib.t9_version().decode()
+
+    def close(self):
+        if self.ctx:
+            self._lib.t9_destroy(self.ctx)
+            self.ctx = None
