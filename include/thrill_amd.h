@@ -105,6 +105,18 @@ int t9_classify_u64(t9_context* ctx, const uint64_t* d_keys, uint64_t n,
                     const uint64_t* d_spl_idx, uint32_t p,
                     uint32_t* d_bucket, uint64_t* d_counts, void* stream);
 
+/* Record classification under the acceptance total order: primary compare
+ * on the precomputed big-endian u64 key prefix (d_k64, from
+ * t9_extract_key64), byte fallback over the remaining record bytes on
+ * prefix ties, then the splitter-index tiebreak. Splitters are whole
+ * records (as in the reference, where splitters are sampled items —
+ * api/sort.hpp:300,368-374). */
+int t9_classify_rec(t9_context* ctx, const uint8_t* d_recs,
+                    const uint64_t* d_k64, uint64_t n, uint64_t gidx0,
+                    const uint8_t* d_spl_recs, const uint64_t* d_spl_k64,
+                    const uint64_t* d_spl_idx, uint32_t p, uint32_t rec_size,
+                    uint32_t* d_bucket, uint64_t* d_counts, void* stream);
+
 /* Workspace bytes for t9_partition_idx. */
 uint64_t t9_partition_idx_workspace(uint64_t n);
 /* Stable counting-sort of the identity permutation by bucket id:

@@ -207,11 +207,11 @@ u64 t9_sort_records_workspace(u64 n, u32 rec_size) {
 int t9_sort_records(t9_context* ctx, const u8* d_in, u8* d_out, u64 n,
                     u32 rec_size, u32 key_len, void* d_workspace,
                     void* stream) {
-    if (!d_in || !d_out || !d_workspace) return T9_EINVAL;
     if (rec_size % 4 || key_len > rec_size || n >= (1ull << 32))
         return T9_EINVAL;
     hipStream_t s = (hipStream_t)stream;
     if (n == 0) return T9_OK;
+    if (!d_in || !d_out || !d_workspace) return T9_EINVAL;
     if (n == 1) {
         HIP_TRY(hipMemcpyAsync(d_out, d_in, rec_size, hipMemcpyDeviceToDevice,
                                s));

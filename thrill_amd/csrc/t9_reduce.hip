@@ -133,11 +133,11 @@ int t9_reduce_build(t9_context* ctx, const u64* d_keys, const u64* d_vals,
                     u64 n, u64* d_tk, u64* d_tv, u64 cap, u64 salt,
                     u32* d_error, void* stream) {
     (void)ctx;
-    if (!d_keys || !d_vals || !d_tk || !d_tv || !d_error || !is_pow2(cap))
-        return T9_EINVAL;
+    if (!d_tk || !d_tv || !d_error || !is_pow2(cap)) return T9_EINVAL;
     hipStream_t s = (hipStream_t)stream;
     HIP_TRY(hipMemsetAsync(d_error, 0, 4, s));
     if (n == 0) return T9_OK;
+    if (!d_keys || !d_vals) return T9_EINVAL;
     hipLaunchKernelGGL(k_reduce_build, dim3(grid_for(n)), dim3(256), 0, s,
                        d_keys, d_vals, n, d_tk, d_tv, cap, salt, d_error);
     T9_LAUNCH_CHECK();

@@ -250,6 +250,56 @@ __global__ __launch_bounds__(256) void k_classify(
                   (unsigned long long)scnt[tid]);
 }
 
+/* Record classification under the acceptance total order: primary compare
+ * on the precomputed big-endian u64 key prefix, byte fallback over the
+ * remaining record bytes only on prefix ties (the reference classifies
+ * with its comparator on whole items — api/sort.hpp:480 compare_function_ —
+ * which under the acceptance order is full-record lexicographic), then the
+ * splitter-index tiebreak (:487-501). */
+__global__ __launch_bounds__(256) void k_classify_rec(
+    const u8* __restrict__ recs, const u64* __restrict__ k64, u64 n,
+    u64 gidx0, const u8* __restrict__ spl_recs,
+    const u64* __restrict__ spl_k64, const u64* __restrict__ spl_idx, u32 p,
+    u32 rec_size, u32* __restrict__ bucket, u64* __restrict__ counts) {
+    __shared__ u64 sk[T9_RADIX], si[T9_RADIX];
+    __shared__ u32 scnt[T9_RADIX];
+    const u32 tid = threadIdx.x;
+    if (tid < p - 1) {
+        sk[tid] = spl_k64[tid];
+        si[tid] = spl_idx[tid];
+    }
+    scnt[tid] = 0;
+    __syncthreads();
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 i = (u64)blockIdx.x * 256 + tid; i < n; i += stride) {
+        const u64 k = k64[i], g = gidx0 + i;
+        u32 b = 0;
+        for (u32 j = 0; j < p - 1; ++j) {
+            bool less;
+            if (sk[j] != k) {
+                less = sk[j] < k;
+            }
+            else {
+                const u8* a = spl_recs + (u64)j * rec_size + 8;
+                const u8* r = recs + i * (u64)rec_size + 8;
+                int c = 0;
+                for (u32 t = 8; t < rec_size && c == 0; ++t) {
+                    u8 x = *a++, y = *r++;
+                    c = (x > y) - (x < y);
+                }
+                less = c < 0 || (c == 0 && si[j] < g);
+            }
+            if (less) b = j + 1;
+        }
+        bucket[i] = b;
+        atomicAdd(&scnt[b], 1u);
+    }
+    __syncthreads();
+    if (tid < p && scnt[tid])
+        atomicAdd((unsigned long long*)&counts[tid],
+                  (unsigned long long)scnt[tid]);
+}
+
 /* count elements whose u64 key equals their left neighbour's (tie probe) */
 __global__ __launch_bounds__(256) void k_count_tied(
     const u64* __restrict__ keys, u64 n, u32* __restrict__ ntied) {
@@ -393,16 +443,37 @@ int t9_classify_u64(t9_context* ctx, const u64* d_keys, u64 n, u64 gidx0,
                     const u64* d_spl_keys, const u64* d_spl_idx, u32 p,
                     u32* d_bucket, u64* d_counts, void* stream) {
     (void)ctx;
-    if (!d_keys || !d_spl_keys || !d_spl_idx || !d_bucket || !d_counts)
-        return T9_EINVAL;
-    if (p < 1 || p > 256) return T9_EINVAL;
+    if (!d_counts || p < 1 || p > 256) return T9_EINVAL;
     hipStream_t s = (hipStream_t)stream;
     HIP_TRY(hipMemsetAsync(d_counts, 0, p * 8, s));
     if (n == 0) return T9_OK;
+    if (!d_keys || !d_spl_keys || !d_spl_idx || !d_bucket) return T9_EINVAL;
     u64 want = t9_ceil_div(n, 256);
     u32 grid = (u32)((want < 2048) ? want : 2048);
     hipLaunchKernelGGL(k_classify, dim3(grid), dim3(256), 0, s, d_keys, n,
                        gidx0, d_spl_keys, d_spl_idx, p, d_bucket, d_counts);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+int t9_classify_rec(t9_context* ctx, const u8* d_recs, const u64* d_k64,
+                    u64 n, u64 gidx0, const u8* d_spl_recs,
+                    const u64* d_spl_k64, const u64* d_spl_idx, u32 p,
+                    u32 rec_size, u32* d_bucket, u64* d_counts,
+                    void* stream) {
+    (void)ctx;
+    if (!d_counts || p < 1 || p > 256 || rec_size < 8) return T9_EINVAL;
+    hipStream_t s = (hipStream_t)stream;
+    HIP_TRY(hipMemsetAsync(d_counts, 0, p * 8, s));
+    if (n == 0) return T9_OK;
+    if (!d_recs || !d_k64 || !d_spl_recs || !d_spl_k64 || !d_spl_idx ||
+        !d_bucket)
+        return T9_EINVAL;
+    u64 want = t9_ceil_div(n, 256);
+    u32 grid = (u32)((want < 2048) ? want : 2048);
+    hipLaunchKernelGGL(k_classify_rec, dim3(grid), dim3(256), 0, s, d_recs,
+                       d_k64, n, gidx0, d_spl_recs, d_spl_k64, d_spl_idx, p,
+                       rec_size, d_bucket, d_counts);
     T9_LAUNCH_CHECK();
     return T9_OK;
 }
@@ -417,13 +488,14 @@ int t9_partition_idx(t9_context* ctx, const u32* d_bucket, u64 n, u32 p,
                      u32* d_perm, u64* d_offsets, void* d_workspace,
                      void* stream) {
     (void)ctx;
-    if (!d_bucket || !d_perm || !d_offsets || !d_workspace) return T9_EINVAL;
-    if (p < 1 || p > 256 || n >= (1ull << 32)) return T9_EINVAL;
+    if (!d_offsets || p < 1 || p > 256 || n >= (1ull << 32))
+        return T9_EINVAL;
     hipStream_t s = (hipStream_t)stream;
     if (n == 0) {
         HIP_TRY(hipMemsetAsync(d_offsets, 0, (p + 1) * 8, s));
         return T9_OK;
     }
+    if (!d_bucket || !d_perm || !d_workspace) return T9_EINVAL;
     char* cp = (char*)d_workspace;
     const u64 B = t9_ceil_div(n, T9_PAIRS_TILE);
     ScanWs w = carve_scan_ws(cp, B);

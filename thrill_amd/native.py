@@ -49,6 +49,8 @@ _SIGS = {
     "t9_sort_records_workspace": (u64, [u64, u32]),
     "t9_sort_records": (i32, [vp, vp, vp, u64, u32, u32, vp, vp]),
     "t9_classify_u64": (i32, [vp, vp, u64, u64, vp, vp, u32, vp, vp, vp]),
+    "t9_classify_rec": (i32, [vp, vp, vp, u64, u64, vp, vp, vp, u32, u32,
+                              vp, vp, vp]),
     "t9_partition_idx_workspace": (u64, [u64]),
     "t9_partition_idx": (i32, [vp, vp, u64, u32, vp, vp, vp, vp]),
     "t9_alltoall": (i32, [vp, vp, vp, vp, vp, vp, vp, u64, vp]),
