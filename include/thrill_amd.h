@@ -179,6 +179,16 @@ int t9_zipf_tokens(t9_context* ctx, uint64_t* d_out, const double* d_cdf,
                    uint64_t N, uint64_t index0, uint64_t n, uint64_t seed,
                    void* stream);
 
+/* ------------------------------------------------------------------ *
+ * Optional per-kernel-class HIP event timing (off by default) for the
+ * bench harness's roofline measurement. Classes: "pair_scatter",
+ * "keys_scatter", "hist_pairs", "hist_keys", "extract", "gather".
+ * ------------------------------------------------------------------ */
+int t9_perf_enable(int on);
+int t9_perf_read(const char* kernel_class, double* total_ms,
+                 uint64_t* launches);
+int t9_perf_reset(void);
+
 #ifdef __cplusplus
 }
 #endif

@@ -70,6 +70,18 @@ __host__ __device__ inline u64 t9_hash128to64(u64 upper, u64 lower) {
 static inline u64 t9_ceil_div(u64 a, u64 b) { return (a + b - 1) / b; }
 static inline u64 t9_align256(u64 x) { return (x + 255) & ~(u64)255; }
 
+/* optional perf-event registry (t9_perf.cpp) */
+bool t9perf_on();
+void* t9perf_begin(hipStream_t s, const char* cls);
+void t9perf_end(void* tok, hipStream_t s);
+
+#define T9_PERF_WRAP(s, cls, launch)                                      \
+    do {                                                                  \
+        void* _tok = t9perf_on() ? t9perf_begin((s), (cls)) : nullptr;    \
+        launch;                                                           \
+        if (_tok) t9perf_end(_tok, (s));                                  \
+    } while (0)
+
 /* radix sort geometry (shared by t9_sort.hip host code and workspace calc) */
 #define T9_RADIX 256
 #define T9_KEYS_TILE 4096   /* elems per block, keys-only scatter */

@@ -158,9 +158,11 @@ int t9_extract_key64(t9_context* ctx, const u8* d_recs, u64 n, u32 rec_size,
     if (rec_size % 4 || key_off % 4 || key_off + 8 > rec_size)
         return T9_EINVAL;
     if (n == 0) return T9_OK;
-    hipLaunchKernelGGL(k_extract_key64, dim3(grid_for(n)), dim3(256), 0,
-                       (hipStream_t)stream, d_recs, n, rec_size / 4, key_off,
-                       d_keys, d_idx);
+    T9_PERF_WRAP((hipStream_t)stream, "extract",
+                 hipLaunchKernelGGL(k_extract_key64, dim3(grid_for(n)),
+                                    dim3(256), 0, (hipStream_t)stream,
+                                    d_recs, n, rec_size / 4, key_off,
+                                    d_keys, d_idx));
     T9_LAUNCH_CHECK();
     return T9_OK;
 }
@@ -170,10 +172,11 @@ int t9_gather_records(t9_context* ctx, const u8* d_recs, const u32* d_idx,
     (void)ctx;
     if (!d_recs || !d_idx || !d_out || rec_size % 4) return T9_EINVAL;
     if (n == 0) return T9_OK;
-    hipLaunchKernelGGL(k_gather_records,
-                       dim3(grid_for(n * (rec_size / 4))), dim3(256), 0,
-                       (hipStream_t)stream, d_recs, d_idx, n, rec_size / 4,
-                       d_out);
+    T9_PERF_WRAP((hipStream_t)stream, "gather",
+                 hipLaunchKernelGGL(k_gather_records,
+                                    dim3(grid_for(n * (rec_size / 4))),
+                                    dim3(256), 0, (hipStream_t)stream,
+                                    d_recs, d_idx, n, rec_size / 4, d_out));
     T9_LAUNCH_CHECK();
     return T9_OK;
 }

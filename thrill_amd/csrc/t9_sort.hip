@@ -382,14 +382,18 @@ int t9_sort_u64(t9_context* ctx, u64* d_keys, u64 n, void* d_workspace,
     u64* bufB = alt;
     for (int pass = 0; pass < 8; ++pass) {
         u32 shift = pass * 8;
-        hipLaunchKernelGGL((k_hist<T9_KEYS_TILE, false>), dim3((u32)B),
-                           dim3(256), 0, s, bufA, nullptr, n, shift, w.hist);
+        T9_PERF_WRAP(s, "hist_keys",
+                     hipLaunchKernelGGL((k_hist<T9_KEYS_TILE, false>),
+                                        dim3((u32)B), dim3(256), 0, s, bufA,
+                                        nullptr, n, shift, w.hist));
         int rc = run_scan(w, s);
         if (rc) return rc;
-        hipLaunchKernelGGL(
-            (k_scatter<T9_KEYS_TILE, true, false, false, false>),
-            dim3((u32)B), dim3(256), 0, s, bufA, nullptr, nullptr, bufB,
-            nullptr, w.hist, n, shift);
+        T9_PERF_WRAP(
+            s, "keys_scatter",
+            hipLaunchKernelGGL(
+                (k_scatter<T9_KEYS_TILE, true, false, false, false>),
+                dim3((u32)B), dim3(256), 0, s, bufA, nullptr, nullptr, bufB,
+                nullptr, w.hist, n, shift));
         T9_LAUNCH_CHECK();
         std::swap(bufA, bufB);
     }
@@ -424,14 +428,18 @@ int t9_sort_pairs_u64_u32(t9_context* ctx, u64* d_keys, u32* d_vals, u64 n,
     u32* vB = alt_v;
     for (int pass = 0; pass < 8; ++pass) {
         u32 shift = pass * 8;
-        hipLaunchKernelGGL((k_hist<T9_PAIRS_TILE, false>), dim3((u32)B),
-                           dim3(256), 0, s, kA, nullptr, n, shift, w.hist);
+        T9_PERF_WRAP(s, "hist_pairs",
+                     hipLaunchKernelGGL((k_hist<T9_PAIRS_TILE, false>),
+                                        dim3((u32)B), dim3(256), 0, s, kA,
+                                        nullptr, n, shift, w.hist));
         int rc = run_scan(w, s);
         if (rc) return rc;
-        hipLaunchKernelGGL(
-            (k_scatter<T9_PAIRS_TILE, true, true, false, false>),
-            dim3((u32)B), dim3(256), 0, s, kA, vA, nullptr, kB, vB, w.hist,
-            n, shift);
+        T9_PERF_WRAP(
+            s, "pair_scatter",
+            hipLaunchKernelGGL(
+                (k_scatter<T9_PAIRS_TILE, true, true, false, false>),
+                dim3((u32)B), dim3(256), 0, s, kA, vA, nullptr, kB, vB,
+                w.hist, n, shift));
         T9_LAUNCH_CHECK();
         std::swap(kA, kB);
         std::swap(vA, vB);

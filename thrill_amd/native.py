@@ -58,6 +58,10 @@ _SIGS = {
     "t9_reduce_build": (i32, [vp, vp, vp, u64, vp, vp, u64, u64, vp, vp]),
     "t9_reduce_drain": (i32, [vp, vp, vp, u64, vp, vp, vp, vp]),
     "t9_zipf_tokens": (i32, [vp, vp, vp, u64, u64, u64, u64, vp]),
+    "t9_perf_enable": (i32, [i32]),
+    "t9_perf_read": (i32, [ctypes.c_char_p, ctypes.POINTER(ctypes.c_double),
+                           ctypes.POINTER(u64)]),
+    "t9_perf_reset": (i32, []),
 }
 
 
@@ -88,6 +92,22 @@ class Native:
                 raise T9Error(f"t9_{name} failed rc={rc}")
             return rc
         return call
+
+    # perf registry functions take no context argument
+    def perf_enable(self, on):
+        self._lib.t9_perf_enable(1 if on else 0)
+
+    def perf_read(self, cls):
+        ms = ctypes.c_double()
+        n = u64()
+        rc = self._lib.t9_perf_read(cls.encode(), ctypes.byref(ms),
+                                    ctypes.byref(n))
+        if rc != 0:
+            raise T9Error(f"t9_perf_read failed rc={rc}")
+        return ms.value, n.value
+
+    def perf_reset(self):
+        self._lib.t9_perf_reset()
 
     def ws(self, name, *args):
         """workspace byte queries (no ctx argument)."""

@@ -1,0 +1,202 @@
+"""Product host orchestration of the TeraSort hot path:
+the role of SortNode::MainOp (thrill/api/sort.hpp:537-663) — sample,
+splitter selection on rank 0, classification, partition, all-to-all
+exchange, local sort — with every bulk step a libt9 HIP kernel and the
+exchange torch.distributed (RCCL) / t9_alltoall. Host python here is
+control plane only (splitters are ~KBs, as in the reference where
+FindAndSendSplitters runs on worker 0's CPU).
+
+No CPU fallback: everything data-sized runs through the C ABI on the GPU.
+"""
+import ctypes
+import math
+
+import numpy as np
+import torch
+
+from .native import Native
+
+REC = 100  # TeraSort record bytes (examples/terasort/terasort.cpp:31-43)
+
+
+def _ptr(t):
+    return ctypes.c_void_p(t.data_ptr())
+
+
+def _stream():
+    return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+def sample_size(n_total):
+    """Reference sample-size law: log2(n) / eps^2, eps = 0.1
+    (thrill/common/reservoir_sampling.hpp:269-274, api/sort.hpp:298)."""
+    if n_total < 2:
+        return 1
+    return max(1, int(math.log2(n_total) * 100.0))
+
+
+def select_splitters(sample_recs, sample_gidx, p):
+    """FindAndSendSplitters (api/sort.hpp:337-378): sort the gathered
+    samples by (record bytes, global index) — LessSampleIndex under the
+    acceptance total order — and pick samples[(size_t)(i * size / p)].
+    sample_recs: (S, 100) uint8; returns (p-1, 100) records + (p-1,) idx."""
+    order = sorted(range(len(sample_gidx)),
+                   key=lambda t: (sample_recs[t].tobytes(),
+                                  int(sample_gidx[t])))
+    step = len(order) / p
+    sel = [order[int(i * step)] for i in range(1, p)]
+    return (np.stack([sample_recs[t] for t in sel]),
+            np.array([sample_gidx[t] for t in sel], dtype=np.uint64))
+
+
+def k64_of_records(recs):
+    """big-endian u64 prefix of each record (host-side, control plane)."""
+    return np.array([int.from_bytes(r[:8].tobytes(), "big") for r in recs],
+                    dtype=np.uint64)
+
+
+class TeraSort:
+    """One rank's TeraSort state. world==1: pure local sort. world>1:
+    sample -> splitters -> classify -> partition -> all-to-all -> local
+    sort; output = this rank's globally-contiguous sorted shard."""
+
+    def __init__(self, n_total, seed, rank=0, world=1, device=0):
+        self.nat = Native(device=device, rank=rank, world=world)
+        self.rank, self.world = rank, world
+        self.n_total, self.seed = n_total, seed
+        base = n_total // world
+        rem = n_total % world
+        self.n_local = base + (1 if rank < rem else 0)
+        self.gidx0 = rank * base + min(rank, rem)
+        self.d_in = torch.empty(self.n_local * REC, dtype=torch.uint8,
+                                device="cuda")
+        self.d_out = torch.empty(self.n_local * REC, dtype=torch.uint8,
+                                 device="cuda")
+        ws_bytes = max(
+            self.nat.ws("sort_records", self.n_local, REC),
+            self.nat.ws("partition_idx", self.n_local),
+        )
+        if world > 1:
+            # headroom for sorting a received shard up to ~1.3x the even
+            # share (splitter imbalance target eps=0.1, api/sort.hpp:298)
+            ws_bytes = max(ws_bytes, self.nat.ws(
+                "sort_records", self.n_local + self.n_local // 3 + 16, REC))
+        self.d_ws = torch.empty(int(ws_bytes), dtype=torch.uint8,
+                                device="cuda")
+        if world > 1:
+            self.d_keys = torch.empty(self.n_local, dtype=torch.int64,
+                                      device="cuda")
+            self.d_idx = torch.empty(self.n_local, dtype=torch.int32,
+                                     device="cuda")
+            self.d_bucket = torch.empty(self.n_local, dtype=torch.int32,
+                                        device="cuda")
+            self.d_counts = torch.empty(world, dtype=torch.int64,
+                                        device="cuda")
+            self.d_perm = torch.empty(self.n_local, dtype=torch.int32,
+                                      device="cuda")
+            self.d_send = torch.empty(self.n_local * REC, dtype=torch.uint8,
+                                      device="cuda")
+
+    def generate(self):
+        self.nat.gen_records(_ptr(self.d_in), self.gidx0, self.n_local,
+                             self.seed, _stream())
+
+    def _splitters(self):
+        """sample locally, gather to rank 0, select, broadcast. Returns
+        device tensors (spl_recs bytes, spl_k64, spl_idx)."""
+        import torch.distributed as dist
+        S = max(1, sample_size(self.n_total) // self.world)
+        stride = max(1, self.n_local // S)
+        pos = torch.arange(0, self.n_local, stride, device="cuda")[:S]
+        # gather sampled records (S x 100 bytes) via byte gather kernel:
+        # use t9_gather_records with the sample positions as indices
+        dpos = pos.to(torch.int32)
+        d_samp = torch.empty(len(pos) * REC, dtype=torch.uint8,
+                             device="cuda")
+        self.nat.gather_records(_ptr(self.d_in), _ptr(dpos), len(pos), REC,
+                                _ptr(d_samp), _stream())
+        gidx = (pos + self.gidx0).to(torch.int64)
+        # equal S per rank not guaranteed (remainders): pad to max S
+        sizes = [None] * self.world
+        dist.all_gather_object(sizes, int(len(pos)))
+        maxS = max(sizes)
+        pad_samp = torch.zeros(maxS * REC, dtype=torch.uint8, device="cuda")
+        pad_samp[:len(pos) * REC] = d_samp
+        pad_idx = torch.zeros(maxS, dtype=torch.int64, device="cuda")
+        pad_idx[:len(pos)] = gidx
+        gs = [torch.empty_like(pad_samp) for _ in range(self.world)]
+        gi = [torch.empty_like(pad_idx) for _ in range(self.world)]
+        dist.all_gather(gs, pad_samp)
+        dist.all_gather(gi, pad_idx)
+        p = self.world
+        spl_recs_t = torch.empty(max(p - 1, 1) * REC, dtype=torch.uint8,
+                                 device="cuda")
+        spl_idx_t = torch.empty(max(p - 1, 1), dtype=torch.int64,
+                                device="cuda")
+        if self.rank == 0:
+            all_recs = np.concatenate([
+                gs[r][:sizes[r] * REC].cpu().numpy().reshape(sizes[r], REC)
+                for r in range(self.world)])
+            all_idx = np.concatenate([
+                gi[r][:sizes[r]].cpu().numpy().astype(np.uint64)
+                for r in range(self.world)])
+            spl_recs, spl_idx = select_splitters(all_recs, all_idx, p)
+            spl_recs_t.copy_(torch.from_numpy(
+                spl_recs.reshape(-1).copy()).cuda())
+            spl_idx_t.copy_(torch.from_numpy(
+                spl_idx.view(np.int64).copy()).cuda())
+        dist.broadcast(spl_recs_t, 0)
+        dist.broadcast(spl_idx_t, 0)
+        spl_recs = spl_recs_t.cpu().numpy().reshape(p - 1, REC)
+        spl_k64 = torch.from_numpy(
+            k64_of_records(spl_recs).view(np.int64)).cuda()
+        return spl_recs_t, spl_k64, spl_idx_t
+
+    def step(self):
+        """One full TeraSort of the (distributed) input. Returns the local
+        output tensor (n_out*100 bytes) and n_out."""
+        nat, s = self.nat, _stream()
+        if self.world == 1:
+            nat.sort_records(_ptr(self.d_in), _ptr(self.d_out),
+                             self.n_local, REC, 10, _ptr(self.d_ws), s)
+            return self.d_out, self.n_local
+
+        import torch.distributed as dist
+        nat.extract_key64(_ptr(self.d_in), self.n_local, REC, 0,
+                          _ptr(self.d_keys), _ptr(self.d_idx), s)
+        spl_recs_t, spl_k64, spl_idx_t = self._splitters()
+        p = self.world
+        nat.classify_rec(_ptr(self.d_in), _ptr(self.d_keys), self.n_local,
+                         self.gidx0, _ptr(spl_recs_t), _ptr(spl_k64),
+                         _ptr(spl_idx_t), p, REC, _ptr(self.d_bucket),
+                         _ptr(self.d_counts), s)
+        d_offs = torch.empty(p + 1, dtype=torch.int64, device="cuda")
+        nat.partition_idx(_ptr(self.d_bucket), self.n_local, p,
+                          _ptr(self.d_perm), _ptr(d_offs), _ptr(self.d_ws),
+                          s)
+        nat.gather_records(_ptr(self.d_in), _ptr(self.d_perm), self.n_local,
+                           REC, _ptr(self.d_send), s)
+        send_counts = self.d_counts.cpu().numpy().astype(np.int64)
+        recv_counts = np.empty(p, dtype=np.int64)
+        sc_t = torch.from_numpy(send_counts).cuda()
+        rc_t = torch.empty(p, dtype=torch.int64, device="cuda")
+        dist.all_to_all_single(rc_t, sc_t)
+        recv_counts = rc_t.cpu().numpy()
+        n_recv = int(recv_counts.sum())
+        d_recv = torch.empty(max(n_recv, 1) * REC, dtype=torch.uint8,
+                             device="cuda")
+        dist.all_to_all_single(
+            d_recv[:n_recv * REC], self.d_send,
+            output_split_sizes=(recv_counts * REC).tolist(),
+            input_split_sizes=(send_counts * REC).tolist())
+        d_sorted = torch.empty(max(n_recv, 1) * REC, dtype=torch.uint8,
+                               device="cuda")
+        ws_need = nat.ws("sort_records", n_recv, REC)
+        ws = self.d_ws if ws_need <= self.d_ws.numel() \
+            else torch.empty(int(ws_need), dtype=torch.uint8, device="cuda")
+        nat.sort_records(_ptr(d_recv), _ptr(d_sorted), n_recv, REC, 10,
+                         _ptr(ws), s)
+        return d_sorted, n_recv
+
+    def close(self):
+        self.nat.close()
