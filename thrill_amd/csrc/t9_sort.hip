@@ -23,6 +23,7 @@
 #include "t9_common.h"
 
 #include <algorithm>
+#include <cstdlib>
 #include <cstring>
 
 /* ------------------------------------------------------------------ *
@@ -103,16 +104,21 @@ __global__ __launch_bounds__(256) void k_finaloffs(
  * yields the same-digit lane mask; rank = running count + earlier-wave
  * counts + popcount of lower same-digit lanes — and reorder into an LDS
  * staging tile, (4) digit-run coalesced global writes at the scanned
- * offsets. */
-template <int TILE, bool HAS_KEY, bool HAS_VAL, bool EXT_DIGIT, bool IOTA_VAL>
+ * offsets.
+ * STAGE_IN=false drops the input staging tiles and re-reads the inputs
+ * from global in phase (3) — the block just read them, so the re-read is
+ * an L1/L2 hit; the freed LDS doubles the tile at equal occupancy, which
+ * doubles the average digit-run length and so the write coalescing. */
+template <int TILE, bool HAS_KEY, bool HAS_VAL, bool EXT_DIGIT,
+          bool IOTA_VAL, bool STAGE_IN>
 __global__ __launch_bounds__(256, 2) void k_scatter(
     const u64* __restrict__ in_keys, const u32* __restrict__ in_vals,
     const u32* __restrict__ ext_digit, u64* __restrict__ out_keys,
     u32* __restrict__ out_vals, const u32* __restrict__ offs, u64 n,
     u32 shift) {
     constexpr int CHUNKS = TILE / 256;
-    __shared__ u64 s_keys[HAS_KEY ? TILE : 1];
-    __shared__ u32 s_vals[HAS_VAL ? TILE : 1];
+    __shared__ u64 s_keys[(HAS_KEY && STAGE_IN) ? TILE : 1];
+    __shared__ u32 s_vals[(HAS_VAL && STAGE_IN) ? TILE : 1];
     __shared__ u64 s_okeys[HAS_KEY ? TILE : 1];
     __shared__ u32 s_ovals[HAS_VAL ? TILE : 1];
     __shared__ u8 s_dig[TILE];
@@ -139,11 +145,11 @@ __global__ __launch_bounds__(256, 2) void k_scatter(
             u64 k = 0;
             if (HAS_KEY) {
                 k = in_keys[base + i];
-                s_keys[i] = k;
+                if (STAGE_IN) s_keys[i] = k;
             }
             u32 d = EXT_DIGIT ? ext_digit[base + i]
                               : ((u32)(k >> shift) & 255u);
-            if (HAS_VAL)
+            if (HAS_VAL && STAGE_IN)
                 s_vals[i] = IOTA_VAL ? (u32)(base + i) : in_vals[base + i];
             s_dig[i] = (u8)d;
             atomicAdd(&s_cnt[d], 1u);
@@ -189,8 +195,12 @@ __global__ __launch_bounds__(256, 2) void k_scatter(
             for (u32 w = 0; w < wave; ++w)
                 before += s_wavecnt[w * T9_RADIX + d];
             u32 pos = s_start[d] + s_run[d] + before + wave_rank;
-            if (HAS_KEY) s_okeys[pos] = s_keys[i];
-            if (HAS_VAL) s_ovals[pos] = s_vals[i];
+            if (HAS_KEY)
+                s_okeys[pos] = STAGE_IN ? s_keys[i] : in_keys[base + i];
+            if (HAS_VAL)
+                s_ovals[pos] = STAGE_IN ? s_vals[i]
+                               : (IOTA_VAL ? (u32)(base + i)
+                                           : in_vals[base + i]);
             s_digof[pos] = (u8)d;
         }
         __syncthreads();
@@ -332,17 +342,25 @@ u64 scan_ws_bytes(u64 B) {
            t9_align256(T9_RADIX * 4);
 }
 
-ScanWs carve_scan_ws(char*& p, u64 B) {
+/* carve buffers sized for B_cap rows but use B_used (<= B_cap) rows */
+ScanWs carve_scan_ws(char*& p, u64 B_cap, u64 B_used) {
     ScanWs w;
-    w.B = B;
-    w.Bc = t9_ceil_div(B, T9_SCAN_CHUNK);
+    w.B = B_used;
+    w.Bc = t9_ceil_div(B_used, T9_SCAN_CHUNK);
     w.hist = (u32*)p;
-    p += t9_align256(B * T9_RADIX * 4);
+    p += t9_align256(B_cap * T9_RADIX * 4);
     w.chunkpart = (u32*)p;
-    p += t9_align256(w.Bc * T9_RADIX * 4);
+    p += t9_align256(t9_ceil_div(B_cap, T9_SCAN_CHUNK) * T9_RADIX * 4);
     w.digit_base = (u32*)p;
     p += t9_align256(T9_RADIX * 4);
     return w;
+}
+
+int env_variant(const char* name, int dflt, int maxv) {
+    const char* e = getenv(name);
+    if (!e) return dflt;
+    int v = atoi(e);
+    return (v >= 1 && v <= maxv) ? v : dflt;
 }
 
 int run_scan(const ScanWs& w, hipStream_t s) {
@@ -375,25 +393,41 @@ int t9_sort_u64(t9_context* ctx, u64* d_keys, u64 n, void* d_workspace,
     char* p = (char*)d_workspace;
     u64* alt = (u64*)p;
     p += t9_align256(n * 8);
-    const u64 B = t9_ceil_div(n, T9_KEYS_TILE);
-    ScanWs w = carve_scan_ws(p, B);
+    const int var = env_variant("T9_KEYS_SCATTER", 2, 2);
+    const u64 tile = var == 1 ? T9_KEYS_TILE : 2 * T9_KEYS_TILE;
+    const u64 B = t9_ceil_div(n, tile);
+    ScanWs w = carve_scan_ws(p, t9_ceil_div(n, (u64)T9_KEYS_TILE), B);
 
     u64* bufA = d_keys;
     u64* bufB = alt;
     for (int pass = 0; pass < 8; ++pass) {
         u32 shift = pass * 8;
-        T9_PERF_WRAP(s, "hist_keys",
-                     hipLaunchKernelGGL((k_hist<T9_KEYS_TILE, false>),
-                                        dim3((u32)B), dim3(256), 0, s, bufA,
-                                        nullptr, n, shift, w.hist));
+        T9_PERF_WRAP(
+            s, "hist_keys",
+            if (var == 1)
+                hipLaunchKernelGGL((k_hist<T9_KEYS_TILE, false>),
+                                   dim3((u32)B), dim3(256), 0, s, bufA,
+                                   nullptr, n, shift, w.hist);
+            else
+                hipLaunchKernelGGL((k_hist<2 * T9_KEYS_TILE, false>),
+                                   dim3((u32)B), dim3(256), 0, s, bufA,
+                                   nullptr, n, shift, w.hist));
         int rc = run_scan(w, s);
         if (rc) return rc;
         T9_PERF_WRAP(
             s, "keys_scatter",
-            hipLaunchKernelGGL(
-                (k_scatter<T9_KEYS_TILE, true, false, false, false>),
-                dim3((u32)B), dim3(256), 0, s, bufA, nullptr, nullptr, bufB,
-                nullptr, w.hist, n, shift));
+            if (var == 1)
+                hipLaunchKernelGGL(
+                    (k_scatter<T9_KEYS_TILE, true, false, false, false,
+                               true>),
+                    dim3((u32)B), dim3(256), 0, s, bufA, nullptr, nullptr,
+                    bufB, nullptr, w.hist, n, shift);
+            else
+                hipLaunchKernelGGL(
+                    (k_scatter<2 * T9_KEYS_TILE, true, false, false, false,
+                               false>),
+                    dim3((u32)B), dim3(256), 0, s, bufA, nullptr, nullptr,
+                    bufB, nullptr, w.hist, n, shift));
         T9_LAUNCH_CHECK();
         std::swap(bufA, bufB);
     }
@@ -419,8 +453,11 @@ int t9_sort_pairs_u64_u32(t9_context* ctx, u64* d_keys, u32* d_vals, u64 n,
     p += t9_align256(n * 8);
     u32* alt_v = (u32*)p;
     p += t9_align256(n * 4);
-    const u64 B = t9_ceil_div(n, T9_PAIRS_TILE);
-    ScanWs w = carve_scan_ws(p, B);
+    const int var = env_variant("T9_PAIR_SCATTER", 2, 3);
+    const u64 tile = var == 1 ? T9_PAIRS_TILE
+                   : (var == 2 ? 2 * T9_PAIRS_TILE : 4 * T9_PAIRS_TILE);
+    const u64 B = t9_ceil_div(n, tile);
+    ScanWs w = carve_scan_ws(p, t9_ceil_div(n, (u64)T9_PAIRS_TILE), B);
 
     u64* kA = d_keys;
     u64* kB = alt_k;
@@ -428,18 +465,42 @@ int t9_sort_pairs_u64_u32(t9_context* ctx, u64* d_keys, u32* d_vals, u64 n,
     u32* vB = alt_v;
     for (int pass = 0; pass < 8; ++pass) {
         u32 shift = pass * 8;
-        T9_PERF_WRAP(s, "hist_pairs",
-                     hipLaunchKernelGGL((k_hist<T9_PAIRS_TILE, false>),
-                                        dim3((u32)B), dim3(256), 0, s, kA,
-                                        nullptr, n, shift, w.hist));
+        T9_PERF_WRAP(
+            s, "hist_pairs",
+            if (var == 1)
+                hipLaunchKernelGGL((k_hist<T9_PAIRS_TILE, false>),
+                                   dim3((u32)B), dim3(256), 0, s, kA,
+                                   nullptr, n, shift, w.hist);
+            else if (var == 2)
+                hipLaunchKernelGGL((k_hist<2 * T9_PAIRS_TILE, false>),
+                                   dim3((u32)B), dim3(256), 0, s, kA,
+                                   nullptr, n, shift, w.hist);
+            else
+                hipLaunchKernelGGL((k_hist<4 * T9_PAIRS_TILE, false>),
+                                   dim3((u32)B), dim3(256), 0, s, kA,
+                                   nullptr, n, shift, w.hist));
         int rc = run_scan(w, s);
         if (rc) return rc;
         T9_PERF_WRAP(
             s, "pair_scatter",
-            hipLaunchKernelGGL(
-                (k_scatter<T9_PAIRS_TILE, true, true, false, false>),
-                dim3((u32)B), dim3(256), 0, s, kA, vA, nullptr, kB, vB,
-                w.hist, n, shift));
+            if (var == 1)
+                hipLaunchKernelGGL(
+                    (k_scatter<T9_PAIRS_TILE, true, true, false, false,
+                               true>),
+                    dim3((u32)B), dim3(256), 0, s, kA, vA, nullptr, kB, vB,
+                    w.hist, n, shift);
+            else if (var == 2)
+                hipLaunchKernelGGL(
+                    (k_scatter<2 * T9_PAIRS_TILE, true, true, false, false,
+                               false>),
+                    dim3((u32)B), dim3(256), 0, s, kA, vA, nullptr, kB, vB,
+                    w.hist, n, shift);
+            else
+                hipLaunchKernelGGL(
+                    (k_scatter<4 * T9_PAIRS_TILE, true, true, false, false,
+                               false>),
+                    dim3((u32)B), dim3(256), 0, s, kA, vA, nullptr, kB, vB,
+                    w.hist, n, shift));
         T9_LAUNCH_CHECK();
         std::swap(kA, kB);
         std::swap(vA, vB);
@@ -506,14 +567,15 @@ int t9_partition_idx(t9_context* ctx, const u32* d_bucket, u64 n, u32 p,
     if (!d_bucket || !d_perm || !d_workspace) return T9_EINVAL;
     char* cp = (char*)d_workspace;
     const u64 B = t9_ceil_div(n, T9_PAIRS_TILE);
-    ScanWs w = carve_scan_ws(cp, B);
+    ScanWs w = carve_scan_ws(cp, B, B);
     hipLaunchKernelGGL((k_hist<T9_PAIRS_TILE, true>), dim3((u32)B),
                        dim3(256), 0, s, nullptr, d_bucket, n, 0, w.hist);
     int rc = run_scan(w, s);
     if (rc) return rc;
-    hipLaunchKernelGGL((k_scatter<T9_PAIRS_TILE, false, true, true, true>),
-                       dim3((u32)B), dim3(256), 0, s, nullptr, nullptr,
-                       d_bucket, nullptr, d_perm, w.hist, n, 0);
+    hipLaunchKernelGGL(
+        (k_scatter<T9_PAIRS_TILE, false, true, true, true, true>),
+        dim3((u32)B), dim3(256), 0, s, nullptr, nullptr, d_bucket, nullptr,
+        d_perm, w.hist, n, 0);
     hipLaunchKernelGGL(k_bucket_offsets, dim3(1), dim3(512), 0, s,
                        w.digit_base, p, n, d_offsets);
     T9_LAUNCH_CHECK();
