@@ -222,6 +222,122 @@ __global__ __launch_bounds__(256, 2) void k_scatter(
     }
 }
 
+/* Wave-autonomous stable scatter: each of the 4 waves owns a contiguous
+ * quarter of the tile and ranks it with NO block barriers — ballot over
+ * the digit bits gives the same-digit lane mask, a wave-private LDS
+ * counter row carries the running per-digit count (wave-ordered LDS, no
+ * races). One barrier, a cross-wave combine (exclusive digit scan +
+ * per-wave bases), one barrier, then the LDS reorder + digit-run
+ * coalesced global writes. Replaces the chunk-serialized k_scatter rank
+ * loop (4 barriers x TILE/256 chunks) which measured latency-bound. */
+template <int TILE, bool HAS_KEY, bool HAS_VAL, bool STAGE_IN>
+__global__ __launch_bounds__(256, 2) void k_scatter_wave(
+    const u64* __restrict__ in_keys, const u32* __restrict__ in_vals,
+    u64* __restrict__ out_keys, u32* __restrict__ out_vals,
+    const u32* __restrict__ offs, u64 n, u32 shift) {
+    constexpr int NW = 4;
+    constexpr int SUB = TILE / NW;
+    constexpr int GROUPS = SUB / 64;
+    __shared__ u64 s_keys[(HAS_KEY && STAGE_IN) ? TILE : 1];
+    __shared__ u32 s_vals[(HAS_VAL && STAGE_IN) ? TILE : 1];
+    __shared__ u64 s_okeys[HAS_KEY ? TILE : 1];
+    __shared__ u32 s_ovals[HAS_VAL ? TILE : 1];
+    __shared__ u8 s_dig[TILE];
+    __shared__ u16 s_rank[TILE];
+    __shared__ u8 s_digof[TILE];
+    __shared__ u32 s_wcnt[NW * T9_RADIX];
+    __shared__ u32 s_woff[NW * T9_RADIX];
+    __shared__ u32 s_start[T9_RADIX];
+    __shared__ u32 s_goff[T9_RADIX];
+
+    const u32 tid = threadIdx.x, wave = tid >> 6, lane = tid & 63;
+    const u64 base = (u64)blockIdx.x * TILE;
+    const u32 tn = (u32)((n - base < (u64)TILE) ? (n - base) : (u64)TILE);
+
+    s_goff[tid] = offs[(u64)blockIdx.x * T9_RADIX + tid];
+    for (u32 t = lane; t < T9_RADIX; t += 64) s_wcnt[wave * T9_RADIX + t] = 0;
+    /* no barrier needed: each wave touches only its own s_wcnt row until
+     * the combine barrier below */
+
+    const u32 wbase = wave * SUB;
+    for (int g = 0; g < GROUPS; ++g) {
+        const u32 i = wbase + g * 64 + lane;
+        const bool valid = i < tn;
+        u64 k = 0;
+        u32 d = 0;
+        if (valid) {
+            k = in_keys[base + i];
+            d = (u32)(k >> shift) & 255u;
+            if (STAGE_IN && HAS_KEY) s_keys[i] = k;
+            if (STAGE_IN && HAS_VAL) s_vals[i] = in_vals[base + i];
+            s_dig[i] = (u8)d;
+        }
+        u64 m = __ballot(valid);
+        for (int bit = 0; bit < 8; ++bit) {
+            u64 bb = __ballot((d >> bit) & 1u);
+            m &= ((d >> bit) & 1u) ? bb : ~bb;
+        }
+        const u32 wr = (u32)__popcll(m & ((1ull << lane) - 1ull));
+        const u32 before = valid ? s_wcnt[wave * T9_RADIX + d] : 0;
+        if (valid) {
+            s_rank[i] = (u16)(before + wr);
+            if (wr == 0)
+                s_wcnt[wave * T9_RADIX + d] =
+                    before + (u32)__popcll(m);
+        }
+    }
+    __syncthreads();
+
+    /* combine: thread tid owns digit tid */
+    const u32 c0 = s_wcnt[0 * T9_RADIX + tid];
+    const u32 c1 = s_wcnt[1 * T9_RADIX + tid];
+    const u32 c2 = s_wcnt[2 * T9_RADIX + tid];
+    const u32 c3 = s_wcnt[3 * T9_RADIX + tid];
+    const u32 total = c0 + c1 + c2 + c3;
+    /* exclusive scan of totals across digits (Hillis-Steele in s_start) */
+    s_start[tid] = total;
+    __syncthreads();
+    for (int off = 1; off < T9_RADIX; off <<= 1) {
+        u32 y = (tid >= (u32)off) ? s_start[tid - off] : 0;
+        __syncthreads();
+        s_start[tid] += y;
+        __syncthreads();
+    }
+    const u32 excl = s_start[tid] - total;
+    __syncthreads();
+    s_start[tid] = excl;
+    s_woff[0 * T9_RADIX + tid] = excl;
+    s_woff[1 * T9_RADIX + tid] = excl + c0;
+    s_woff[2 * T9_RADIX + tid] = excl + c0 + c1;
+    s_woff[3 * T9_RADIX + tid] = excl + c0 + c1 + c2;
+    __syncthreads();
+
+    for (int g = 0; g < GROUPS; ++g) {
+        const u32 i = wbase + g * 64 + lane;
+        if (i < tn) {
+            const u32 d = s_dig[i];
+            const u32 pos = s_woff[wave * T9_RADIX + d] + s_rank[i];
+            if (HAS_KEY)
+                s_okeys[pos] = STAGE_IN ? s_keys[i] : in_keys[base + i];
+            if (HAS_VAL)
+                s_ovals[pos] = STAGE_IN ? s_vals[i] : in_vals[base + i];
+            s_digof[pos] = (u8)d;
+        }
+    }
+    __syncthreads();
+
+    constexpr int CHUNKS = TILE / 256;
+    for (int c = 0; c < CHUNKS; ++c) {
+        const u32 j = c * 256 + tid;
+        if (j < tn) {
+            const u32 d = s_digof[j];
+            const u64 gpos = (u64)s_goff[d] + (j - s_start[d]);
+            if (HAS_KEY) out_keys[gpos] = s_okeys[j];
+            if (HAS_VAL) out_vals[gpos] = s_ovals[j];
+        }
+    }
+}
+
 /* bucket offsets (u64, p+1 entries) from the digit base array */
 __global__ __launch_bounds__(512) void k_bucket_offsets(
     const u32* __restrict__ digit_base, u32 p, u64 n,
@@ -393,8 +509,8 @@ int t9_sort_u64(t9_context* ctx, u64* d_keys, u64 n, void* d_workspace,
     char* p = (char*)d_workspace;
     u64* alt = (u64*)p;
     p += t9_align256(n * 8);
-    const int var = env_variant("T9_KEYS_SCATTER", 2, 2);
-    const u64 tile = var == 1 ? T9_KEYS_TILE : 2 * T9_KEYS_TILE;
+    const int var = env_variant("T9_KEYS_SCATTER", 3, 3);
+    const u64 tile = var == 2 ? 2 * T9_KEYS_TILE : T9_KEYS_TILE;
     const u64 B = t9_ceil_div(n, tile);
     ScanWs w = carve_scan_ws(p, t9_ceil_div(n, (u64)T9_KEYS_TILE), B);
 
@@ -404,12 +520,12 @@ int t9_sort_u64(t9_context* ctx, u64* d_keys, u64 n, void* d_workspace,
         u32 shift = pass * 8;
         T9_PERF_WRAP(
             s, "hist_keys",
-            if (var == 1)
-                hipLaunchKernelGGL((k_hist<T9_KEYS_TILE, false>),
+            if (var == 2)
+                hipLaunchKernelGGL((k_hist<2 * T9_KEYS_TILE, false>),
                                    dim3((u32)B), dim3(256), 0, s, bufA,
                                    nullptr, n, shift, w.hist);
             else
-                hipLaunchKernelGGL((k_hist<2 * T9_KEYS_TILE, false>),
+                hipLaunchKernelGGL((k_hist<T9_KEYS_TILE, false>),
                                    dim3((u32)B), dim3(256), 0, s, bufA,
                                    nullptr, n, shift, w.hist));
         int rc = run_scan(w, s);
@@ -422,12 +538,17 @@ int t9_sort_u64(t9_context* ctx, u64* d_keys, u64 n, void* d_workspace,
                                true>),
                     dim3((u32)B), dim3(256), 0, s, bufA, nullptr, nullptr,
                     bufB, nullptr, w.hist, n, shift);
-            else
+            else if (var == 2)
                 hipLaunchKernelGGL(
                     (k_scatter<2 * T9_KEYS_TILE, true, false, false, false,
                                false>),
                     dim3((u32)B), dim3(256), 0, s, bufA, nullptr, nullptr,
-                    bufB, nullptr, w.hist, n, shift));
+                    bufB, nullptr, w.hist, n, shift);
+            else
+                hipLaunchKernelGGL(
+                    (k_scatter_wave<T9_KEYS_TILE, true, false, false>),
+                    dim3((u32)B), dim3(256), 0, s, bufA, nullptr, bufB,
+                    nullptr, w.hist, n, shift));
         T9_LAUNCH_CHECK();
         std::swap(bufA, bufB);
     }
@@ -453,9 +574,9 @@ int t9_sort_pairs_u64_u32(t9_context* ctx, u64* d_keys, u32* d_vals, u64 n,
     p += t9_align256(n * 8);
     u32* alt_v = (u32*)p;
     p += t9_align256(n * 4);
-    const int var = env_variant("T9_PAIR_SCATTER", 2, 3);
-    const u64 tile = var == 1 ? T9_PAIRS_TILE
-                   : (var == 2 ? 2 * T9_PAIRS_TILE : 4 * T9_PAIRS_TILE);
+    const int var = env_variant("T9_PAIR_SCATTER", 5, 5);
+    const u64 tile = (var == 1 || var == 4) ? T9_PAIRS_TILE
+                   : (var == 3 ? 4 * T9_PAIRS_TILE : 2 * T9_PAIRS_TILE);
     const u64 B = t9_ceil_div(n, tile);
     ScanWs w = carve_scan_ws(p, t9_ceil_div(n, (u64)T9_PAIRS_TILE), B);
 
@@ -467,16 +588,16 @@ int t9_sort_pairs_u64_u32(t9_context* ctx, u64* d_keys, u32* d_vals, u64 n,
         u32 shift = pass * 8;
         T9_PERF_WRAP(
             s, "hist_pairs",
-            if (var == 1)
+            if (var == 1 || var == 4)
                 hipLaunchKernelGGL((k_hist<T9_PAIRS_TILE, false>),
                                    dim3((u32)B), dim3(256), 0, s, kA,
                                    nullptr, n, shift, w.hist);
-            else if (var == 2)
-                hipLaunchKernelGGL((k_hist<2 * T9_PAIRS_TILE, false>),
+            else if (var == 3)
+                hipLaunchKernelGGL((k_hist<4 * T9_PAIRS_TILE, false>),
                                    dim3((u32)B), dim3(256), 0, s, kA,
                                    nullptr, n, shift, w.hist);
             else
-                hipLaunchKernelGGL((k_hist<4 * T9_PAIRS_TILE, false>),
+                hipLaunchKernelGGL((k_hist<2 * T9_PAIRS_TILE, false>),
                                    dim3((u32)B), dim3(256), 0, s, kA,
                                    nullptr, n, shift, w.hist));
         int rc = run_scan(w, s);
@@ -495,12 +616,22 @@ int t9_sort_pairs_u64_u32(t9_context* ctx, u64* d_keys, u32* d_vals, u64 n,
                                false>),
                     dim3((u32)B), dim3(256), 0, s, kA, vA, nullptr, kB, vB,
                     w.hist, n, shift);
-            else
+            else if (var == 3)
                 hipLaunchKernelGGL(
                     (k_scatter<4 * T9_PAIRS_TILE, true, true, false, false,
                                false>),
                     dim3((u32)B), dim3(256), 0, s, kA, vA, nullptr, kB, vB,
-                    w.hist, n, shift));
+                    w.hist, n, shift);
+            else if (var == 4)
+                hipLaunchKernelGGL(
+                    (k_scatter_wave<T9_PAIRS_TILE, true, true, true>),
+                    dim3((u32)B), dim3(256), 0, s, kA, vA, kB, vB, w.hist,
+                    n, shift);
+            else
+                hipLaunchKernelGGL(
+                    (k_scatter_wave<2 * T9_PAIRS_TILE, true, true, false>),
+                    dim3((u32)B), dim3(256), 0, s, kA, vA, kB, vB, w.hist,
+                    n, shift));
         T9_LAUNCH_CHECK();
         std::swap(kA, kB);
         std::swap(vA, vB);
