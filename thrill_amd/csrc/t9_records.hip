@@ -109,13 +109,10 @@ __global__ __launch_bounds__(256) void k_gather_records(
         }
         else {
             for (int t = 0; t < 4 && g0 + t < total_words; ++t) {
-                u64 g = g0 + t;
-                if (w + t >= rec_words) {
-                    rec = g / rec_words;
-                    w = (u32)(g - rec * rec_words) - t;
-                    src = (u64)idx[rec] * rec_words + w;
-                }
-                ((u32*)out)[g] = rin[src + t];
+                const u64 g = g0 + t;
+                const u64 r = g / rec_words;
+                const u32 ww = (u32)(g - r * rec_words);
+                ((u32*)out)[g] = rin[(u64)idx[r] * rec_words + ww];
             }
         }
     }
