@@ -81,40 +81,22 @@ __global__ __launch_bounds__(256) void k_extract_key64(
     }
 }
 
-/* out[i] = recs[idx[i]]: each lane handles 4 consecutive output words and
- * writes them as one 16-byte store (the output is 16 B aligned per lane);
- * reads touch whole source records (contiguous lines), consecutive lanes
- * cover consecutive output positions. */
+/* out[i] = recs[idx[i]], word-wise: consecutive threads write consecutive
+ * output words; a wave's 64 word-reads touch only ~3 source records
+ * (contiguous lines), which measured faster than 16-byte-per-lane
+ * variants that spread each read instruction over ~10 random records. */
 __global__ __launch_bounds__(256) void k_gather_records(
     const u8* __restrict__ recs, const u32* __restrict__ idx, u64 n,
     u32 rec_words, u8* __restrict__ out) {
     const u64 total_words = n * rec_words;
-    const u64 quads = (total_words + 3) / 4;
     const u64 stride = (u64)gridDim.x * 256;
     const u32* rin = (const u32*)recs;
-    for (u64 q = (u64)blockIdx.x * 256 + threadIdx.x; q < quads;
-         q += stride) {
-        const u64 g0 = q * 4;
-        u64 rec = g0 / rec_words;
-        u32 w = (u32)(g0 - rec * rec_words);
-        u64 src = (u64)idx[rec] * rec_words + w;
-        u32 v[4];
-        if (g0 + 4 <= total_words && w + 4 <= rec_words) {
-            /* all four words in one source record */
-            v[0] = rin[src];
-            v[1] = rin[src + 1];
-            v[2] = rin[src + 2];
-            v[3] = rin[src + 3];
-            *(uint4*)(out + g0 * 4) = make_uint4(v[0], v[1], v[2], v[3]);
-        }
-        else {
-            for (int t = 0; t < 4 && g0 + t < total_words; ++t) {
-                const u64 g = g0 + t;
-                const u64 r = g / rec_words;
-                const u32 ww = (u32)(g - r * rec_words);
-                ((u32*)out)[g] = rin[(u64)idx[r] * rec_words + ww];
-            }
-        }
+    u32* rout = (u32*)out;
+    for (u64 g = (u64)blockIdx.x * 256 + threadIdx.x; g < total_words;
+         g += stride) {
+        u64 rec = g / rec_words;
+        u32 w = (u32)(g - rec * rec_words);
+        rout[g] = rin[(u64)idx[rec] * rec_words + w];
     }
 }
 

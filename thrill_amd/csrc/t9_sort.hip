@@ -615,7 +615,7 @@ int t9_sort_u64(t9_context* ctx, u64* d_keys, u64 n, void* d_workspace,
     char* p = (char*)d_workspace;
     u64* alt = (u64*)p;
     p += t9_align256(n * 8);
-    const int var = env_variant("T9_KEYS_SCATTER", 4, 5);
+    const int var = env_variant("T9_KEYS_SCATTER", 5, 5);
     const u64 tile = (var == 2 || var == 5) ? 2 * T9_KEYS_TILE
                                             : T9_KEYS_TILE;
     const u64 B = t9_ceil_div(n, tile);
@@ -692,7 +692,7 @@ int t9_sort_pairs_u64_u32(t9_context* ctx, u64* d_keys, u32* d_vals, u64 n,
     p += t9_align256(n * 8);
     u32* alt_v = (u32*)p;
     p += t9_align256(n * 4);
-    const int var = env_variant("T9_PAIR_SCATTER", 6, 7);
+    const int var = env_variant("T9_PAIR_SCATTER", 7, 7);
     const u64 tile = (var == 1 || var == 4) ? T9_PAIRS_TILE
                    : ((var == 3 || var == 7) ? 4 * T9_PAIRS_TILE
                                              : 2 * T9_PAIRS_TILE);
