@@ -154,6 +154,14 @@ int t9_alltoall(t9_context* ctx, const void* d_send,
  * >= 2x the number of distinct keys (no grow/spill: 288 GB HBM holds the
  * table — reference grow/spill machinery is subsumed by sizing).
  * ------------------------------------------------------------------ */
+/* bucket = Hash128to64(salt, key) % p, the reference's ReduceByHash
+ * partition mapping (core/reduce_functional.hpp:60-72; libstdc++
+ * std::hash<u64> is the identity) — splits pre-reduced pairs across ranks.
+ * d_counts[p] (u64, zeroed by the call) accumulates per-rank totals. */
+int t9_hash_bucket(t9_context* ctx, const uint64_t* d_keys, uint64_t n,
+                   uint64_t salt, uint32_t p, uint32_t* d_bucket,
+                   uint64_t* d_counts, void* stream);
+
 int t9_reduce_init(t9_context* ctx, uint64_t* d_table_keys,
                    uint64_t* d_table_vals, uint64_t capacity, void* stream);
 /* Accumulate n (key, value) pairs into the table; value reduce = u64 add.

@@ -124,3 +124,35 @@ def test_bacon_ipsum_kat_gpu(nat, oracle):
     back = {h: w for w, h in wid.items()}
     result = {back[int(k)]: int(v) for k, v in zip(gk, gv)}
     assert result == table
+
+
+def test_hash_bucket_matches_reference_mapping(nat, oracle):
+    n, p = 200_000, 8
+    keys = oracle.gen_u64(n, seed=31)
+    dk = G.dev(keys)
+    db = G.empty(n, np.uint32)
+    dc = G.empty(p, np.uint64)
+    nat.hash_bucket(G.ptr(dk), n, 0, p, G.ptr(db), G.ptr(dc), G.stream())
+    got = G.host(db, np.uint32)
+    expect = np.array([oracle.partition_of_u64(int(k), 0, p)
+                       for k in keys[:2000]], dtype=np.uint32)
+    assert np.array_equal(got[:2000], expect)
+    counts = G.host(dc, np.uint64)
+    assert counts.sum() == n
+
+
+def test_wordcount_pipeline_single_gpu(nat, oracle):
+    from thrill_amd.pipeline import WordCount, zipf_cdf
+    n, vocab = 1 << 20, 50_000
+    wc = WordCount(n, vocab, 1.1, seed=5, rank=0, world=1, device=0)
+    wc.generate()
+    ok, ov, m = wc.step()
+    gk = G.host(ok, np.uint64)
+    gv = G.host(ov, np.uint64)
+    order = np.argsort(gk)
+    # oracle on the SAME cdf table -> identical token stream
+    toks = oracle.zipf_tokens(zipf_cdf(vocab, 1.1), n, seed=5)
+    ek, ev = oracle.reduce_u64(toks, np.ones(n, np.uint64))
+    assert np.array_equal(gk[order], ek)
+    assert np.array_equal(gv[order], ev)
+    wc.close()

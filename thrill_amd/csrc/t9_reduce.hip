@@ -109,6 +109,28 @@ __global__ __launch_bounds__(256) void k_reduce_drain(
     }
 }
 
+/* bucket = Hash128to64(salt, key) % p — the ReduceByHash partition mapping
+ * (core/reduce_functional.hpp:60-72) used to split pre-reduced pairs
+ * across ranks before the all-to-all. d_counts[p] accumulates totals. */
+__global__ __launch_bounds__(256) void k_hash_bucket(
+    const u64* __restrict__ keys, u64 n, u64 salt, u32 p,
+    u32* __restrict__ bucket, u64* __restrict__ counts) {
+    __shared__ u32 scnt[256];
+    const u32 tid = threadIdx.x;
+    scnt[tid] = 0;
+    __syncthreads();
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 i = (u64)blockIdx.x * 256 + tid; i < n; i += stride) {
+        u32 b = (u32)(t9_hash128to64(salt, keys[i]) % p);
+        bucket[i] = b;
+        atomicAdd(&scnt[b], 1u);
+    }
+    __syncthreads();
+    if (tid < p && scnt[tid])
+        atomicAdd((unsigned long long*)&counts[tid],
+                  (unsigned long long)scnt[tid]);
+}
+
 namespace {
 u32 grid_for(u64 work) {
     u64 want = (work + 255) / 256;
@@ -118,6 +140,20 @@ bool is_pow2(u64 x) { return x && !(x & (x - 1)); }
 } // namespace
 
 extern "C" {
+
+int t9_hash_bucket(t9_context* ctx, const u64* d_keys, u64 n, u64 salt,
+                   u32 p, u32* d_bucket, u64* d_counts, void* stream) {
+    (void)ctx;
+    if (!d_counts || p < 1 || p > 256) return T9_EINVAL;
+    hipStream_t s = (hipStream_t)stream;
+    HIP_TRY(hipMemsetAsync(d_counts, 0, p * 8, s));
+    if (n == 0) return T9_OK;
+    if (!d_keys || !d_bucket) return T9_EINVAL;
+    hipLaunchKernelGGL(k_hash_bucket, dim3(grid_for(n)), dim3(256), 0, s,
+                       d_keys, n, salt, p, d_bucket, d_counts);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
 
 int t9_reduce_init(t9_context* ctx, u64* d_tk, u64* d_tv, u64 cap,
                    void* stream) {

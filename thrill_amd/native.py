@@ -54,6 +54,7 @@ _SIGS = {
     "t9_partition_idx_workspace": (u64, [u64]),
     "t9_partition_idx": (i32, [vp, vp, u64, u32, vp, vp, vp, vp]),
     "t9_alltoall": (i32, [vp, vp, vp, vp, vp, vp, vp, u64, vp]),
+    "t9_hash_bucket": (i32, [vp, vp, u64, u64, u32, vp, vp, vp]),
     "t9_reduce_init": (i32, [vp, vp, vp, u64, vp]),
     "t9_reduce_build": (i32, [vp, vp, vp, u64, vp, vp, u64, u64, vp, vp]),
     "t9_reduce_drain": (i32, [vp, vp, vp, u64, vp, vp, vp, vp]),

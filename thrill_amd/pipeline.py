@@ -55,6 +55,113 @@ def k64_of_records(recs):
                     dtype=np.uint64)
 
 
+def zipf_cdf(N, s, q=0.0):
+    """CDF of the Zipf-Mandelbrot mass function 1/(H*(k+q)^s)
+    (thrill/common/zipf_distribution.hpp:55-120). The canonical table for
+    the product path; tests feed the SAME table to the oracle sampler, so
+    parity is over the sampling given the table."""
+    w = 1.0 / np.power(np.arange(1, N + 1, dtype=np.float64) + q, s)
+    cdf = np.cumsum(w)
+    cdf /= cdf[-1]
+    return cdf
+
+
+class WordCount:
+    """One rank's ReduceByKey (word_count) state: Zipf tokens -> local
+    pre-reduce (ReducePrePhase role, core/reduce_pre_phase.hpp) -> hash
+    partition (Hash128to64 % p, core/reduce_functional.hpp:60-72) ->
+    one all-to-all -> final reduce (post phase)."""
+
+    def __init__(self, n_total, vocab, s, seed, rank=0, world=1, device=0):
+        self.nat = Native(device=device, rank=rank, world=world)
+        self.rank, self.world = rank, world
+        self.n_total, self.seed = n_total, seed
+        base = n_total // world
+        rem = n_total % world
+        self.n_local = base + (1 if rank < rem else 0)
+        self.tok0 = rank * base + min(rank, rem)
+        self.vocab = vocab
+        cdf = zipf_cdf(vocab, s)
+        self.d_cdf = torch.from_numpy(cdf).cuda()
+        self.d_toks = torch.empty(self.n_local, dtype=torch.int64,
+                                  device="cuda")
+        self.d_ones = torch.ones(self.n_local, dtype=torch.int64,
+                                 device="cuda")
+        cap = 1 << max(10, int(math.ceil(math.log2(2 * vocab + 2))))
+        self.cap = cap
+        self.d_tk = torch.empty(cap + 1, dtype=torch.int64, device="cuda")
+        self.d_tv = torch.empty(cap + 1, dtype=torch.int64, device="cuda")
+        self.d_ok = torch.empty(cap + 1, dtype=torch.int64, device="cuda")
+        self.d_ov = torch.empty(cap + 1, dtype=torch.int64, device="cuda")
+        self.d_err = torch.empty(1, dtype=torch.int32, device="cuda")
+        self.d_n = torch.empty(1, dtype=torch.int64, device="cuda")
+
+    def generate(self):
+        self.nat.zipf_tokens(_ptr(self.d_toks), _ptr(self.d_cdf),
+                             self.vocab, self.tok0, self.n_local,
+                             self.seed, _stream())
+
+    def _reduce(self, d_keys, d_vals, n, salt=0):
+        nat, s = self.nat, _stream()
+        nat.reduce_init(_ptr(self.d_tk), _ptr(self.d_tv), self.cap, s)
+        nat.reduce_build(_ptr(d_keys), _ptr(d_vals), n, _ptr(self.d_tk),
+                         _ptr(self.d_tv), self.cap, salt, _ptr(self.d_err),
+                         s)
+        nat.reduce_drain(_ptr(self.d_tk), _ptr(self.d_tv), self.cap,
+                         _ptr(self.d_ok), _ptr(self.d_ov), _ptr(self.d_n),
+                         s)
+        m = int(self.d_n.cpu().item())
+        assert int(self.d_err.cpu().item()) == 0, "reduce table overflow"
+        return self.d_ok[:m], self.d_ov[:m], m
+
+    def step(self):
+        """One full ReduceByKey of the (distributed) token stream. Returns
+        (keys tensor, vals tensor, m) of this rank's final pairs."""
+        nat, s = self.nat, _stream()
+        ok, ov, m = self._reduce(self.d_toks, self.d_ones, self.n_local)
+        if self.world == 1:
+            return ok.clone(), ov.clone(), m
+
+        import torch.distributed as dist
+        p = self.world
+        keys = ok.clone()
+        vals = ov.clone()
+        d_bucket = torch.empty(max(m, 1), dtype=torch.int32, device="cuda")
+        d_counts = torch.empty(p, dtype=torch.int64, device="cuda")
+        nat.hash_bucket(_ptr(keys), m, 0, p, _ptr(d_bucket),
+                        _ptr(d_counts), s)
+        d_perm = torch.empty(max(m, 1), dtype=torch.int32, device="cuda")
+        d_offs = torch.empty(p + 1, dtype=torch.int64, device="cuda")
+        d_ws = torch.empty(max(int(self.nat.ws("partition_idx", m)), 256),
+                           dtype=torch.uint8, device="cuda")
+        nat.partition_idx(_ptr(d_bucket), m, p, _ptr(d_perm), _ptr(d_offs),
+                          _ptr(d_ws), s)
+        ks = torch.empty_like(keys)
+        vs = torch.empty_like(vals)
+        if m:
+            nat.gather_records(_ptr(keys), _ptr(d_perm), m, 8, _ptr(ks), s)
+            nat.gather_records(_ptr(vals), _ptr(d_perm), m, 8, _ptr(vs), s)
+        send_counts = d_counts.cpu().numpy().astype(np.int64)
+        sc_t = torch.from_numpy(send_counts).cuda()
+        rc_t = torch.empty(p, dtype=torch.int64, device="cuda")
+        dist.all_to_all_single(rc_t, sc_t)
+        recv_counts = rc_t.cpu().numpy()
+        n_recv = int(recv_counts.sum())
+        rk = torch.empty(max(n_recv, 1), dtype=torch.int64, device="cuda")
+        rv = torch.empty(max(n_recv, 1), dtype=torch.int64, device="cuda")
+        dist.all_to_all_single(rk[:n_recv], ks[:m],
+                               output_split_sizes=recv_counts.tolist(),
+                               input_split_sizes=send_counts.tolist())
+        dist.all_to_all_single(rv[:n_recv], vs[:m],
+                               output_split_sizes=recv_counts.tolist(),
+                               input_split_sizes=send_counts.tolist())
+        ok2, ov2, m2 = self._reduce(rk, rv, n_recv)
+        return ok2.clone(), ov2.clone(), m2
+
+    def close(self):
+        self.nat.close()
+
+
 class TeraSort:
     """One rank's TeraSort state. world==1: pure local sort. world>1:
     sample -> splitters -> classify -> partition -> all-to-all -> local
