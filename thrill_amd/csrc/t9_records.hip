@@ -92,11 +92,22 @@ __global__ __launch_bounds__(256) void k_gather_records(
     const u64 stride = (u64)gridDim.x * 256;
     const u32* rin = (const u32*)recs;
     u32* rout = (u32*)out;
-    for (u64 g = (u64)blockIdx.x * 256 + threadIdx.x; g < total_words;
-         g += stride) {
+    u64 g = (u64)blockIdx.x * 256 + threadIdx.x;
+    /* 4 independent loads in flight per iteration */
+    for (; g + 3 * stride < total_words; g += 4 * stride) {
+        u32 v[4];
+        for (int j = 0; j < 4; ++j) {
+            u64 gj = g + (u64)j * stride;
+            u64 rec = gj / rec_words;
+            v[j] = rin[(u64)idx[rec] * rec_words +
+                       (u32)(gj - rec * rec_words)];
+        }
+        for (int j = 0; j < 4; ++j) rout[g + (u64)j * stride] = v[j];
+    }
+    for (; g < total_words; g += stride) {
         u64 rec = g / rec_words;
-        u32 w = (u32)(g - rec * rec_words);
-        rout[g] = rin[(u64)idx[rec] * rec_words + w];
+        rout[g] = rin[(u64)idx[rec] * rec_words +
+                      (u32)(g - rec * rec_words)];
     }
 }
 
