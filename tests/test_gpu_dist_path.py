@@ -1,0 +1,65 @@
+"""Execute the REAL distributed TeraSort/WordCount code path (nccl/RCCL
+process group, broadcast, all_to_all_single, receive-side sort) on a
+single GPU via a world-size-1 group and T9_FORCE_DIST — so the exact
+branch the driver's 8-GPU scaling run takes is exercised before it runs.
+"""
+import os
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from tests import _gpu as G
+
+
+@pytest.fixture()
+def dist_world1():
+    import torch.distributed as dist
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29533")
+    torch.cuda.set_device(0)
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    os.environ["T9_FORCE_DIST"] = "1"
+    yield dist
+    del os.environ["T9_FORCE_DIST"]
+    dist.destroy_process_group()
+
+
+def test_terasort_distributed_branch_world1(dist_world1, oracle):
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from thrill_amd.pipeline import TeraSort
+    n = 200_000
+    ts = TeraSort(n, seed=0x71, rank=0, world=1, device=0)
+    ts.generate()
+    out, n_out = ts.step()
+    assert n_out == n
+    got = G.host(out, np.uint8)[:n * 100].reshape(n, 100)
+    expect = oracle.sort_records(oracle.gen_records(n, seed=0x71))
+    assert np.array_equal(got, expect)
+    ts.close()
+
+
+def test_wordcount_distributed_branch_world1(dist_world1, oracle):
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from thrill_amd.pipeline import WordCount, zipf_cdf
+
+    # force the distributed reduce path too: monkeypatch world attr only
+    # for the exchange logic by running world=1 through step()'s
+    # distributed branch
+    n, vocab = 1 << 18, 20_000
+    wc = WordCount(n, vocab, 1.1, seed=9, rank=0, world=1, device=0)
+    wc.generate()
+    ok, ov, m = wc.step()  # T9_FORCE_DIST: hash partition + all-to-all +
+    # final reduce all execute (self-exchange)
+    gk, gv = G.host(ok, np.uint64), G.host(ov, np.uint64)
+    order = np.argsort(gk)
+    toks = oracle.zipf_tokens(zipf_cdf(vocab, 1.1), n, seed=9)
+    ek, ev = oracle.reduce_u64(toks, np.ones(n, np.uint64))
+    assert np.array_equal(gk[order], ek)
+    assert np.array_equal(gv[order], ev)
+    wc.close()

@@ -117,9 +117,10 @@ class WordCount:
     def step(self):
         """One full ReduceByKey of the (distributed) token stream. Returns
         (keys tensor, vals tensor, m) of this rank's final pairs."""
+        import os
         nat, s = self.nat, _stream()
         ok, ov, m = self._reduce(self.d_toks, self.d_ones, self.n_local)
-        if self.world == 1:
+        if self.world == 1 and not os.environ.get("T9_FORCE_DIST"):
             return ok.clone(), ov.clone(), m
 
         import torch.distributed as dist
@@ -212,6 +213,12 @@ class TeraSort:
         """sample locally, gather to rank 0, select, broadcast. Returns
         device tensors (spl_recs bytes, spl_k64, spl_idx)."""
         import torch.distributed as dist
+        if self.world == 1:
+            # forced-distributed self-exchange: no splitters needed
+            z = torch.zeros(REC, dtype=torch.uint8, device="cuda")
+            zk = torch.zeros(1, dtype=torch.int64, device="cuda")
+            zi = torch.zeros(1, dtype=torch.int64, device="cuda")
+            return z, zk, zi
         S = max(1, sample_size(self.n_total) // self.world)
         stride = max(1, self.n_local // S)
         pos = torch.arange(0, self.n_local, stride, device="cuda")[:S]
@@ -261,9 +268,14 @@ class TeraSort:
 
     def step(self):
         """One full TeraSort of the (distributed) input. Returns the local
-        output tensor (n_out*100 bytes) and n_out."""
+        output tensor (n_out*100 bytes) and n_out.
+
+        T9_FORCE_DIST=1 routes world==1 through the distributed branch
+        (self-exchange) so the exact multi-rank code path is testable on
+        one GPU."""
+        import os
         nat, s = self.nat, _stream()
-        if self.world == 1:
+        if self.world == 1 and not os.environ.get("T9_FORCE_DIST"):
             nat.sort_records(_ptr(self.d_in), _ptr(self.d_out),
                              self.n_local, REC, 10, _ptr(self.d_ws), s)
             return self.d_out, self.n_local
