@@ -191,7 +191,8 @@ class TeraSort:
                 "sort_records", self.n_local + self.n_local // 3 + 16, REC))
         self.d_ws = torch.empty(int(ws_bytes), dtype=torch.uint8,
                                 device="cuda")
-        if world > 1:
+        import os
+        if world > 1 or os.environ.get("T9_FORCE_DIST"):
             self.d_keys = torch.empty(self.n_local, dtype=torch.int64,
                                       device="cuda")
             self.d_idx = torch.empty(self.n_local, dtype=torch.int32,
