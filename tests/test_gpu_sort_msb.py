@@ -1,0 +1,118 @@
+"""GPU parity tests for the two-level MSB radix path (T9_SORT_ALGO=msb),
+including the skew fallbacks (oversize sub-buckets -> ranged LSD; equal
+low-48-bit sub-buckets -> skip)."""
+import os
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from tests import _gpu as G
+    from thrill_amd import Native
+
+
+@pytest.fixture(scope="module")
+def nat():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    n = Native(device=0)
+    yield n
+    n.close()
+
+
+@pytest.fixture(autouse=True)
+def msb_env():
+    os.environ["T9_SORT_ALGO"] = "msb"
+    yield
+    del os.environ["T9_SORT_ALGO"]
+
+
+def sort_pairs(nat, keys, vals):
+    n = len(keys)
+    dk, dv = G.dev(keys), G.dev(vals)
+    w = G.ws(nat.ws("sort_pairs", n))
+    nat.sort_pairs_u64_u32(G.ptr(dk), G.ptr(dv), n, G.ptr(w), G.stream())
+    return G.host(dk, np.uint64), G.host(dv, np.uint32)
+
+
+def check_stable(keys, vals, gk, gv):
+    assert np.array_equal(gk, np.sort(keys))
+    order = np.argsort(keys, kind="stable").astype(np.uint32)
+    assert np.array_equal(gv, vals[order.astype(np.int64)])
+
+
+@pytest.mark.parametrize("n", [1 << 14, 100_001, 1 << 20, (1 << 22) + 17])
+def test_msb_uniform_parity_and_stability(nat, oracle, n):
+    keys = oracle.gen_u64(n, seed=n)
+    vals = np.arange(n, dtype=np.uint32)
+    gk, gv = sort_pairs(nat, keys, vals)
+    check_stable(keys, vals, gk, gv)
+
+
+def test_msb_all_equal_keys(nat):
+    # one (b7,b6) sub-bucket holds everything -> oversize -> ranged LSD;
+    # and the low-48 equal-skip path inside it
+    n = 1 << 18
+    keys = np.full(n, 0xABCDEF0123456789, dtype=np.uint64)
+    vals = np.arange(n, dtype=np.uint32)
+    gk, gv = sort_pairs(nat, keys, vals)
+    check_stable(keys, vals, gk, gv)
+
+
+def test_msb_equal_top16_varying_low(nat):
+    # single oversize sub-bucket with differing low bits -> ranged LSD
+    n = 1 << 18
+    rng = np.random.default_rng(1)
+    keys = (np.uint64(0x7777) << np.uint64(48)) | \
+        rng.integers(0, 1 << 48, n).astype(np.uint64)
+    vals = np.arange(n, dtype=np.uint32)
+    gk, gv = sort_pairs(nat, keys, vals)
+    check_stable(keys, vals, gk, gv)
+
+
+def test_msb_equal_low48(nat):
+    # sub-buckets all take the equal-low-48 skip path
+    n = 1 << 18
+    rng = np.random.default_rng(2)
+    keys = rng.integers(0, 1 << 16, n).astype(np.uint64) << np.uint64(48)
+    vals = np.arange(n, dtype=np.uint32)
+    gk, gv = sort_pairs(nat, keys, vals)
+    check_stable(keys, vals, gk, gv)
+
+
+def test_msb_few_values(nat):
+    # 8 distinct keys: 8 sub-buckets oversize (ranged LSD each, <=64)
+    n = 1 << 18
+    rng = np.random.default_rng(3)
+    keys = rng.integers(0, 8, n).astype(np.uint64) * np.uint64(2**61)
+    vals = np.arange(n, dtype=np.uint32)
+    gk, gv = sort_pairs(nat, keys, vals)
+    check_stable(keys, vals, gk, gv)
+
+
+def test_msb_moderate_skew_many_oversize(nat):
+    # 1024 distinct top-16 values at n=2^20 -> ~1024 oversize sub-buckets
+    # (> 64) -> full-LSD fallback path
+    n = 1 << 20
+    rng = np.random.default_rng(4)
+    keys = (rng.integers(0, 1024, n).astype(np.uint64) << np.uint64(48)) | \
+        rng.integers(0, 1 << 30, n).astype(np.uint64)
+    vals = np.arange(n, dtype=np.uint32)
+    gk, gv = sort_pairs(nat, keys, vals)
+    check_stable(keys, vals, gk, gv)
+
+
+def test_msb_records_end_to_end(nat, oracle):
+    # record sort routed through MSB (sort_records -> sort_pairs dispatch)
+    n = 300_000
+    recs = oracle.gen_records(n, seed=77)
+    din = G.dev(recs.reshape(-1))
+    dout = G.empty(n * 100, np.uint8)
+    w = G.ws(nat.ws("sort_records", n, 100))
+    nat.sort_records(G.ptr(din), G.ptr(dout), n, 100, 10, G.ptr(w),
+                     G.stream())
+    got = G.host(dout, np.uint8).reshape(n, 100)
+    assert np.array_equal(got, oracle.sort_records(recs))

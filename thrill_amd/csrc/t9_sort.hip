@@ -338,111 +338,7 @@ __global__ __launch_bounds__(256, 2) void k_scatter_wave(
     }
 }
 
-/* 512-thread (8-wave) wave-autonomous scatter, always global re-read:
- * doubles waves/SIMD to 4 at the same LDS footprint — the PMC-measured
- * limiter of the 256-thread version was 81% SQ_WAIT_ANY at 2 waves/SIMD.
- * Digit recomputed from the L1/L2-resident re-read instead of an LDS
- * cache. */
-template <int TILE, int BLOCK, bool HAS_KEY, bool HAS_VAL>
-__global__ __launch_bounds__(BLOCK, 4) void k_scatter_wave512(
-    const u64* __restrict__ in_keys, const u32* __restrict__ in_vals,
-    u64* __restrict__ out_keys, u32* __restrict__ out_vals,
-    const u32* __restrict__ offs, u64 n, u32 shift) {
-    constexpr int NW = BLOCK / 64;
-    constexpr int SUB = TILE / NW;
-    constexpr int GROUPS = SUB / 64;
-    __shared__ u64 s_okeys[HAS_KEY ? TILE : 1];
-    __shared__ u32 s_ovals[HAS_VAL ? TILE : 1];
-    __shared__ u16 s_rank[TILE];
-    __shared__ u8 s_digof[TILE];
-    __shared__ u32 s_wcnt[NW * T9_RADIX];
-    __shared__ u32 s_woff[NW * T9_RADIX];
-    __shared__ u32 s_start[T9_RADIX];
-    __shared__ u32 s_goff[T9_RADIX];
-
-    const u32 tid = threadIdx.x, wave = tid >> 6, lane = tid & 63;
-    const u64 base = (u64)blockIdx.x * TILE;
-    const u32 tn = (u32)((n - base < (u64)TILE) ? (n - base) : (u64)TILE);
-
-    if (tid < T9_RADIX)
-        s_goff[tid] = offs[(u64)blockIdx.x * T9_RADIX + tid];
-    for (u32 t = lane; t < T9_RADIX; t += 64) s_wcnt[wave * T9_RADIX + t] = 0;
-
-    const u32 wbase = wave * SUB;
-    for (int g = 0; g < GROUPS; ++g) {
-        const u32 i = wbase + g * 64 + lane;
-        const bool valid = i < tn;
-        u32 d = 0;
-        if (valid) d = (u32)(in_keys[base + i] >> shift) & 255u;
-        u64 m = __ballot(valid);
-        for (int bit = 0; bit < 8; ++bit) {
-            u64 bb = __ballot((d >> bit) & 1u);
-            m &= ((d >> bit) & 1u) ? bb : ~bb;
-        }
-        const u32 wr = (u32)__popcll(m & ((1ull << lane) - 1ull));
-        const u32 before = valid ? s_wcnt[wave * T9_RADIX + d] : 0;
-        if (valid) {
-            s_rank[i] = (u16)(before + wr);
-            if (wr == 0)
-                s_wcnt[wave * T9_RADIX + d] = before + (u32)__popcll(m);
-        }
-    }
-    __syncthreads();
-
-    /* combine (threads 0..255 own digit tid; every barrier is executed by
-     * ALL 512 threads — guarded work, unguarded barriers) */
-    u32 total = 0;
-    if (tid < T9_RADIX) {
-        u32 run = 0;
-        for (int w = 0; w < NW; ++w) {
-            s_woff[w * T9_RADIX + tid] = run;   /* wave-relative for now */
-            run += s_wcnt[w * T9_RADIX + tid];
-        }
-        total = run;                            /* per-digit total */
-        s_start[tid] = run;
-    }
-    __syncthreads();
-    /* exclusive scan of totals (Hillis-Steele over 256 in s_start) */
-    for (int off = 1; off < T9_RADIX; off <<= 1) {
-        u32 y = (tid < T9_RADIX && tid >= (u32)off) ? s_start[tid - off]
-                                                    : 0;
-        __syncthreads();
-        if (tid < T9_RADIX) s_start[tid] += y;
-        __syncthreads();
-    }
-    u32 excl = 0;
-    if (tid < T9_RADIX) excl = s_start[tid] - total;
-    __syncthreads();
-    if (tid < T9_RADIX) {
-        s_start[tid] = excl;
-        for (int w = 0; w < NW; ++w) s_woff[w * T9_RADIX + tid] += excl;
-    }
-    __syncthreads();
-
-    for (int g = 0; g < GROUPS; ++g) {
-        const u32 i = wbase + g * 64 + lane;
-        if (i < tn) {
-            const u64 k = HAS_KEY ? in_keys[base + i] : 0;
-            const u32 d = (u32)(k >> shift) & 255u;
-            const u32 pos = s_woff[wave * T9_RADIX + d] + s_rank[i];
-            if (HAS_KEY) s_okeys[pos] = k;
-            if (HAS_VAL) s_ovals[pos] = in_vals[base + i];
-            s_digof[pos] = (u8)d;
-        }
-    }
-    __syncthreads();
-
-    constexpr int CHUNKS = TILE / BLOCK;
-    for (int c = 0; c < CHUNKS; ++c) {
-        const u32 j = c * BLOCK + tid;
-        if (j < tn) {
-            const u32 d = s_digof[j];
-            const u64 gpos = (u64)s_goff[d] + (j - s_start[d]);
-            if (HAS_KEY) out_keys[gpos] = s_okeys[j];
-            if (HAS_VAL) out_vals[gpos] = s_ovals[j];
-        }
-    }
-}
+#include "t9_rank_scatter.h"
 
 /* bucket offsets (u64, p+1 entries) from the digit base array */
 __global__ __launch_bounds__(512) void k_bucket_offsets(
@@ -674,14 +570,46 @@ int t9_sort_u64(t9_context* ctx, u64* d_keys, u64 n, void* d_workspace,
     return T9_OK;
 }
 
-u64 t9_sort_pairs_workspace(u64 n) {
+u64 t9i_sort_pairs_lsd_workspace(u64 n) {
     if (n < 2) return 256;
     u64 B = t9_ceil_div(n, T9_PAIRS_TILE);
     return t9_align256(n * 8) + t9_align256(n * 4) + scan_ws_bytes(B);
 }
 
+/* MSB implementation (t9_sort_msb.hip) */
+u64 t9i_sort_pairs_msb_workspace(u64 n);
+int t9i_sort_pairs_msb(t9_context*, u64*, u32*, u64, void*, void*);
+/* LSD body, defined below */
+int t9i_sort_pairs_lsd(t9_context*, u64*, u32*, u64, void*, void*);
+
+u64 t9_sort_pairs_workspace(u64 n) {
+    if (n < 2) return 256;
+    u64 a = t9i_sort_pairs_lsd_workspace(n);
+    u64 b = t9i_sort_pairs_msb_workspace(n);
+    return a > b ? a : b;
+}
+
+/* algorithm dispatch: two-level MSB + LDS sub-sort for large n (see
+ * t9_sort_msb.hip), 8-pass LSD otherwise; T9_SORT_ALGO=lsd|msb
+ * overrides. */
 int t9_sort_pairs_u64_u32(t9_context* ctx, u64* d_keys, u32* d_vals, u64 n,
                           void* d_workspace, void* stream) {
+    if (n < 2) return T9_OK;
+    if (!d_keys || !d_vals || !d_workspace || n >= (1ull << 32))
+        return T9_EINVAL;
+    const char* e = getenv("T9_SORT_ALGO");
+    bool use_msb = n >= (1ull << 22);
+    if (e && strcmp(e, "lsd") == 0) use_msb = false;
+    if (e && strcmp(e, "msb") == 0) use_msb = n >= (1ull << 14);
+    if (use_msb)
+        return t9i_sort_pairs_msb(ctx, d_keys, d_vals, n, d_workspace,
+                                  stream);
+    return t9i_sort_pairs_lsd(ctx, d_keys, d_vals, n, d_workspace, stream);
+}
+
+/* the 8-pass LSD pipeline (also the fallback for skewed MSB inputs) */
+int t9i_sort_pairs_lsd(t9_context* ctx, u64* d_keys, u32* d_vals, u64 n,
+                       void* d_workspace, void* stream) {
     (void)ctx;
     if (n < 2) return T9_OK;
     if (!d_keys || !d_vals || !d_workspace || n >= (1ull << 32))

@@ -1,0 +1,553 @@
+/* t9_sort_msb.hip — two-level MSB radix sort for (u64 key, u32 idx) pairs.
+ *
+ * Replaces the 8-pass LSD pipeline for large n with:
+ *   pass 1: stable scatter by key byte 7 into 256 buckets whose starts are
+ *           TILE-aligned (padding), so every pass-2 tile lies in exactly
+ *           one bucket;
+ *   pass 2: stable scatter by key byte 6, segmented per bucket (a
+ *           per-bucket column scan computes the (bucket, digit) bases),
+ *           landing at final compact positions grouped by the top 16 bits;
+ *   level 3: each of the 65536 (b7, b6) sub-buckets (~n/65536 pairs) is
+ *           sorted IN LDS over the remaining 48 bits (6 ballot-ranked
+ *           passes, no HBM traffic beyond one read + one write).
+ *
+ * HBM traffic ≈ 2 scatter passes + 1 LDS-sort read/write ≈ 3×24 B/pair vs
+ * the LSD pipeline's 8×24 B. Every stage is stable, so the composition is
+ * stable (the tie-fix and the parity tests rely on that).
+ *
+ * Skew handling: sub-buckets larger than T9_SUBMAX fall back — few: each
+ * is re-sorted in place by the (stable) LSD path on its range; many
+ * (pathological, e.g. all-equal top bytes): one full LSD sort of the
+ * current (stable-permuted) array. Sub-buckets whose remaining 48 bits are
+ * all equal are skipped entirely (already in stable order), so the
+ * all-equal-key case degrades to ~2 passes, not 8.
+ */
+
+#include "t9_common.h"
+#include "t9_rank_scatter.h"
+
+#include <cstdlib>
+#include <vector>
+
+#define T9_MSB_TILE 8192       /* pass tile (matches scatter v7 geometry) */
+#define T9_SUBMAX 2048         /* max sub-bucket for the LDS sort */
+#define T9_MSB_MIN (1u << 22)  /* below this, plain LSD */
+
+/* kernels defined in t9_sort.hip (non-template, external linkage) */
+__global__ void k_colsum(const u32*, u64, u32*);
+__global__ void k_chunkscan(u32*, u64, u32*);
+__global__ void k_finaloffs(u32*, u64, const u32*, const u32*);
+
+extern "C" int t9i_sort_pairs_lsd(t9_context*, u64*, u32*, u64, void*,
+                                  void*);
+extern "C" u64 t9i_sort_pairs_lsd_workspace(u64 n);
+
+/* ------------------------------------------------------------------ */
+
+/* plain per-block histogram (tile = T9_MSB_TILE, digit from key>>shift) */
+__global__ __launch_bounds__(256) void k_hist_msb(
+    const u64* __restrict__ keys, u64 n, u32 shift, u32* __restrict__ hist) {
+    __shared__ u32 s_cnt[T9_RADIX];
+    const u32 tid = threadIdx.x;
+    const u64 base = (u64)blockIdx.x * T9_MSB_TILE;
+    const u32 tn =
+        (u32)((n - base < (u64)T9_MSB_TILE) ? (n - base) : (u64)T9_MSB_TILE);
+    s_cnt[tid] = 0;
+    __syncthreads();
+    for (u32 i = tid; i < tn; i += 256)
+        atomicAdd(&s_cnt[(u32)(keys[base + i] >> shift) & 255u], 1u);
+    __syncthreads();
+    hist[(u64)blockIdx.x * T9_RADIX + tid] = s_cnt[tid];
+}
+
+/* aligned bucket bases from the pass-1 exclusive digit bases.
+ * true_base = copy of the exclusive scan (compact positions);
+ * abase[d] = TILE-aligned start of bucket d in the padded layout,
+ * abase[256] = padded total; bucket_n[d] = count. One block. */
+__global__ __launch_bounds__(256) void k_align_bases(
+    const u32* __restrict__ digit_base, u64 n, u32* __restrict__ true_base,
+    u32* __restrict__ abase, u32* __restrict__ bucket_n) {
+    const u32 tid = threadIdx.x;
+    __shared__ u32 s[T9_RADIX];
+    const u32 b0 = digit_base[tid];
+    const u32 b1 = (tid == 255) ? (u32)n : digit_base[tid + 1];
+    const u32 cnt = b1 - b0;
+    true_base[tid] = b0;
+    bucket_n[tid] = cnt;
+    const u32 padded =
+        (cnt + T9_MSB_TILE - 1) / T9_MSB_TILE * T9_MSB_TILE;
+    s[tid] = padded;
+    __syncthreads();
+    for (int off = 1; off < T9_RADIX; off <<= 1) {
+        u32 y = (tid >= (u32)off) ? s[tid - off] : 0;
+        __syncthreads();
+        s[tid] += y;
+        __syncthreads();
+    }
+    abase[tid] = s[tid] - padded;
+    if (tid == 255) abase[256] = s[255];
+}
+
+/* find the bucket containing padded position `pos` (abase in LDS) */
+__device__ inline u32 bucket_of(const u32* s_abase, u32 pos) {
+    u32 lo = 0, hi = 255;
+    while (lo < hi) {
+        u32 mid = (lo + hi + 1) >> 1;
+        if (s_abase[mid] <= pos) lo = mid; else hi = mid - 1;
+    }
+    return lo;
+}
+
+/* segmented per-block histogram over the padded pass-1 output */
+__global__ __launch_bounds__(256) void k_hist_seg(
+    const u64* __restrict__ keys, const u32* __restrict__ abase,
+    const u32* __restrict__ bucket_n, u32 shift, u32* __restrict__ hist) {
+    __shared__ u32 s_cnt[T9_RADIX];
+    __shared__ u32 s_abase[257];
+    const u32 tid = threadIdx.x;
+    s_abase[tid] = abase[tid];
+    if (tid == 0) s_abase[256] = abase[256];
+    s_cnt[tid] = 0;
+    __syncthreads();
+    const u32 tbase = (u32)(blockIdx.x * T9_MSB_TILE);
+    if (tbase < s_abase[256]) {
+        const u32 b = bucket_of(s_abase, tbase);
+        const u32 off = tbase - s_abase[b];
+        const u32 tn = (bucket_n[b] > off)
+                           ? ((bucket_n[b] - off < T9_MSB_TILE)
+                                  ? bucket_n[b] - off
+                                  : T9_MSB_TILE)
+                           : 0;
+        for (u32 i = tid; i < tn; i += 256)
+            atomicAdd(&s_cnt[(u32)(keys[tbase + i] >> shift) & 255u], 1u);
+    }
+    __syncthreads();
+    hist[(u64)blockIdx.x * T9_RADIX + tid] = s_cnt[tid];
+}
+
+/* per-bucket segmented column scan: block b scans its bucket's hist rows,
+ * producing final global offsets (compact positions, from true_base) in
+ * place, plus sub_start/sub_n for the 256 (b, b6) sub-buckets. */
+__global__ __launch_bounds__(256) void k_seg_scan(
+    u32* __restrict__ hist, const u32* __restrict__ abase,
+    const u32* __restrict__ bucket_n, const u32* __restrict__ true_base,
+    u32* __restrict__ sub_start, u32* __restrict__ sub_n) {
+    const u32 b = blockIdx.x;
+    const u32 tid = threadIdx.x;
+    const u32 r0 = abase[b] / T9_MSB_TILE;
+    const u32 rows = (bucket_n[b] + T9_MSB_TILE - 1) / T9_MSB_TILE;
+    u32 total = 0;
+    for (u32 r = 0; r < rows; ++r)
+        total += hist[(u64)(r0 + r) * T9_RADIX + tid];
+    /* exclusive scan of totals across digits */
+    __shared__ u32 s[T9_RADIX];
+    s[tid] = total;
+    __syncthreads();
+    for (int off = 1; off < T9_RADIX; off <<= 1) {
+        u32 y = (tid >= (u32)off) ? s[tid - off] : 0;
+        __syncthreads();
+        s[tid] += y;
+        __syncthreads();
+    }
+    const u32 excl = s[tid] - total;
+    const u32 gstart = true_base[b] + excl;
+    sub_start[(u64)b * T9_RADIX + tid] = gstart;
+    sub_n[(u64)b * T9_RADIX + tid] = total;
+    u32 run = gstart;
+    for (u32 r = 0; r < rows; ++r) {
+        u32 v = hist[(u64)(r0 + r) * T9_RADIX + tid];
+        hist[(u64)(r0 + r) * T9_RADIX + tid] = run;
+        run += v;
+    }
+}
+
+/* segmented 1024-thread wave-autonomous stable scatter (pass 2): like
+ * k_scatter_wave512 but the tile's element range comes from the padded
+ * bucket layout. */
+__global__ __launch_bounds__(1024, 4) void k_scatter_seg(
+    const u64* __restrict__ in_keys, const u32* __restrict__ in_vals,
+    const u32* __restrict__ abase, const u32* __restrict__ bucket_n,
+    u64* __restrict__ out_keys, u32* __restrict__ out_vals,
+    const u32* __restrict__ offs, u32 shift) {
+    constexpr int TILE = T9_MSB_TILE;
+    constexpr int NW = 16;
+    constexpr int SUB = TILE / NW;
+    constexpr int GROUPS = SUB / 64;
+    __shared__ u64 s_okeys[TILE];
+    __shared__ u32 s_ovals[TILE];
+    __shared__ u16 s_rank[TILE];
+    __shared__ u8 s_digof[TILE];
+    __shared__ u32 s_wcnt[NW * T9_RADIX];
+    __shared__ u32 s_woff[NW * T9_RADIX];
+    __shared__ u32 s_start[T9_RADIX];
+    __shared__ u32 s_goff[T9_RADIX];
+    __shared__ u32 s_abase[257];
+
+    const u32 tid = threadIdx.x, wave = tid >> 6, lane = tid & 63;
+    if (tid < 257) s_abase[tid] = abase[tid];
+    __syncthreads();
+    const u32 tbase = (u32)(blockIdx.x * TILE);
+    if (tbase >= s_abase[256]) return;
+    const u32 b = bucket_of(s_abase, tbase);
+    const u32 off_in_bucket = tbase - s_abase[b];
+    const u32 bn = bucket_n[b];
+    const u32 tn = (bn > off_in_bucket)
+                       ? ((bn - off_in_bucket < (u32)TILE)
+                              ? bn - off_in_bucket
+                              : (u32)TILE)
+                       : 0;
+
+    if (tid < T9_RADIX)
+        s_goff[tid] = offs[(u64)blockIdx.x * T9_RADIX + tid];
+    for (u32 t = lane; t < T9_RADIX; t += 64) s_wcnt[wave * T9_RADIX + t] = 0;
+
+    const u32 wbase = wave * SUB;
+    for (int g = 0; g < GROUPS; ++g) {
+        const u32 i = wbase + g * 64 + lane;
+        const bool valid = i < tn;
+        u32 d = 0;
+        if (valid) d = (u32)(in_keys[tbase + i] >> shift) & 255u;
+        u64 m = __ballot(valid);
+        for (int bit = 0; bit < 8; ++bit) {
+            u64 bb = __ballot((d >> bit) & 1u);
+            m &= ((d >> bit) & 1u) ? bb : ~bb;
+        }
+        const u32 wr = (u32)__popcll(m & ((1ull << lane) - 1ull));
+        const u32 before = valid ? s_wcnt[wave * T9_RADIX + d] : 0;
+        if (valid) {
+            s_rank[i] = (u16)(before + wr);
+            if (wr == 0)
+                s_wcnt[wave * T9_RADIX + d] = before + (u32)__popcll(m);
+        }
+    }
+    __syncthreads();
+
+    u32 total = 0;
+    if (tid < T9_RADIX) {
+        u32 run = 0;
+        for (int w = 0; w < NW; ++w) {
+            s_woff[w * T9_RADIX + tid] = run;
+            run += s_wcnt[w * T9_RADIX + tid];
+        }
+        total = run;
+        s_start[tid] = run;
+    }
+    __syncthreads();
+    for (int off2 = 1; off2 < T9_RADIX; off2 <<= 1) {
+        u32 y = (tid < T9_RADIX && tid >= (u32)off2) ? s_start[tid - off2]
+                                                     : 0;
+        __syncthreads();
+        if (tid < T9_RADIX) s_start[tid] += y;
+        __syncthreads();
+    }
+    u32 excl = 0;
+    if (tid < T9_RADIX) excl = s_start[tid] - total;
+    __syncthreads();
+    if (tid < T9_RADIX) {
+        s_start[tid] = excl;
+        for (int w = 0; w < NW; ++w) s_woff[w * T9_RADIX + tid] += excl;
+    }
+    __syncthreads();
+
+    for (int g = 0; g < GROUPS; ++g) {
+        const u32 i = wbase + g * 64 + lane;
+        if (i < tn) {
+            const u64 k = in_keys[tbase + i];
+            const u32 d = (u32)(k >> shift) & 255u;
+            const u32 pos = s_woff[wave * T9_RADIX + d] + s_rank[i];
+            s_okeys[pos] = k;
+            s_ovals[pos] = in_vals[tbase + i];
+            s_digof[pos] = (u8)d;
+        }
+    }
+    __syncthreads();
+
+    constexpr int CHUNKS = TILE / 1024;
+    for (int c = 0; c < CHUNKS; ++c) {
+        const u32 j = c * 1024 + tid;
+        if (j < tn) {
+            const u32 d = s_digof[j];
+            const u64 gpos = (u64)s_goff[d] + (j - s_start[d]);
+            out_keys[gpos] = s_okeys[j];
+            out_vals[gpos] = s_ovals[j];
+        }
+    }
+}
+
+/* collect oversize sub-buckets */
+__global__ __launch_bounds__(256) void k_oversize(
+    const u32* __restrict__ sub_n, u32 nsub, u32 submax,
+    u32* __restrict__ count, u32* __restrict__ list) {
+    const u32 i = blockIdx.x * 256 + threadIdx.x;
+    if (i < nsub && sub_n[i] > submax) {
+        u32 pos = atomicAdd(count, 1u);
+        list[pos] = i;
+    }
+}
+
+/* level 3: sort one sub-bucket (<= T9_SUBMAX pairs) in LDS over the low
+ * 48 key bits — 6 stable ballot-ranked passes, then write back. 512
+ * threads (8 waves). Sub-buckets with all-equal low-48 bits are skipped
+ * (already in stable order, in place). */
+__global__ __launch_bounds__(512, 2) void k_lds_sort_sub(
+    u64* __restrict__ keys, u32* __restrict__ vals,
+    const u32* __restrict__ sub_start, const u32* __restrict__ sub_n) {
+    constexpr int NW = 8;
+    constexpr int SUBQ = T9_SUBMAX / NW;     /* 256 */
+    constexpr int GROUPS = SUBQ / 64;        /* 4 */
+    __shared__ u64 s_k[2][T9_SUBMAX];
+    __shared__ u32 s_v[2][T9_SUBMAX];
+    __shared__ u16 s_rank[T9_SUBMAX];
+    __shared__ u32 s_wcnt[NW * T9_RADIX];
+    __shared__ u32 s_woff[NW * T9_RADIX];
+    __shared__ u32 s_start[T9_RADIX];
+    __shared__ u32 s_differ;
+
+    const u32 sb = blockIdx.x;
+    const u32 ns = sub_n[sb];
+    if (ns <= 1 || ns > T9_SUBMAX) return;
+    const u32 gbase = sub_start[sb];
+    const u32 tid = threadIdx.x, wave = tid >> 6, lane = tid & 63;
+
+    if (tid == 0) s_differ = 0;
+    __syncthreads();
+    /* load + equal-check over the low 48 bits */
+    const u64 mask48 = 0x0000FFFFFFFFFFFFull;
+    u64 k0ref = keys[gbase] & mask48;
+    for (u32 i = tid; i < ns; i += 512) {
+        u64 k = keys[gbase + i];
+        s_k[0][i] = k;
+        s_v[0][i] = vals[gbase + i];
+        if ((k & mask48) != k0ref) s_differ = 1;
+    }
+    __syncthreads();
+    if (!s_differ) return;   /* stable order already — nothing to do */
+
+    int cur = 0;
+    for (int pass = 0; pass < 6; ++pass) {
+        const u32 shift = pass * 8;
+        for (u32 t = lane; t < T9_RADIX; t += 64)
+            s_wcnt[wave * T9_RADIX + t] = 0;
+        __syncthreads();
+        const u32 wbase = wave * SUBQ;
+        for (int g = 0; g < GROUPS; ++g) {
+            const u32 i = wbase + g * 64 + lane;
+            const bool valid = i < ns;
+            u32 d = 0;
+            if (valid) d = (u32)(s_k[cur][i] >> shift) & 255u;
+            u64 m = __ballot(valid);
+            for (int bit = 0; bit < 8; ++bit) {
+                u64 bb = __ballot((d >> bit) & 1u);
+                m &= ((d >> bit) & 1u) ? bb : ~bb;
+            }
+            const u32 wr = (u32)__popcll(m & ((1ull << lane) - 1ull));
+            const u32 before = valid ? s_wcnt[wave * T9_RADIX + d] : 0;
+            if (valid) {
+                s_rank[i] = (u16)(before + wr);
+                if (wr == 0)
+                    s_wcnt[wave * T9_RADIX + d] =
+                        before + (u32)__popcll(m);
+            }
+        }
+        __syncthreads();
+        u32 total = 0;
+        if (tid < T9_RADIX) {
+            u32 run = 0;
+            for (int w = 0; w < NW; ++w) {
+                s_woff[w * T9_RADIX + tid] = run;
+                run += s_wcnt[w * T9_RADIX + tid];
+            }
+            total = run;
+            s_start[tid] = run;
+        }
+        __syncthreads();
+        for (int off = 1; off < T9_RADIX; off <<= 1) {
+            u32 y = (tid < T9_RADIX && tid >= (u32)off)
+                        ? s_start[tid - off]
+                        : 0;
+            __syncthreads();
+            if (tid < T9_RADIX) s_start[tid] += y;
+            __syncthreads();
+        }
+        u32 excl = 0;
+        if (tid < T9_RADIX) excl = s_start[tid] - total;
+        __syncthreads();
+        if (tid < T9_RADIX)
+            for (int w = 0; w < NW; ++w)
+                s_woff[w * T9_RADIX + tid] = excl + s_woff[w * T9_RADIX + tid];
+        __syncthreads();
+        for (int g = 0; g < GROUPS; ++g) {
+            const u32 i = wbase + g * 64 + lane;
+            if (i < ns) {
+                const u64 k = s_k[cur][i];
+                const u32 d = (u32)(k >> shift) & 255u;
+                const u32 pos = s_woff[wave * T9_RADIX + d] + s_rank[i];
+                s_k[cur ^ 1][pos] = k;
+                s_v[cur ^ 1][pos] = s_v[cur][i];
+            }
+        }
+        __syncthreads();
+        cur ^= 1;
+    }
+    for (u32 i = tid; i < ns; i += 512) {
+        keys[gbase + i] = s_k[cur][i];
+        vals[gbase + i] = s_v[cur][i];
+    }
+}
+
+/* ------------------------------------------------------------------ *
+ * host
+ * ------------------------------------------------------------------ */
+
+namespace {
+struct MsbWs {
+    u64* alt_k;
+    u32* alt_v;
+    u32* hist;
+    u32* chunkpart;
+    u32* digit_base;
+    u32* true_base;
+    u32* abase;
+    u32* bucket_n;
+    u32* sub_start;
+    u32* sub_n;
+    u32* ovr;        /* [0] = count, [1..] = list */
+    u64 B1, B2max;
+};
+
+constexpr u32 NSUB = 256 * 256;
+
+MsbWs carve_msb(char* p, u64 n) {
+    MsbWs w;
+    const u64 npad = n + 256ull * T9_MSB_TILE;
+    w.B1 = t9_ceil_div(n, T9_MSB_TILE);
+    w.B2max = t9_ceil_div(n, T9_MSB_TILE) + 256;
+    w.alt_k = (u64*)p;
+    p += t9_align256(npad * 8);
+    w.alt_v = (u32*)p;
+    p += t9_align256(npad * 4);
+    w.hist = (u32*)p;
+    p += t9_align256(w.B2max * T9_RADIX * 4);
+    w.chunkpart = (u32*)p;
+    p += t9_align256(t9_ceil_div(w.B2max, T9_SCAN_CHUNK) * T9_RADIX * 4);
+    w.digit_base = (u32*)p;
+    p += t9_align256(T9_RADIX * 4);
+    w.true_base = (u32*)p;
+    p += t9_align256(T9_RADIX * 4);
+    w.abase = (u32*)p;
+    p += t9_align256(257 * 4);
+    w.bucket_n = (u32*)p;
+    p += t9_align256(T9_RADIX * 4);
+    w.sub_start = (u32*)p;
+    p += t9_align256((u64)NSUB * 4);
+    w.sub_n = (u32*)p;
+    p += t9_align256((u64)NSUB * 4);
+    w.ovr = (u32*)p;
+    p += t9_align256((u64)(NSUB + 1) * 4);
+    return w;
+}
+
+u64 msb_ws_bytes(u64 n) {
+    const u64 npad = n + 256ull * T9_MSB_TILE;
+    const u64 B2max = t9_ceil_div(n, T9_MSB_TILE) + 256;
+    return t9_align256(npad * 8) + t9_align256(npad * 4) +
+           t9_align256(B2max * T9_RADIX * 4) +
+           t9_align256(t9_ceil_div(B2max, T9_SCAN_CHUNK) * T9_RADIX * 4) +
+           4 * t9_align256(T9_RADIX * 4) + t9_align256(257 * 4) +
+           2 * t9_align256((u64)NSUB * 4) +
+           t9_align256((u64)(NSUB + 1) * 4);
+}
+} // namespace
+
+extern "C" u64 t9i_sort_pairs_msb_workspace(u64 n) { return msb_ws_bytes(n); }
+
+extern "C" int t9i_sort_pairs_msb(t9_context* ctx, u64* d_keys, u32* d_vals,
+                                  u64 n, void* d_workspace, void* stream) {
+    hipStream_t s = (hipStream_t)stream;
+    MsbWs w = carve_msb((char*)d_workspace, n);
+
+    /* ---- pass 1: byte 7 into TILE-aligned buckets (input -> alt) ---- */
+    {
+        const u64 B = w.B1;
+        const u64 Bc = t9_ceil_div(B, T9_SCAN_CHUNK);
+        T9_PERF_WRAP(s, "hist_pairs",
+                     hipLaunchKernelGGL(k_hist_msb, dim3((u32)B), dim3(256),
+                                        0, s, d_keys, n, 56, w.hist));
+        hipLaunchKernelGGL(k_colsum, dim3((u32)Bc), dim3(256), 0, s, w.hist,
+                           B, w.chunkpart);
+        hipLaunchKernelGGL(k_chunkscan, dim3(1), dim3(256), 0, s,
+                           w.chunkpart, Bc, w.digit_base);
+        hipLaunchKernelGGL(k_align_bases, dim3(1), dim3(256), 0, s,
+                           w.digit_base, n, w.true_base, w.abase,
+                           w.bucket_n);
+        hipLaunchKernelGGL(k_finaloffs, dim3((u32)Bc), dim3(256), 0, s,
+                           w.hist, B, w.chunkpart, w.abase);
+        T9_PERF_WRAP(
+            s, "pair_scatter",
+            hipLaunchKernelGGL(
+                (k_scatter_wave512<T9_MSB_TILE, 1024, true, true>),
+                dim3((u32)B), dim3(1024), 0, s, d_keys, d_vals, w.alt_k,
+                w.alt_v, w.hist, n, 56));
+        T9_LAUNCH_CHECK();
+    }
+
+    /* ---- pass 2: byte 6, segmented per bucket (alt -> input, compact
+     * final positions by the top 16 bits) ---- */
+    {
+        const u64 B2 = w.B2max;   /* blocks beyond the padded end exit */
+        T9_PERF_WRAP(s, "hist_pairs",
+                     hipLaunchKernelGGL(k_hist_seg, dim3((u32)B2),
+                                        dim3(256), 0, s, w.alt_k, w.abase,
+                                        w.bucket_n, 48, w.hist));
+        hipLaunchKernelGGL(k_seg_scan, dim3(256), dim3(256), 0, s, w.hist,
+                           w.abase, w.bucket_n, w.true_base, w.sub_start,
+                           w.sub_n);
+        T9_PERF_WRAP(
+            s, "pair_scatter",
+            hipLaunchKernelGGL(k_scatter_seg, dim3((u32)B2), dim3(1024), 0,
+                               s, w.alt_k, w.alt_v, w.abase, w.bucket_n,
+                               d_keys, d_vals, w.hist, 48));
+        T9_LAUNCH_CHECK();
+    }
+
+    /* ---- level 3: in-LDS sort of each (b7, b6) sub-bucket ---- */
+    HIP_TRY(hipMemsetAsync(w.ovr, 0, 4, s));
+    hipLaunchKernelGGL(k_oversize, dim3(NSUB / 256), dim3(256), 0, s,
+                       w.sub_n, NSUB, T9_SUBMAX, w.ovr, w.ovr + 1);
+    T9_PERF_WRAP(s, "lds_sort",
+                 hipLaunchKernelGGL(k_lds_sort_sub, dim3(NSUB), dim3(512),
+                                    0, s, d_keys, d_vals, w.sub_start,
+                                    w.sub_n));
+    T9_LAUNCH_CHECK();
+
+    u32 novr = 0;
+    HIP_TRY(hipMemcpyAsync(&novr, w.ovr, 4, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    if (novr == 0) return T9_OK;
+
+    if (novr > 64) {
+        /* heavy skew: one full stable LSD sort of the current (already
+         * stable-permuted) array — correct, at LSD speed */
+        return t9i_sort_pairs_lsd(ctx, d_keys, d_vals, n, d_workspace,
+                                  stream);
+    }
+    /* few oversize sub-buckets: stable LSD on each range (shares the
+     * workspace, so copy the metadata out first) */
+    std::vector<u32> list(novr);
+    HIP_TRY(hipMemcpy(list.data(), w.ovr + 1, novr * 4,
+                      hipMemcpyDeviceToHost));
+    std::vector<u32> starts(novr), counts(novr);
+    for (u32 i = 0; i < novr; ++i) {
+        HIP_TRY(hipMemcpy(&starts[i], w.sub_start + list[i], 4,
+                          hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(&counts[i], w.sub_n + list[i], 4,
+                          hipMemcpyDeviceToHost));
+    }
+    for (u32 i = 0; i < novr; ++i) {
+        int rc = t9i_sort_pairs_lsd(ctx, d_keys + starts[i],
+                                    d_vals + starts[i], counts[i],
+                                    d_workspace, stream);
+        if (rc) return rc;
+    }
+    return T9_OK;
+}
