@@ -30,7 +30,7 @@
 #include <vector>
 
 #define T9_MSB_TILE 8192       /* pass tile (matches scatter v7 geometry) */
-#define T9_SUBMAX 2048         /* max sub-bucket for the LDS sort */
+#define T9_SUBMAX 4096         /* max sub-bucket for the LDS sort */
 #define T9_MSB_MIN (1u << 22)  /* below this, plain LSD */
 
 /* kernels defined in t9_sort.hip (non-template, external linkage) */
@@ -286,13 +286,14 @@ __global__ __launch_bounds__(256) void k_oversize(
 }
 
 /* level 3: sort one sub-bucket (<= T9_SUBMAX pairs) in LDS over the low
- * 48 key bits — 6 stable ballot-ranked passes, then write back. 512
- * threads (8 waves). Sub-buckets with all-equal low-48 bits are skipped
- * (already in stable order, in place). */
-__global__ __launch_bounds__(512, 2) void k_lds_sort_sub(
+ * 48 key bits — 6 stable ballot-ranked passes, then write back. 1024
+ * threads (16 waves, 137 KB LDS, 4 waves/SIMD). Sub-buckets with
+ * all-equal low-48 bits are skipped (already in stable order, in
+ * place). */
+__global__ __launch_bounds__(1024, 4) void k_lds_sort_sub(
     u64* __restrict__ keys, u32* __restrict__ vals,
     const u32* __restrict__ sub_start, const u32* __restrict__ sub_n) {
-    constexpr int NW = 8;
+    constexpr int NW = 16;
     constexpr int SUBQ = T9_SUBMAX / NW;     /* 256 */
     constexpr int GROUPS = SUBQ / 64;        /* 4 */
     __shared__ u64 s_k[2][T9_SUBMAX];
@@ -314,7 +315,7 @@ __global__ __launch_bounds__(512, 2) void k_lds_sort_sub(
     /* load + equal-check over the low 48 bits */
     const u64 mask48 = 0x0000FFFFFFFFFFFFull;
     u64 k0ref = keys[gbase] & mask48;
-    for (u32 i = tid; i < ns; i += 512) {
+    for (u32 i = tid; i < ns; i += 1024) {
         u64 k = keys[gbase + i];
         s_k[0][i] = k;
         s_v[0][i] = vals[gbase + i];
@@ -389,7 +390,7 @@ __global__ __launch_bounds__(512, 2) void k_lds_sort_sub(
         __syncthreads();
         cur ^= 1;
     }
-    for (u32 i = tid; i < ns; i += 512) {
+    for (u32 i = tid; i < ns; i += 1024) {
         keys[gbase + i] = s_k[cur][i];
         vals[gbase + i] = s_v[cur][i];
     }
@@ -515,7 +516,7 @@ extern "C" int t9i_sort_pairs_msb(t9_context* ctx, u64* d_keys, u32* d_vals,
     hipLaunchKernelGGL(k_oversize, dim3(NSUB / 256), dim3(256), 0, s,
                        w.sub_n, NSUB, T9_SUBMAX, w.ovr, w.ovr + 1);
     T9_PERF_WRAP(s, "lds_sort",
-                 hipLaunchKernelGGL(k_lds_sort_sub, dim3(NSUB), dim3(512),
+                 hipLaunchKernelGGL(k_lds_sort_sub, dim3(NSUB), dim3(1024),
                                     0, s, d_keys, d_vals, w.sub_start,
                                     w.sub_n));
     T9_LAUNCH_CHECK();
