@@ -149,7 +149,7 @@ def main():
     ts.nat.perf_enable(False)
     scat_ms, scat_n = ts.nat.perf_read("pair_scatter")
     perf_breakdown = {}
-    for cls in ["pair_scatter", "hist_pairs", "extract", "gather"]:
+    for cls in ["pair_scatter", "hist_pairs", "lds_sort", "extract", "gather"]:
         ms, cnt = ts.nat.perf_read(cls)
         perf_breakdown[cls] = {"total_ms": round(ms, 3), "launches": cnt}
     ts.nat.perf_reset()

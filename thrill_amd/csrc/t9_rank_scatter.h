@@ -4,6 +4,28 @@
 #pragma once
 #include "t9_common.h"
 
+/* exclusive scan of 256 LDS entries by wave 0 alone (4 shfl-scanned
+ * 64-lane chunks with a running carry) — replaces the 256-thread
+ * Hillis-Steele scan and its 16 block barriers with one. Caller barriers
+ * before (values visible to wave 0) and after. */
+__device__ inline void t9_scan256_onewave(u32* s_vals, u32 tid) {
+    if (tid < 64) {
+        const u32 lane = tid;
+        u32 carry = 0;
+        for (int c = 0; c < 4; ++c) {
+            const u32 v0 = s_vals[c * 64 + lane];
+            u32 v = v0;
+            for (int off = 1; off < 64; off <<= 1) {
+                u32 y = __shfl_up(v, off);
+                if (lane >= (u32)off) v += y;
+            }
+            const u32 total = __shfl(v, 63);
+            s_vals[c * 64 + lane] = (v - v0) + carry;
+            carry += total;
+        }
+    }
+}
+
 /* 512-thread (8-wave) wave-autonomous scatter, always global re-read:
  * doubles waves/SIMD to 4 at the same LDS footprint — the PMC-measured
  * limiter of the 256-thread version was 81% SQ_WAIT_ANY at 2 waves/SIMD.
@@ -55,32 +77,20 @@ __global__ __launch_bounds__(BLOCK, 4) void k_scatter_wave512(
     }
     __syncthreads();
 
-    /* combine (threads 0..255 own digit tid; every barrier is executed by
-     * ALL 512 threads — guarded work, unguarded barriers) */
-    u32 total = 0;
+    /* combine (threads 0..255 own digit tid) */
     if (tid < T9_RADIX) {
         u32 run = 0;
         for (int w = 0; w < NW; ++w) {
             s_woff[w * T9_RADIX + tid] = run;   /* wave-relative for now */
             run += s_wcnt[w * T9_RADIX + tid];
         }
-        total = run;                            /* per-digit total */
         s_start[tid] = run;
     }
     __syncthreads();
-    /* exclusive scan of totals (Hillis-Steele over 256 in s_start) */
-    for (int off = 1; off < T9_RADIX; off <<= 1) {
-        u32 y = (tid < T9_RADIX && tid >= (u32)off) ? s_start[tid - off]
-                                                    : 0;
-        __syncthreads();
-        if (tid < T9_RADIX) s_start[tid] += y;
-        __syncthreads();
-    }
-    u32 excl = 0;
-    if (tid < T9_RADIX) excl = s_start[tid] - total;
+    t9_scan256_onewave(s_start, tid);   /* s_start: totals -> exclusive */
     __syncthreads();
     if (tid < T9_RADIX) {
-        s_start[tid] = excl;
+        const u32 excl = s_start[tid];
         for (int w = 0; w < NW; ++w) s_woff[w * T9_RADIX + tid] += excl;
     }
     __syncthreads();

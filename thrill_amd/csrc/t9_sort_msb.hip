@@ -222,29 +222,19 @@ __global__ __launch_bounds__(1024, 4) void k_scatter_seg(
     }
     __syncthreads();
 
-    u32 total = 0;
     if (tid < T9_RADIX) {
         u32 run = 0;
         for (int w = 0; w < NW; ++w) {
-            s_woff[w * T9_RADIX + tid] = run;
+            s_woff[w * T9_RADIX + tid] = run;   /* wave-relative for now */
             run += s_wcnt[w * T9_RADIX + tid];
         }
-        total = run;
         s_start[tid] = run;
     }
     __syncthreads();
-    for (int off2 = 1; off2 < T9_RADIX; off2 <<= 1) {
-        u32 y = (tid < T9_RADIX && tid >= (u32)off2) ? s_start[tid - off2]
-                                                     : 0;
-        __syncthreads();
-        if (tid < T9_RADIX) s_start[tid] += y;
-        __syncthreads();
-    }
-    u32 excl = 0;
-    if (tid < T9_RADIX) excl = s_start[tid] - total;
+    t9_scan256_onewave(s_start, tid);   /* s_start: totals -> exclusive */
     __syncthreads();
     if (tid < T9_RADIX) {
-        s_start[tid] = excl;
+        const u32 excl = s_start[tid];
         for (int w = 0; w < NW; ++w) s_woff[w * T9_RADIX + tid] += excl;
     }
     __syncthreads();
@@ -351,31 +341,21 @@ __global__ __launch_bounds__(1024, 4) void k_lds_sort_sub(
             }
         }
         __syncthreads();
-        u32 total = 0;
         if (tid < T9_RADIX) {
             u32 run = 0;
             for (int w = 0; w < NW; ++w) {
-                s_woff[w * T9_RADIX + tid] = run;
+                s_woff[w * T9_RADIX + tid] = run;   /* wave-relative for now */
                 run += s_wcnt[w * T9_RADIX + tid];
             }
-            total = run;
             s_start[tid] = run;
         }
         __syncthreads();
-        for (int off = 1; off < T9_RADIX; off <<= 1) {
-            u32 y = (tid < T9_RADIX && tid >= (u32)off)
-                        ? s_start[tid - off]
-                        : 0;
-            __syncthreads();
-            if (tid < T9_RADIX) s_start[tid] += y;
-            __syncthreads();
-        }
-        u32 excl = 0;
-        if (tid < T9_RADIX) excl = s_start[tid] - total;
+        t9_scan256_onewave(s_start, tid);   /* s_start: totals -> exclusive */
         __syncthreads();
-        if (tid < T9_RADIX)
-            for (int w = 0; w < NW; ++w)
-                s_woff[w * T9_RADIX + tid] = excl + s_woff[w * T9_RADIX + tid];
+        if (tid < T9_RADIX) {
+            const u32 excl = s_start[tid];
+            for (int w = 0; w < NW; ++w) s_woff[w * T9_RADIX + tid] += excl;
+        }
         __syncthreads();
         for (int g = 0; g < GROUPS; ++g) {
             const u32 i = wbase + g * 64 + lane;
