@@ -264,31 +264,38 @@ __global__ __launch_bounds__(1024, 4) void k_scatter_seg(
     }
 }
 
-/* collect oversize sub-buckets */
-__global__ __launch_bounds__(256) void k_oversize(
-    const u32* __restrict__ sub_n, u32 nsub, u32 submax,
-    u32* __restrict__ count, u32* __restrict__ list) {
+/* sub-bucket stats: info[0] = max size (atomicMax), info[1] = count of
+ * sub-buckets above hardmax, list = their indices */
+__global__ __launch_bounds__(256) void k_subinfo(
+    const u32* __restrict__ sub_n, u32 nsub, u32 hardmax,
+    u32* __restrict__ info, u32* __restrict__ list) {
     const u32 i = blockIdx.x * 256 + threadIdx.x;
-    if (i < nsub && sub_n[i] > submax) {
-        u32 pos = atomicAdd(count, 1u);
-        list[pos] = i;
+    if (i < nsub) {
+        const u32 v = sub_n[i];
+        atomicMax(&info[0], v);
+        if (v > hardmax) {
+            u32 pos = atomicAdd(&info[1], 1u);
+            list[pos] = i;
+        }
     }
 }
 
-/* level 3: sort one sub-bucket (<= T9_SUBMAX pairs) in LDS over the low
- * 48 key bits — 6 stable ballot-ranked passes, then write back. 1024
- * threads (16 waves, 137 KB LDS, 4 waves/SIMD). Sub-buckets with
- * all-equal low-48 bits are skipped (already in stable order, in
- * place). */
-__global__ __launch_bounds__(1024, 4) void k_lds_sort_sub(
+/* level 3: sort one sub-bucket (<= SUBMAX pairs) in LDS over the low
+ * 48 key bits — 6 stable ballot-ranked passes, then write back.
+ * <2048,512>: 70 KB LDS, 2 blocks/CU (cross-block overlap hides the
+ * pass barriers); <4096,1024>: 137 KB, 1 block/CU — chosen by the host
+ * from the measured max sub-bucket size. Sub-buckets with all-equal
+ * low-48 bits are skipped (already in stable order, in place). */
+template <int SUBMAX, int BLOCK>
+__global__ __launch_bounds__(BLOCK, 4) void k_lds_sort_sub(
     u64* __restrict__ keys, u32* __restrict__ vals,
     const u32* __restrict__ sub_start, const u32* __restrict__ sub_n) {
-    constexpr int NW = 16;
-    constexpr int SUBQ = T9_SUBMAX / NW;     /* 256 */
+    constexpr int NW = BLOCK / 64;
+    constexpr int SUBQ = SUBMAX / NW;        /* 256 */
     constexpr int GROUPS = SUBQ / 64;        /* 4 */
-    __shared__ u64 s_k[2][T9_SUBMAX];
-    __shared__ u32 s_v[2][T9_SUBMAX];
-    __shared__ u16 s_rank[T9_SUBMAX];
+    __shared__ u64 s_k[2][SUBMAX];
+    __shared__ u32 s_v[2][SUBMAX];
+    __shared__ u16 s_rank[SUBMAX];
     __shared__ u32 s_wcnt[NW * T9_RADIX];
     __shared__ u32 s_woff[NW * T9_RADIX];
     __shared__ u32 s_start[T9_RADIX];
@@ -296,7 +303,7 @@ __global__ __launch_bounds__(1024, 4) void k_lds_sort_sub(
 
     const u32 sb = blockIdx.x;
     const u32 ns = sub_n[sb];
-    if (ns <= 1 || ns > T9_SUBMAX) return;
+    if (ns <= 1 || ns > (u32)SUBMAX) return;
     const u32 gbase = sub_start[sb];
     const u32 tid = threadIdx.x, wave = tid >> 6, lane = tid & 63;
 
@@ -305,7 +312,7 @@ __global__ __launch_bounds__(1024, 4) void k_lds_sort_sub(
     /* load + equal-check over the low 48 bits */
     const u64 mask48 = 0x0000FFFFFFFFFFFFull;
     u64 k0ref = keys[gbase] & mask48;
-    for (u32 i = tid; i < ns; i += 1024) {
+    for (u32 i = tid; i < ns; i += BLOCK) {
         u64 k = keys[gbase + i];
         s_k[0][i] = k;
         s_v[0][i] = vals[gbase + i];
@@ -370,7 +377,7 @@ __global__ __launch_bounds__(1024, 4) void k_lds_sort_sub(
         __syncthreads();
         cur ^= 1;
     }
-    for (u32 i = tid; i < ns; i += 1024) {
+    for (u32 i = tid; i < ns; i += BLOCK) {
         keys[gbase + i] = s_k[cur][i];
         vals[gbase + i] = s_v[cur][i];
     }
@@ -492,18 +499,24 @@ extern "C" int t9i_sort_pairs_msb(t9_context* ctx, u64* d_keys, u32* d_vals,
     }
 
     /* ---- level 3: in-LDS sort of each (b7, b6) sub-bucket ---- */
-    HIP_TRY(hipMemsetAsync(w.ovr, 0, 4, s));
-    hipLaunchKernelGGL(k_oversize, dim3(NSUB / 256), dim3(256), 0, s,
-                       w.sub_n, NSUB, T9_SUBMAX, w.ovr, w.ovr + 1);
-    T9_PERF_WRAP(s, "lds_sort",
-                 hipLaunchKernelGGL(k_lds_sort_sub, dim3(NSUB), dim3(1024),
-                                    0, s, d_keys, d_vals, w.sub_start,
-                                    w.sub_n));
-    T9_LAUNCH_CHECK();
-
-    u32 novr = 0;
-    HIP_TRY(hipMemcpyAsync(&novr, w.ovr, 4, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipMemsetAsync(w.ovr, 0, 8, s));
+    hipLaunchKernelGGL(k_subinfo, dim3(NSUB / 256), dim3(256), 0, s,
+                       w.sub_n, NSUB, T9_SUBMAX, w.ovr, w.ovr + 2);
+    u32 info[2] = { 0, 0 };
+    HIP_TRY(hipMemcpyAsync(info, w.ovr, 8, hipMemcpyDeviceToHost, s));
     HIP_TRY(hipStreamSynchronize(s));
+    const u32 maxsub = info[0], novr = info[1];
+    T9_PERF_WRAP(
+        s, "lds_sort",
+        if (maxsub <= 2048)
+            hipLaunchKernelGGL((k_lds_sort_sub<2048, 512>), dim3(NSUB),
+                               dim3(512), 0, s, d_keys, d_vals,
+                               w.sub_start, w.sub_n);
+        else
+            hipLaunchKernelGGL((k_lds_sort_sub<T9_SUBMAX, 1024>),
+                               dim3(NSUB), dim3(1024), 0, s, d_keys,
+                               d_vals, w.sub_start, w.sub_n));
+    T9_LAUNCH_CHECK();
     if (novr == 0) return T9_OK;
 
     if (novr > 64) {
@@ -514,8 +527,9 @@ extern "C" int t9i_sort_pairs_msb(t9_context* ctx, u64* d_keys, u32* d_vals,
     }
     /* few oversize sub-buckets: stable LSD on each range (shares the
      * workspace, so copy the metadata out first) */
+    HIP_TRY(hipStreamSynchronize(s));
     std::vector<u32> list(novr);
-    HIP_TRY(hipMemcpy(list.data(), w.ovr + 1, novr * 4,
+    HIP_TRY(hipMemcpy(list.data(), w.ovr + 2, novr * 4,
                       hipMemcpyDeviceToHost));
     std::vector<u32> starts(novr), counts(novr);
     for (u32 i = 0; i < novr; ++i) {
