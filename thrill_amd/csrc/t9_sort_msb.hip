@@ -431,7 +431,7 @@ MsbWs carve_msb(char* p, u64 n) {
     w.sub_n = (u32*)p;
     p += t9_align256((u64)NSUB * 4);
     w.ovr = (u32*)p;
-    p += t9_align256((u64)(NSUB + 1) * 4);
+    p += t9_align256((u64)(NSUB + 2) * 4);
     return w;
 }
 
@@ -443,7 +443,7 @@ u64 msb_ws_bytes(u64 n) {
            t9_align256(t9_ceil_div(B2max, T9_SCAN_CHUNK) * T9_RADIX * 4) +
            4 * t9_align256(T9_RADIX * 4) + t9_align256(257 * 4) +
            2 * t9_align256((u64)NSUB * 4) +
-           t9_align256((u64)(NSUB + 1) * 4);
+           t9_align256((u64)(NSUB + 2) * 4);
 }
 } // namespace
 
