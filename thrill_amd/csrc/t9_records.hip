@@ -84,10 +84,15 @@ __global__ __launch_bounds__(256) void k_extract_key64(
 /* out[i] = recs[idx[i]], word-wise: consecutive threads write consecutive
  * output words; a wave's 64 word-reads touch only ~3 source records
  * (contiguous lines), which measured faster than 16-byte-per-lane
- * variants that spread each read instruction over ~10 random records. */
+ * variants that spread each read instruction over ~10 random records.
+ * RW is the compile-time record width in words (0 = runtime): the
+ * word->record division then compiles to a multiply-shift instead of a
+ * ~30-instruction emulated u64 division per word. */
+template <int RW>
 __global__ __launch_bounds__(256) void k_gather_records(
     const u8* __restrict__ recs, const u32* __restrict__ idx, u64 n,
-    u32 rec_words, u8* __restrict__ out) {
+    u32 rec_words_rt, u8* __restrict__ out) {
+    const u32 rec_words = RW ? (u32)RW : rec_words_rt;
     const u64 total_words = n * rec_words;
     const u64 stride = (u64)gridDim.x * 256;
     const u32* rin = (const u32*)recs;
@@ -185,11 +190,23 @@ int t9_gather_records(t9_context* ctx, const u8* d_recs, const u32* d_idx,
     (void)ctx;
     if (!d_recs || !d_idx || !d_out || rec_size % 4) return T9_EINVAL;
     if (n == 0) return T9_OK;
-    T9_PERF_WRAP((hipStream_t)stream, "gather",
-                 hipLaunchKernelGGL(k_gather_records,
-                                    dim3(grid_for(n * (rec_size / 4))),
-                                    dim3(256), 0, (hipStream_t)stream,
-                                    d_recs, d_idx, n, rec_size / 4, d_out));
+    const u32 rw = rec_size / 4;
+    const dim3 grid(grid_for(n * rw));
+    hipStream_t s = (hipStream_t)stream;
+    T9_PERF_WRAP(
+        s, "gather",
+        if (rw == 25)
+            hipLaunchKernelGGL((k_gather_records<25>), grid, dim3(256), 0,
+                               s, d_recs, d_idx, n, rw, d_out);
+        else if (rw == 32)
+            hipLaunchKernelGGL((k_gather_records<32>), grid, dim3(256), 0,
+                               s, d_recs, d_idx, n, rw, d_out);
+        else if (rw == 2)
+            hipLaunchKernelGGL((k_gather_records<2>), grid, dim3(256), 0,
+                               s, d_recs, d_idx, n, rw, d_out);
+        else
+            hipLaunchKernelGGL((k_gather_records<0>), grid, dim3(256), 0,
+                               s, d_recs, d_idx, n, rw, d_out));
     T9_LAUNCH_CHECK();
     return T9_OK;
 }
