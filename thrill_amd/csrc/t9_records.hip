@@ -73,22 +73,7 @@ __global__ __launch_bounds__(256) void k_extract_key64(
     u64* __restrict__ keys, u32* __restrict__ idx) {
     const u64 stride = (u64)gridDim.x * 256;
     const u32* r32 = (const u32*)recs;
-    u64 i = (u64)blockIdx.x * 256 + threadIdx.x;
-    for (; i + 7 * stride < n; i += 8 * stride) {
-        u32 w0[8], w1[8];
-        for (int j = 0; j < 8; ++j) {
-            u64 w = (i + (u64)j * stride) * rec_words + key_off / 4;
-            w0[j] = r32[w];
-            w1[j] = r32[w + 1];
-        }
-        for (int j = 0; j < 8; ++j) {
-            u64 ij = i + (u64)j * stride;
-            keys[ij] = ((u64)__builtin_bswap32(w0[j]) << 32) |
-                       __builtin_bswap32(w1[j]);
-            idx[ij] = (u32)ij;
-        }
-    }
-    for (; i < n; i += stride) {
+    for (u64 i = (u64)blockIdx.x * 256 + threadIdx.x; i < n; i += stride) {
         u64 w = (u64)i * rec_words + key_off / 4;
         u32 w0 = r32[w], w1 = r32[w + 1];
         keys[i] = ((u64)__builtin_bswap32(w0) << 32) | __builtin_bswap32(w1);
@@ -113,16 +98,16 @@ __global__ __launch_bounds__(256) void k_gather_records(
     const u32* rin = (const u32*)recs;
     u32* rout = (u32*)out;
     u64 g = (u64)blockIdx.x * 256 + threadIdx.x;
-    /* 8 independent loads in flight per iteration */
-    for (; g + 7 * stride < total_words; g += 8 * stride) {
-        u32 v[8];
-        for (int j = 0; j < 8; ++j) {
+    /* 4 independent loads in flight per iteration */
+    for (; g + 3 * stride < total_words; g += 4 * stride) {
+        u32 v[4];
+        for (int j = 0; j < 4; ++j) {
             u64 gj = g + (u64)j * stride;
             u64 rec = gj / rec_words;
             v[j] = rin[(u64)idx[rec] * rec_words +
                        (u32)(gj - rec * rec_words)];
         }
-        for (int j = 0; j < 8; ++j) rout[g + (u64)j * stride] = v[j];
+        for (int j = 0; j < 4; ++j) rout[g + (u64)j * stride] = v[j];
     }
     for (; g < total_words; g += stride) {
         u64 rec = g / rec_words;
