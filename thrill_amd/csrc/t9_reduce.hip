@@ -174,8 +174,10 @@ int t9_reduce_build(t9_context* ctx, const u64* d_keys, const u64* d_vals,
     HIP_TRY(hipMemsetAsync(d_error, 0, 4, s));
     if (n == 0) return T9_OK;
     if (!d_keys || !d_vals) return T9_EINVAL;
-    hipLaunchKernelGGL(k_reduce_build, dim3(grid_for(n)), dim3(256), 0, s,
-                       d_keys, d_vals, n, d_tk, d_tv, cap, salt, d_error);
+    T9_PERF_WRAP(s, "reduce_build",
+                 hipLaunchKernelGGL(k_reduce_build, dim3(grid_for(n)),
+                                    dim3(256), 0, s, d_keys, d_vals, n,
+                                    d_tk, d_tv, cap, salt, d_error));
     T9_LAUNCH_CHECK();
     return T9_OK;
 }
