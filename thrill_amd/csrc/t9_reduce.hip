@@ -19,6 +19,8 @@
 
 #include "t9_common.h"
 
+#include <cstdlib>
+
 #define T9_EMPTY 0xFFFFFFFFFFFFFFFFull
 
 __global__ __launch_bounds__(256) void k_reduce_init(u64* __restrict__ tk,
@@ -35,7 +37,7 @@ __global__ __launch_bounds__(256) void k_reduce_init(u64* __restrict__ tk,
 __global__ __launch_bounds__(256) void k_reduce_build(
     const u64* __restrict__ keys, const u64* __restrict__ vals, u64 n,
     u64* __restrict__ tk, u64* __restrict__ tv, u64 cap, u64 salt,
-    u32* __restrict__ err) {
+    u32* __restrict__ err, int combine) {
     const u64 gsz = (u64)gridDim.x * 256;
     const u32 lane = threadIdx.x & 63;
     for (u64 base = (u64)blockIdx.x * 256; base < n; base += gsz) {
@@ -53,7 +55,7 @@ __global__ __launch_bounds__(256) void k_reduce_build(
         bool leader = valid;
         u64 gsum = v;
         u32 cnt_same = valid ? 1u : 0u;
-        for (int ofs = 1; ofs < 64; ++ofs) {
+        for (int ofs = 1; combine && ofs < 64; ++ofs) {
             const u64 k_up = __shfl_up(k, ofs);
             const int val_up = __shfl_up((int)valid, ofs);
             if (lane >= (u32)ofs && valid && val_up && k_up == k)
@@ -91,6 +93,100 @@ __global__ __launch_bounds__(256) void k_reduce_build(
                         break;
                     }
                 }
+            }
+        }
+    }
+}
+
+/* LDS-accumulated build: a 2048-slot per-block LDS table absorbs the hot
+ * keys of a skewed stream at LDS-atomic speed (first-come slot claim, up
+ * to 4 probes); misses insert directly into the global table (cold keys
+ * are near-unique, so global contention is negligible). The block flushes
+ * its LDS table once at the end. Plays the role the reference's
+ * in-cache pre-table plays for its CPU cache
+ * (core/reduce_pre_phase.hpp). */
+#define T9_LDS_SLOTS 2048
+__global__ __launch_bounds__(256) void k_reduce_build_lds(
+    const u64* __restrict__ keys, const u64* __restrict__ vals, u64 n,
+    u64* __restrict__ tk, u64* __restrict__ tv, u64 cap, u64 salt,
+    u32* __restrict__ err) {
+    __shared__ u64 lk[T9_LDS_SLOTS];
+    __shared__ u64 lv[T9_LDS_SLOTS];
+    const u32 tid = threadIdx.x;
+    for (u32 s = tid; s < T9_LDS_SLOTS; s += 256) {
+        lk[s] = T9_EMPTY;
+        lv[s] = 0;
+    }
+    __syncthreads();
+
+    const u64 gsz = (u64)gridDim.x * 256;
+    for (u64 i = (u64)blockIdx.x * 256 + tid; i < n; i += gsz) {
+        const u64 k = keys[i];
+        const u64 v = vals[i];
+        if (k == T9_EMPTY) {
+            atomicAdd((unsigned long long*)&tk[cap], 1ull);
+            atomicAdd((unsigned long long*)&tv[cap],
+                      (unsigned long long)v);
+            continue;
+        }
+        const u64 h = t9_hash128to64(salt, k);
+        u32 ls = (u32)(h >> 48) & (T9_LDS_SLOTS - 1);
+        bool done = false;
+        for (int p = 0; p < 4; ++p) {
+            u64 prev = atomicCAS((unsigned long long*)&lk[ls],
+                                 (unsigned long long)T9_EMPTY,
+                                 (unsigned long long)k);
+            if (prev == T9_EMPTY || prev == k) {
+                atomicAdd((unsigned long long*)&lv[ls],
+                          (unsigned long long)v);
+                done = true;
+                break;
+            }
+            ls = (ls + 1) & (T9_LDS_SLOTS - 1);
+        }
+        if (!done) {
+            /* cold path: straight to the global table */
+            u64 slot = h & (cap - 1);
+            u64 probes = 0;
+            for (;;) {
+                u64 prev = atomicCAS((unsigned long long*)&tk[slot],
+                                     (unsigned long long)T9_EMPTY,
+                                     (unsigned long long)k);
+                if (prev == T9_EMPTY || prev == k) {
+                    atomicAdd((unsigned long long*)&tv[slot],
+                              (unsigned long long)v);
+                    break;
+                }
+                slot = (slot + 1) & (cap - 1);
+                if (++probes > cap) {
+                    atomicExch(err, 1u);
+                    break;
+                }
+            }
+        }
+    }
+    __syncthreads();
+
+    /* flush the block's LDS accumulators into the global table */
+    for (u32 s = tid; s < T9_LDS_SLOTS; s += 256) {
+        const u64 k = lk[s];
+        if (k == T9_EMPTY) continue;
+        const u64 v = lv[s];
+        u64 slot = t9_hash128to64(salt, k) & (cap - 1);
+        u64 probes = 0;
+        for (;;) {
+            u64 prev = atomicCAS((unsigned long long*)&tk[slot],
+                                 (unsigned long long)T9_EMPTY,
+                                 (unsigned long long)k);
+            if (prev == T9_EMPTY || prev == k) {
+                atomicAdd((unsigned long long*)&tv[slot],
+                          (unsigned long long)v);
+                break;
+            }
+            slot = (slot + 1) & (cap - 1);
+            if (++probes > cap) {
+                atomicExch(err, 1u);
+                break;
             }
         }
     }
@@ -180,10 +276,18 @@ int t9_reduce_build(t9_context* ctx, const u64* d_keys, const u64* d_vals,
     HIP_TRY(hipMemsetAsync(d_error, 0, 4, s));
     if (n == 0) return T9_OK;
     if (!d_keys || !d_vals) return T9_EINVAL;
-    T9_PERF_WRAP(s, "reduce_build",
-                 hipLaunchKernelGGL(k_reduce_build, dim3(grid_for(n)),
-                                    dim3(256), 0, s, d_keys, d_vals, n,
-                                    d_tk, d_tv, cap, salt, d_error));
+    const char* ce = getenv("T9_REDUCE_COMBINE");
+    const int mode = ce ? atoi(ce) : 2;   /* 2 = LDS table (default) */
+    T9_PERF_WRAP(
+        s, "reduce_build",
+        if (mode == 2)
+            hipLaunchKernelGGL(k_reduce_build_lds, dim3(grid_for(n)),
+                               dim3(256), 0, s, d_keys, d_vals, n, d_tk,
+                               d_tv, cap, salt, d_error);
+        else
+            hipLaunchKernelGGL(k_reduce_build, dim3(grid_for(n)),
+                               dim3(256), 0, s, d_keys, d_vals, n, d_tk,
+                               d_tv, cap, salt, d_error, mode));
     T9_LAUNCH_CHECK();
     return T9_OK;
 }
