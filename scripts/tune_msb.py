@@ -47,4 +47,25 @@ for algo in ["lsd", "msb"]:
     print(json.dumps({algo: {"sort_s": round(t, 4),
                              "Mpairs_per_s": round(n / t / 1e6, 1)}}),
           flush=True)
+wk = G.ws(nat.ws("sort_u64", n))
+for algo in ["lsd", "msb"]:
+    os.environ["T9_SORT_ALGO"] = algo
+
+    def runk():
+        dk.copy_(src_k)
+        nat.sort_u64(G.ptr(dk), n, G.ptr(wk), s)
+
+    runk()
+    torch.cuda.synchronize()
+    signed = dk ^ (-2 ** 63)
+    assert bool((signed[1:] >= signed[:-1]).all().item()), algo
+    ts = []
+    for _ in range(3):
+        t0 = time.perf_counter()
+        runk()
+        torch.cuda.synchronize()
+        ts.append(time.perf_counter() - t0)
+    t = min(ts)
+    print(json.dumps({f"keys_{algo}": {"sort_s": round(t, 4),
+                      "Mkeys_per_s": round(n / t / 1e6, 1)}}), flush=True)
 nat.close()

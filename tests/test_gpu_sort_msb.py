@@ -116,3 +116,24 @@ def test_msb_records_end_to_end(nat, oracle):
                      G.stream())
     got = G.host(dout, np.uint8).reshape(n, 100)
     assert np.array_equal(got, oracle.sort_records(recs))
+
+
+@pytest.mark.parametrize("case", ["uniform", "all_equal", "few_values",
+                                  "high_bits_only", "low48_equal"])
+def test_msb_keys_only(nat, oracle, case):
+    n = 1 << 18
+    rng = np.random.default_rng(hash(case) % 2**31)
+    if case == "uniform":
+        keys = oracle.gen_u64(n, seed=1)
+    elif case == "all_equal":
+        keys = np.full(n, 0x123456789ABCDEF0, dtype=np.uint64)
+    elif case == "few_values":
+        keys = rng.integers(0, 5, n).astype(np.uint64) * np.uint64(2**62)
+    elif case == "high_bits_only":
+        keys = rng.integers(0, 1 << 16, n).astype(np.uint64) << np.uint64(48)
+    else:
+        keys = rng.integers(0, 1 << 30, n).astype(np.uint64)
+    d = G.dev(keys)
+    w = G.ws(nat.ws("sort_u64", n))
+    nat.sort_u64(G.ptr(d), n, G.ptr(w), G.stream())
+    assert np.array_equal(G.host(d, np.uint64), np.sort(keys))

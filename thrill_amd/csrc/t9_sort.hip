@@ -445,6 +445,8 @@ __global__ __launch_bounds__(256) void k_count_tied(
  * host orchestration
  * ------------------------------------------------------------------ */
 
+extern "C" u64 t9i_sort_pairs_msb_workspace(u64 n);
+
 namespace {
 
 struct ScanWs {
@@ -496,14 +498,39 @@ int run_scan(const ScanWs& w, hipStream_t s) {
 
 extern "C" {
 
-u64 t9_sort_u64_workspace(u64 n) {
+u64 t9i_sort_keys_lsd_workspace(u64 n) {
     if (n < 2) return 256;
     u64 B = t9_ceil_div(n, T9_KEYS_TILE);
     return t9_align256(n * 8) + scan_ws_bytes(B);
 }
 
+/* MSB keys implementation (t9_sort_msb.hip) */
+int t9i_sort_keys_msb(t9_context*, u64*, u64, void*, void*);
+int t9i_sort_keys_lsd(t9_context*, u64*, u64, void*, void*);
+
+u64 t9_sort_u64_workspace(u64 n) {
+    if (n < 2) return 256;
+    u64 a = t9i_sort_keys_lsd_workspace(n);
+    u64 b = t9i_sort_pairs_msb_workspace(n);
+    return a > b ? a : b;
+}
+
+/* algorithm dispatch, as for pairs */
 int t9_sort_u64(t9_context* ctx, u64* d_keys, u64 n, void* d_workspace,
                 void* stream) {
+    if (n < 2) return T9_OK;
+    if (!d_keys || !d_workspace || n >= (1ull << 32)) return T9_EINVAL;
+    const char* e = getenv("T9_SORT_ALGO");
+    bool use_msb = n >= (1ull << 22);
+    if (e && strcmp(e, "lsd") == 0) use_msb = false;
+    if (e && strcmp(e, "msb") == 0) use_msb = n >= (1ull << 14);
+    if (use_msb)
+        return t9i_sort_keys_msb(ctx, d_keys, n, d_workspace, stream);
+    return t9i_sort_keys_lsd(ctx, d_keys, n, d_workspace, stream);
+}
+
+int t9i_sort_keys_lsd(t9_context* ctx, u64* d_keys, u64 n,
+                      void* d_workspace, void* stream) {
     (void)ctx;
     if (n < 2) return T9_OK;
     if (!d_keys || !d_workspace || n >= (1ull << 32)) return T9_EINVAL;

@@ -41,6 +41,7 @@ __global__ void k_finaloffs(u32*, u64, const u32*, const u32*);
 extern "C" int t9i_sort_pairs_lsd(t9_context*, u64*, u32*, u64, void*,
                                   void*);
 extern "C" u64 t9i_sort_pairs_lsd_workspace(u64 n);
+extern "C" int t9i_sort_keys_lsd(t9_context*, u64*, u64, void*, void*);
 
 /* ------------------------------------------------------------------ */
 
@@ -164,6 +165,7 @@ __global__ __launch_bounds__(256) void k_seg_scan(
 /* segmented 1024-thread wave-autonomous stable scatter (pass 2): like
  * k_scatter_wave512 but the tile's element range comes from the padded
  * bucket layout. */
+template <bool HAS_VAL>
 __global__ __launch_bounds__(1024, 4) void k_scatter_seg(
     const u64* __restrict__ in_keys, const u32* __restrict__ in_vals,
     const u32* __restrict__ abase, const u32* __restrict__ bucket_n,
@@ -174,7 +176,7 @@ __global__ __launch_bounds__(1024, 4) void k_scatter_seg(
     constexpr int SUB = TILE / NW;
     constexpr int GROUPS = SUB / 64;
     __shared__ u64 s_okeys[TILE];
-    __shared__ u32 s_ovals[TILE];
+    __shared__ u32 s_ovals[HAS_VAL ? TILE : 1];
     __shared__ u16 s_rank[TILE];
     __shared__ u8 s_digof[TILE];
     __shared__ u32 s_wcnt[NW * T9_RADIX];
@@ -246,7 +248,7 @@ __global__ __launch_bounds__(1024, 4) void k_scatter_seg(
             const u32 d = (u32)(k >> shift) & 255u;
             const u32 pos = s_woff[wave * T9_RADIX + d] + s_rank[i];
             s_okeys[pos] = k;
-            s_ovals[pos] = in_vals[tbase + i];
+            if (HAS_VAL) s_ovals[pos] = in_vals[tbase + i];
             s_digof[pos] = (u8)d;
         }
     }
@@ -259,7 +261,7 @@ __global__ __launch_bounds__(1024, 4) void k_scatter_seg(
             const u32 d = s_digof[j];
             const u64 gpos = (u64)s_goff[d] + (j - s_start[d]);
             out_keys[gpos] = s_okeys[j];
-            out_vals[gpos] = s_ovals[j];
+            if (HAS_VAL) out_vals[gpos] = s_ovals[j];
         }
     }
 }
@@ -286,7 +288,7 @@ __global__ __launch_bounds__(256) void k_subinfo(
  * pass barriers); <4096,1024>: 137 KB, 1 block/CU — chosen by the host
  * from the measured max sub-bucket size. Sub-buckets with all-equal
  * low-48 bits are skipped (already in stable order, in place). */
-template <int SUBMAX, int BLOCK>
+template <int SUBMAX, int BLOCK, bool HAS_VAL>
 __global__ __launch_bounds__(BLOCK, 4) void k_lds_sort_sub(
     u64* __restrict__ keys, u32* __restrict__ vals,
     const u32* __restrict__ sub_start, const u32* __restrict__ sub_n) {
@@ -294,7 +296,7 @@ __global__ __launch_bounds__(BLOCK, 4) void k_lds_sort_sub(
     constexpr int SUBQ = SUBMAX / NW;        /* 256 */
     constexpr int GROUPS = SUBQ / 64;        /* 4 */
     __shared__ u64 s_k[2][SUBMAX];
-    __shared__ u32 s_v[2][SUBMAX];
+    __shared__ u32 s_v[2][HAS_VAL ? SUBMAX : 1];
     __shared__ u16 s_rank[SUBMAX];
     __shared__ u32 s_wcnt[NW * T9_RADIX];
     __shared__ u32 s_woff[NW * T9_RADIX];
@@ -315,7 +317,7 @@ __global__ __launch_bounds__(BLOCK, 4) void k_lds_sort_sub(
     for (u32 i = tid; i < ns; i += BLOCK) {
         u64 k = keys[gbase + i];
         s_k[0][i] = k;
-        s_v[0][i] = vals[gbase + i];
+        if (HAS_VAL) s_v[0][i] = vals[gbase + i];
         if ((k & mask48) != k0ref) s_differ = 1;
     }
     __syncthreads();
@@ -371,7 +373,7 @@ __global__ __launch_bounds__(BLOCK, 4) void k_lds_sort_sub(
                 const u32 d = (u32)(k >> shift) & 255u;
                 const u32 pos = s_woff[wave * T9_RADIX + d] + s_rank[i];
                 s_k[cur ^ 1][pos] = k;
-                s_v[cur ^ 1][pos] = s_v[cur][i];
+                if (HAS_VAL) s_v[cur ^ 1][pos] = s_v[cur][i];
             }
         }
         __syncthreads();
@@ -379,7 +381,7 @@ __global__ __launch_bounds__(BLOCK, 4) void k_lds_sort_sub(
     }
     for (u32 i = tid; i < ns; i += BLOCK) {
         keys[gbase + i] = s_k[cur][i];
-        vals[gbase + i] = s_v[cur][i];
+        if (HAS_VAL) vals[gbase + i] = s_v[cur][i];
     }
 }
 
@@ -449,8 +451,9 @@ u64 msb_ws_bytes(u64 n) {
 
 extern "C" u64 t9i_sort_pairs_msb_workspace(u64 n) { return msb_ws_bytes(n); }
 
-extern "C" int t9i_sort_pairs_msb(t9_context* ctx, u64* d_keys, u32* d_vals,
-                                  u64 n, void* d_workspace, void* stream) {
+template <bool HAS_VAL>
+static int sort_msb_impl(t9_context* ctx, u64* d_keys, u32* d_vals, u64 n,
+                         void* d_workspace, void* stream) {
     hipStream_t s = (hipStream_t)stream;
     MsbWs w = carve_msb((char*)d_workspace, n);
 
@@ -473,7 +476,7 @@ extern "C" int t9i_sort_pairs_msb(t9_context* ctx, u64* d_keys, u32* d_vals,
         T9_PERF_WRAP(
             s, "pair_scatter",
             hipLaunchKernelGGL(
-                (k_scatter_wave512<T9_MSB_TILE, 1024, true, true>),
+                (k_scatter_wave512<T9_MSB_TILE, 1024, true, HAS_VAL>),
                 dim3((u32)B), dim3(1024), 0, s, d_keys, d_vals, w.alt_k,
                 w.alt_v, w.hist, n, 56));
         T9_LAUNCH_CHECK();
@@ -492,9 +495,9 @@ extern "C" int t9i_sort_pairs_msb(t9_context* ctx, u64* d_keys, u32* d_vals,
                            w.sub_n);
         T9_PERF_WRAP(
             s, "pair_scatter",
-            hipLaunchKernelGGL(k_scatter_seg, dim3((u32)B2), dim3(1024), 0,
-                               s, w.alt_k, w.alt_v, w.abase, w.bucket_n,
-                               d_keys, d_vals, w.hist, 48));
+            hipLaunchKernelGGL((k_scatter_seg<HAS_VAL>), dim3((u32)B2),
+                               dim3(1024), 0, s, w.alt_k, w.alt_v, w.abase,
+                               w.bucket_n, d_keys, d_vals, w.hist, 48));
         T9_LAUNCH_CHECK();
     }
 
@@ -509,11 +512,11 @@ extern "C" int t9i_sort_pairs_msb(t9_context* ctx, u64* d_keys, u32* d_vals,
     T9_PERF_WRAP(
         s, "lds_sort",
         if (maxsub <= 2048)
-            hipLaunchKernelGGL((k_lds_sort_sub<2048, 512>), dim3(NSUB),
-                               dim3(512), 0, s, d_keys, d_vals,
+            hipLaunchKernelGGL((k_lds_sort_sub<2048, 512, HAS_VAL>),
+                               dim3(NSUB), dim3(512), 0, s, d_keys, d_vals,
                                w.sub_start, w.sub_n);
         else
-            hipLaunchKernelGGL((k_lds_sort_sub<T9_SUBMAX, 1024>),
+            hipLaunchKernelGGL((k_lds_sort_sub<T9_SUBMAX, 1024, HAS_VAL>),
                                dim3(NSUB), dim3(1024), 0, s, d_keys,
                                d_vals, w.sub_start, w.sub_n));
     T9_LAUNCH_CHECK();
@@ -522,8 +525,10 @@ extern "C" int t9i_sort_pairs_msb(t9_context* ctx, u64* d_keys, u32* d_vals,
     if (novr > 64) {
         /* heavy skew: one full stable LSD sort of the current (already
          * stable-permuted) array — correct, at LSD speed */
-        return t9i_sort_pairs_lsd(ctx, d_keys, d_vals, n, d_workspace,
-                                  stream);
+        return HAS_VAL ? t9i_sort_pairs_lsd(ctx, d_keys, d_vals, n,
+                                            d_workspace, stream)
+                       : t9i_sort_keys_lsd(ctx, d_keys, n, d_workspace,
+                                           stream);
     }
     /* few oversize sub-buckets: stable LSD on each range (shares the
      * workspace, so copy the metadata out first) */
@@ -539,10 +544,25 @@ extern "C" int t9i_sort_pairs_msb(t9_context* ctx, u64* d_keys, u32* d_vals,
                           hipMemcpyDeviceToHost));
     }
     for (u32 i = 0; i < novr; ++i) {
-        int rc = t9i_sort_pairs_lsd(ctx, d_keys + starts[i],
-                                    d_vals + starts[i], counts[i],
-                                    d_workspace, stream);
+        int rc = HAS_VAL
+                     ? t9i_sort_pairs_lsd(ctx, d_keys + starts[i],
+                                          d_vals + starts[i], counts[i],
+                                          d_workspace, stream)
+                     : t9i_sort_keys_lsd(ctx, d_keys + starts[i],
+                                         counts[i], d_workspace, stream);
         if (rc) return rc;
     }
     return T9_OK;
+}
+
+extern "C" int t9i_sort_pairs_msb(t9_context* ctx, u64* d_keys,
+                                  u32* d_vals, u64 n, void* d_workspace,
+                                  void* stream) {
+    return sort_msb_impl<true>(ctx, d_keys, d_vals, n, d_workspace, stream);
+}
+
+extern "C" int t9i_sort_keys_msb(t9_context* ctx, u64* d_keys, u64 n,
+                                 void* d_workspace, void* stream) {
+    return sort_msb_impl<false>(ctx, d_keys, nullptr, n, d_workspace,
+                                stream);
 }
