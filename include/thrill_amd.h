@@ -179,6 +179,24 @@ int t9_reduce_drain(t9_context* ctx, const uint64_t* d_table_keys,
                     uint64_t* d_out_keys, uint64_t* d_out_vals,
                     uint64_t* d_out_n, void* stream);
 
+/* ReduceToIndex — reference
+ * api/reduce_to_index.hpp + core/reduce_by_index_post_phase.hpp with the
+ * ReduceByIndex mapping (core/reduce_functional.hpp:84-149): keys are
+ * dense indices in [begin, begin+size); d_dense (size u64, zeroed by the
+ * call) accumulates the per-index sums; absent indices stay at the
+ * neutral 0, as the reference's by-index post phase emits. d_error set
+ * nonzero on out-of-range keys. */
+int t9_reduce_by_index(t9_context* ctx, const uint64_t* d_keys,
+                       const uint64_t* d_vals, uint64_t n, uint64_t begin,
+                       uint64_t size, uint64_t* d_dense, uint32_t* d_error,
+                       void* stream);
+
+/* by-index partition: bucket = (key-begin)*p/size
+ * (core/reduce_functional.hpp:113-128) */
+int t9_index_bucket(t9_context* ctx, const uint64_t* d_keys, uint64_t n,
+                    uint64_t begin, uint64_t size, uint32_t p,
+                    uint32_t* d_bucket, uint64_t* d_counts, void* stream);
+
 /* Zipf(s, q, N) token sampling by inverse CDF (bit-identical to the
  * oracle's t9o_zipf_tokens given the same d_cdf table — the CDF itself is
  * computed once by the oracle/host and copied to the device). Restates

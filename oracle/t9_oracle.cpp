@@ -363,6 +363,24 @@ uint64_t t9o_reduce_u64(const uint64_t* keys, const uint64_t* vals,
 }
 
 /* ------------------------------------------------------------------ */
+/* ReduceToIndex — thrill/api/reduce_to_index.hpp with the ReduceByIndex
+ * mapping (core/reduce_functional.hpp:84-149): keys are dense indices in
+ * [begin, begin+size); the result is the dense value array with the sum
+ * per index (absent indices keep the neutral value 0, as the reference's
+ * by-index post phase emits neutral elements). Returns 0, or -1 if a key
+ * is out of range. */
+int t9o_reduce_by_index(const uint64_t* keys, const uint64_t* vals,
+                        uint64_t n, uint64_t begin, uint64_t size,
+                        uint64_t* dense) {
+    for (uint64_t i = 0; i < size; ++i) dense[i] = 0;
+    for (uint64_t i = 0; i < n; ++i) {
+        if (keys[i] < begin || keys[i] - begin >= size) return -1;
+        dense[keys[i] - begin] += vals[i];
+    }
+    return 0;
+}
+
+/* ------------------------------------------------------------------ */
 /* Zipf(s, N) token sampling by inverse CDF. The reference's
  * common/zipf_distribution.hpp:55-120 draws from std::discrete_distribution
  * over weights 1/(k+q)^s; we restate the same mass function with an explicit

@@ -44,6 +44,8 @@ class Oracle:
         lib.t9o_reduce_u64.restype = u64
         lib.t9o_reduce_u64.argtypes = [
             u64p, u64p, u64, u64, u64, u64p, u64p, u64]
+        lib.t9o_reduce_by_index.restype = ctypes.c_int
+        lib.t9o_reduce_by_index.argtypes = [u64p, u64p, u64, u64, u64, u64p]
         lib.t9o_zipf_cdf.argtypes = [f64p, u64, f64, f64]
         lib.t9o_zipf_tokens.argtypes = [u64p, f64p, u64, u64, u64, u64]
 
@@ -115,6 +117,15 @@ class Oracle:
             keys, vals, len(keys), salt, num_partitions, ok, ov, cap)
         assert m != np.iinfo(np.uint64).max, "oracle reduce: cap exceeded"
         return ok[:m].copy(), ov[:m].copy()
+
+    def reduce_by_index(self, keys, vals, begin, size):
+        keys = np.ascontiguousarray(keys, dtype=np.uint64)
+        vals = np.ascontiguousarray(vals, dtype=np.uint64)
+        dense = np.empty(size, dtype=np.uint64)
+        rc = self._lib.t9o_reduce_by_index(keys, vals, len(keys), begin,
+                                           size, dense)
+        assert rc == 0, "key out of range"
+        return dense
 
     def zipf_cdf(self, N, s, q=0.0):
         cdf = np.empty(N, dtype=np.float64)

@@ -156,3 +156,35 @@ def test_wordcount_pipeline_single_gpu(nat, oracle):
     assert np.array_equal(gk[order], ek)
     assert np.array_equal(gv[order], ev)
     wc.close()
+
+
+def test_reduce_by_index_parity(nat, oracle):
+    # ReduceToIndex: dense per-index u64 sums.
+    rng = np.random.default_rng(17)
+    n, begin, size = 1 << 20, 1000, 5000
+    keys = rng.integers(begin, begin + size, n).astype(np.uint64)
+    vals = rng.integers(0, 1 << 40, n).astype(np.uint64)
+    dk, dv = G.dev(keys), G.dev(vals)
+    dd = G.empty(size, np.uint64)
+    de = G.empty(1, np.uint32)
+    nat.reduce_by_index(G.ptr(dk), G.ptr(dv), n, begin, size, G.ptr(dd),
+                        G.ptr(de), G.stream())
+    assert int(G.host(de, np.uint32)[0]) == 0
+    got = G.host(dd, np.uint64)
+    assert np.array_equal(got, oracle.reduce_by_index(keys, vals, begin,
+                                                      size))
+
+
+def test_index_bucket_mapping(nat):
+    # bucket = (k-begin)*p/size (core/reduce_functional.hpp:113-128)
+    n, begin, size, p = 100_000, 50, 1024, 8
+    rng = np.random.default_rng(18)
+    keys = rng.integers(begin, begin + size, n).astype(np.uint64)
+    dk = G.dev(keys)
+    db = G.empty(n, np.uint32)
+    dc = G.empty(p, np.uint64)
+    nat.index_bucket(G.ptr(dk), n, begin, size, p, G.ptr(db), G.ptr(dc),
+                     G.stream())
+    got = G.host(db, np.uint32)
+    expect = ((keys - begin) * p // size).astype(np.uint32)
+    assert np.array_equal(got, expect)

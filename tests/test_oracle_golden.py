@@ -252,3 +252,15 @@ def test_zipf_tokens_shape(oracle):
     # Zipf(1.1): token 1 must dominate; loose sanity bounds only.
     frac1 = np.mean(toks == 1)
     assert 0.05 < frac1 < 0.5
+
+
+def test_reduce_by_index_oracle(oracle):
+    # ReduceToIndex restatement (api/reduce_to_index.hpp semantics):
+    # dense per-index sums, absent indices neutral.
+    rng = np.random.default_rng(12)
+    keys = rng.integers(100, 150, 5000).astype(np.uint64)
+    vals = rng.integers(0, 1 << 30, 5000).astype(np.uint64)
+    dense = oracle.reduce_by_index(keys, vals, begin=100, size=60)
+    expect = np.zeros(60, dtype=np.uint64)
+    np.add.at(expect, (keys - 100).astype(np.int64), vals)
+    assert np.array_equal(dense, expect)

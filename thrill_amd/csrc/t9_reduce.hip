@@ -192,6 +192,48 @@ __global__ __launch_bounds__(256) void k_reduce_build_lds(
     }
 }
 
+/* ReduceToIndex — reference api/reduce_to_index.hpp with the
+ * ReduceByIndex mapping (core/reduce_functional.hpp:84-149): keys are
+ * dense indices; dense[key-begin] accumulates the u64 sum. Out-of-range
+ * keys set the error flag. */
+__global__ __launch_bounds__(256) void k_reduce_by_index(
+    const u64* __restrict__ keys, const u64* __restrict__ vals, u64 n,
+    u64 begin, u64 size, u64* __restrict__ dense, u32* __restrict__ err) {
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 i = (u64)blockIdx.x * 256 + threadIdx.x; i < n; i += stride) {
+        const u64 k = keys[i];
+        if (k < begin || k - begin >= size) {
+            atomicExch(err, 1u);
+            continue;
+        }
+        atomicAdd((unsigned long long*)&dense[k - begin],
+                  (unsigned long long)vals[i]);
+    }
+}
+
+/* partition for the by-index mapping: bucket = (k-begin)*p / size
+ * (reduce_functional.hpp:113-128 with num_buckets = p) */
+__global__ __launch_bounds__(256) void k_index_bucket(
+    const u64* __restrict__ keys, u64 n, u64 begin, u64 size, u32 p,
+    u32* __restrict__ bucket, u64* __restrict__ counts) {
+    __shared__ u32 scnt[256];
+    const u32 tid = threadIdx.x;
+    scnt[tid] = 0;
+    __syncthreads();
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 i = (u64)blockIdx.x * 256 + tid; i < n; i += stride) {
+        u64 g = keys[i] - begin;
+        if (g >= size) g = size - 1;
+        const u32 b = (u32)(g * p / size);
+        bucket[i] = b;
+        atomicAdd(&scnt[b], 1u);
+    }
+    __syncthreads();
+    if (tid < p && scnt[tid])
+        atomicAdd((unsigned long long*)&counts[tid],
+                  (unsigned long long)scnt[tid]);
+}
+
 __global__ __launch_bounds__(256) void k_reduce_drain(
     const u64* __restrict__ tk, const u64* __restrict__ tv, u64 cap,
     u64* __restrict__ ok, u64* __restrict__ ov, u64* __restrict__ out_n) {
@@ -253,6 +295,38 @@ int t9_hash_bucket(t9_context* ctx, const u64* d_keys, u64 n, u64 salt,
     if (!d_keys || !d_bucket) return T9_EINVAL;
     hipLaunchKernelGGL(k_hash_bucket, dim3(grid_for(n)), dim3(256), 0, s,
                        d_keys, n, salt, p, d_bucket, d_counts);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+int t9_reduce_by_index(t9_context* ctx, const u64* d_keys,
+                       const u64* d_vals, u64 n, u64 begin, u64 size,
+                       u64* d_dense, u32* d_error, void* stream) {
+    (void)ctx;
+    if (!d_dense || !d_error || size == 0) return T9_EINVAL;
+    hipStream_t s = (hipStream_t)stream;
+    HIP_TRY(hipMemsetAsync(d_error, 0, 4, s));
+    HIP_TRY(hipMemsetAsync(d_dense, 0, size * 8, s));
+    if (n == 0) return T9_OK;
+    if (!d_keys || !d_vals) return T9_EINVAL;
+    hipLaunchKernelGGL(k_reduce_by_index, dim3(grid_for(n)), dim3(256), 0,
+                       s, d_keys, d_vals, n, begin, size, d_dense,
+                       d_error);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+int t9_index_bucket(t9_context* ctx, const u64* d_keys, u64 n, u64 begin,
+                    u64 size, u32 p, u32* d_bucket, u64* d_counts,
+                    void* stream) {
+    (void)ctx;
+    if (!d_counts || p < 1 || p > 256 || size == 0) return T9_EINVAL;
+    hipStream_t s = (hipStream_t)stream;
+    HIP_TRY(hipMemsetAsync(d_counts, 0, p * 8, s));
+    if (n == 0) return T9_OK;
+    if (!d_keys || !d_bucket) return T9_EINVAL;
+    hipLaunchKernelGGL(k_index_bucket, dim3(grid_for(n)), dim3(256), 0, s,
+                       d_keys, n, begin, size, p, d_bucket, d_counts);
     T9_LAUNCH_CHECK();
     return T9_OK;
 }
