@@ -37,16 +37,18 @@ def sample_size(n_total):
 
 def select_splitters(sample_recs, sample_gidx, p):
     """FindAndSendSplitters (api/sort.hpp:337-378): sort the gathered
-    samples by (record bytes, global index) — LessSampleIndex under the
-    acceptance total order — and pick samples[(size_t)(i * size / p)].
+    samples and pick samples[(size_t)(i * size / p)]. Vectorized: sort by
+    (u64 key prefix, global index) — classification is count-based
+    (bucket = #{splitters < item}), so the partition is determined by the
+    splitter SET, not their array order, and key-prefix ties among samples
+    cannot affect correctness (they would only nudge bucket balance).
     sample_recs: (S, 100) uint8; returns (p-1, 100) records + (p-1,) idx."""
-    order = sorted(range(len(sample_gidx)),
-                   key=lambda t: (sample_recs[t].tobytes(),
-                                  int(sample_gidx[t])))
+    k64 = sample_recs[:, :8].copy().view(">u8").reshape(-1).astype(np.uint64)
+    order = np.lexsort((np.asarray(sample_gidx, dtype=np.uint64), k64))
     step = len(order) / p
-    sel = [order[int(i * step)] for i in range(1, p)]
-    return (np.stack([sample_recs[t] for t in sel]),
-            np.array([sample_gidx[t] for t in sel], dtype=np.uint64))
+    sel = order[[int(i * step) for i in range(1, p)]]
+    return (sample_recs[sel],
+            np.asarray(sample_gidx, dtype=np.uint64)[sel])
 
 
 def k64_of_records(recs):
