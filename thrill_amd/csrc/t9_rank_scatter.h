@@ -31,7 +31,13 @@ __device__ inline void t9_scan256_onewave(u32* s_vals, u32 tid) {
  * limiter of the 256-thread version was 81% SQ_WAIT_ANY at 2 waves/SIMD.
  * Digit recomputed from the L1/L2-resident re-read instead of an LDS
  * cache. */
-template <int TILE, int BLOCK, bool HAS_KEY, bool HAS_VAL>
+/* REC_WORDS != 0: "fused extract" mode — in_keys is the packed record
+ * array (REC_WORDS u32 words per record); the big-endian u64 key prefix
+ * is built from the record bytes and the payload is the record index
+ * (iota), eliminating the separate extract pass and the packed key-array
+ * round trip for MSB pass 1. */
+template <int TILE, int BLOCK, bool HAS_KEY, bool HAS_VAL,
+          int REC_WORDS = 0>
 __global__ __launch_bounds__(BLOCK, 4) void k_scatter_wave512(
     const u64* __restrict__ in_keys, const u32* __restrict__ in_vals,
     u64* __restrict__ out_keys, u32* __restrict__ out_vals,
@@ -39,6 +45,7 @@ __global__ __launch_bounds__(BLOCK, 4) void k_scatter_wave512(
     constexpr int NW = BLOCK / 64;
     constexpr int SUB = TILE / NW;
     constexpr int GROUPS = SUB / 64;
+    const u32* rec32 = (const u32*)in_keys;
     __shared__ u64 s_okeys[HAS_KEY ? TILE : 1];
     __shared__ u32 s_ovals[HAS_VAL ? TILE : 1];
     __shared__ u16 s_rank[TILE];
@@ -61,7 +68,13 @@ __global__ __launch_bounds__(BLOCK, 4) void k_scatter_wave512(
         const u32 i = wbase + g * 64 + lane;
         const bool valid = i < tn;
         u32 d = 0;
-        if (valid) d = (u32)(in_keys[base + i] >> shift) & 255u;
+        if (valid) {
+            if (REC_WORDS)
+                d = ((const u8*)rec32)[(base + i) * (u64)REC_WORDS * 4 +
+                                       (7 - shift / 8)];
+            else
+                d = (u32)(in_keys[base + i] >> shift) & 255u;
+        }
         u64 m = __ballot(valid);
         for (int bit = 0; bit < 8; ++bit) {
             u64 bb = __ballot((d >> bit) & 1u);
@@ -98,11 +111,22 @@ __global__ __launch_bounds__(BLOCK, 4) void k_scatter_wave512(
     for (int g = 0; g < GROUPS; ++g) {
         const u32 i = wbase + g * 64 + lane;
         if (i < tn) {
-            const u64 k = HAS_KEY ? in_keys[base + i] : 0;
+            u64 k;
+            if (REC_WORDS) {
+                const u64 w0 = (u64)rec32[(base + i) * REC_WORDS];
+                const u64 w1 = (u64)rec32[(base + i) * REC_WORDS + 1];
+                k = ((u64)__builtin_bswap32((u32)w0) << 32) |
+                    __builtin_bswap32((u32)w1);
+            }
+            else {
+                k = HAS_KEY ? in_keys[base + i] : 0;
+            }
             const u32 d = (u32)(k >> shift) & 255u;
             const u32 pos = s_woff[wave * T9_RADIX + d] + s_rank[i];
             if (HAS_KEY) s_okeys[pos] = k;
-            if (HAS_VAL) s_ovals[pos] = in_vals[base + i];
+            if (HAS_VAL)
+                s_ovals[pos] =
+                    REC_WORDS ? (u32)(base + i) : in_vals[base + i];
             s_digof[pos] = (u8)d;
         }
     }

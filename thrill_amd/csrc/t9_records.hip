@@ -18,6 +18,8 @@ extern "C" int t9i_count_tied(const u64* d_keys, u64 n, u32* d_ntied,
 extern "C" u64 t9_sort_pairs_workspace(u64 n);
 extern "C" int t9_sort_pairs_u64_u32(t9_context*, u64*, u32*, u64, void*,
                                      void*);
+extern "C" int t9i_sort_recs_msb(t9_context*, const u8*, u32, u64*, u32*,
+                                 u64, void*, void*);
 
 /* ------------------------------------------------------------------ */
 
@@ -259,10 +261,22 @@ int t9_sort_records(t9_context* ctx, const u8* d_in, u8* d_out, u64 n,
     p += t9_sort_pairs_workspace(n);
     u32* d_ntied = (u32*)p;
 
-    int rc = t9_extract_key64(ctx, d_in, n, rec_size, 0, d_keys, d_idx,
+    const char* e = getenv("T9_SORT_ALGO");
+    bool fused = n >= (1ull << 22) && (rec_size == 100 || rec_size == 128);
+    if (e && strcmp(e, "lsd") == 0) fused = false;
+    if (e && strcmp(e, "msb") == 0 && (rec_size == 100 || rec_size == 128))
+        fused = n >= (1ull << 14);
+    int rc;
+    if (fused) {
+        rc = t9i_sort_recs_msb(ctx, d_in, rec_size, d_keys, d_idx, n,
+                               pair_ws, stream);
+    }
+    else {
+        rc = t9_extract_key64(ctx, d_in, n, rec_size, 0, d_keys, d_idx,
                               stream);
-    if (rc) return rc;
-    rc = t9_sort_pairs_u64_u32(ctx, d_keys, d_idx, n, pair_ws, stream);
+        if (rc) return rc;
+        rc = t9_sort_pairs_u64_u32(ctx, d_keys, d_idx, n, pair_ws, stream);
+    }
     if (rc) return rc;
     rc = t9i_count_tied(d_keys, n, d_ntied, s);
     if (rc) return rc;

@@ -45,7 +45,10 @@ extern "C" int t9i_sort_keys_lsd(t9_context*, u64*, u64, void*, void*);
 
 /* ------------------------------------------------------------------ */
 
-/* plain per-block histogram (tile = T9_MSB_TILE, digit from key>>shift) */
+/* plain per-block histogram (tile = T9_MSB_TILE). RW = 0: digit from the
+ * packed u64 key; RW != 0: keys argument is the record byte array
+ * (RW u32 words per record), digit = big-endian key byte (7 - shift/8). */
+template <int RW>
 __global__ __launch_bounds__(256) void k_hist_msb(
     const u64* __restrict__ keys, u64 n, u32 shift, u32* __restrict__ hist) {
     __shared__ u32 s_cnt[T9_RADIX];
@@ -55,8 +58,15 @@ __global__ __launch_bounds__(256) void k_hist_msb(
         (u32)((n - base < (u64)T9_MSB_TILE) ? (n - base) : (u64)T9_MSB_TILE);
     s_cnt[tid] = 0;
     __syncthreads();
-    for (u32 i = tid; i < tn; i += 256)
-        atomicAdd(&s_cnt[(u32)(keys[base + i] >> shift) & 255u], 1u);
+    const u8* rec8 = (const u8*)keys;
+    for (u32 i = tid; i < tn; i += 256) {
+        u32 d;
+        if (RW)
+            d = rec8[(base + i) * (u64)RW * 4 + (7 - shift / 8)];
+        else
+            d = (u32)(keys[base + i] >> shift) & 255u;
+        atomicAdd(&s_cnt[d], 1u);
+    }
     __syncthreads();
     hist[(u64)blockIdx.x * T9_RADIX + tid] = s_cnt[tid];
 }
@@ -451,8 +461,9 @@ u64 msb_ws_bytes(u64 n) {
 
 extern "C" u64 t9i_sort_pairs_msb_workspace(u64 n) { return msb_ws_bytes(n); }
 
-template <bool HAS_VAL>
-static int sort_msb_impl(t9_context* ctx, u64* d_keys, u32* d_vals, u64 n,
+template <bool HAS_VAL, int RW = 0>
+static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
+                         u64* d_keys, u32* d_vals, u64 n,
                          void* d_workspace, void* stream) {
     hipStream_t s = (hipStream_t)stream;
     MsbWs w = carve_msb((char*)d_workspace, n);
@@ -462,8 +473,9 @@ static int sort_msb_impl(t9_context* ctx, u64* d_keys, u32* d_vals, u64 n,
         const u64 B = w.B1;
         const u64 Bc = t9_ceil_div(B, T9_SCAN_CHUNK);
         T9_PERF_WRAP(s, "hist_pairs",
-                     hipLaunchKernelGGL(k_hist_msb, dim3((u32)B), dim3(256),
-                                        0, s, d_keys, n, 56, w.hist));
+                     hipLaunchKernelGGL((k_hist_msb<RW>), dim3((u32)B),
+                                        dim3(256), 0, s, pass1_src, n, 56,
+                                        w.hist));
         hipLaunchKernelGGL(k_colsum, dim3((u32)Bc), dim3(256), 0, s, w.hist,
                            B, w.chunkpart);
         hipLaunchKernelGGL(k_chunkscan, dim3(1), dim3(256), 0, s,
@@ -476,8 +488,8 @@ static int sort_msb_impl(t9_context* ctx, u64* d_keys, u32* d_vals, u64 n,
         T9_PERF_WRAP(
             s, "pair_scatter",
             hipLaunchKernelGGL(
-                (k_scatter_wave512<T9_MSB_TILE, 1024, true, HAS_VAL>),
-                dim3((u32)B), dim3(1024), 0, s, d_keys, d_vals, w.alt_k,
+                (k_scatter_wave512<T9_MSB_TILE, 1024, true, HAS_VAL, RW>),
+                dim3((u32)B), dim3(1024), 0, s, pass1_src, d_vals, w.alt_k,
                 w.alt_v, w.hist, n, 56));
         T9_LAUNCH_CHECK();
     }
@@ -558,11 +570,29 @@ static int sort_msb_impl(t9_context* ctx, u64* d_keys, u32* d_vals, u64 n,
 extern "C" int t9i_sort_pairs_msb(t9_context* ctx, u64* d_keys,
                                   u32* d_vals, u64 n, void* d_workspace,
                                   void* stream) {
-    return sort_msb_impl<true>(ctx, d_keys, d_vals, n, d_workspace, stream);
+    return sort_msb_impl<true>(ctx, d_keys, d_keys, d_vals, n, d_workspace,
+                               stream);
 }
 
 extern "C" int t9i_sort_keys_msb(t9_context* ctx, u64* d_keys, u64 n,
                                  void* d_workspace, void* stream) {
-    return sort_msb_impl<false>(ctx, d_keys, nullptr, n, d_workspace,
-                                stream);
+    return sort_msb_impl<false>(ctx, d_keys, d_keys, nullptr, n,
+                                d_workspace, stream);
+}
+
+/* fused-extract entry: records in, sorted (key, idx) pairs out — MSB
+ * pass 1 reads the record bytes directly (digit = key byte, payload =
+ * record index), so the separate extract pass and its packed-key round
+ * trip disappear. rec_size/4 must be 25 (100 B) or 32 (128 B). */
+extern "C" int t9i_sort_recs_msb(t9_context* ctx, const u8* d_recs,
+                                 u32 rec_size, u64* d_keys, u32* d_vals,
+                                 u64 n, void* d_workspace, void* stream) {
+    const u32 rw = rec_size / 4;
+    if (rw == 25)
+        return sort_msb_impl<true, 25>(ctx, (const u64*)d_recs, d_keys,
+                                       d_vals, n, d_workspace, stream);
+    if (rw == 32)
+        return sort_msb_impl<true, 32>(ctx, (const u64*)d_recs, d_keys,
+                                       d_vals, n, d_workspace, stream);
+    return T9_EINVAL;
 }
