@@ -261,11 +261,13 @@ int t9_sort_records(t9_context* ctx, const u8* d_in, u8* d_out, u64 n,
     p += t9_sort_pairs_workspace(n);
     u32* d_ntied = (u32*)p;
 
-    const char* e = getenv("T9_SORT_ALGO");
-    bool fused = n >= (1ull << 22) && (rec_size == 100 || rec_size == 128);
-    if (e && strcmp(e, "lsd") == 0) fused = false;
-    if (e && strcmp(e, "msb") == 0 && (rec_size == 100 || rec_size == 128))
-        fused = n >= (1ull << 14);
+    /* fused-extract measured SLOWER than extract + packed-key passes on
+     * 100 B records (byte reads at record stride fetch a full 64 B line
+     * per lane: 17.3 vs 15.5 ms per 10 GiB sort) — kept opt-in for
+     * re-evaluation with wider tiles. */
+    const char* fe = getenv("T9_FUSED_EXTRACT");
+    bool fused = fe && atoi(fe) && n >= (1ull << 14) &&
+                 (rec_size == 100 || rec_size == 128);
     int rc;
     if (fused) {
         rc = t9i_sort_recs_msb(ctx, d_in, rec_size, d_keys, d_idx, n,
