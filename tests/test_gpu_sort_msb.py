@@ -26,8 +26,10 @@ def nat():
 @pytest.fixture(autouse=True)
 def msb_env():
     os.environ["T9_SORT_ALGO"] = "msb"
+    os.environ["T9_FUSED_EXTRACT"] = "1"   # record tests cover both paths
     yield
     del os.environ["T9_SORT_ALGO"]
+    del os.environ["T9_FUSED_EXTRACT"]
 
 
 def sort_pairs(nat, keys, vals):

@@ -90,7 +90,7 @@ __global__ __launch_bounds__(256) void k_extract_key64(
  * RW is the compile-time record width in words (0 = runtime): the
  * word->record division then compiles to a multiply-shift instead of a
  * ~30-instruction emulated u64 division per word. */
-template <int RW>
+template <int RW, bool NT>
 __global__ __launch_bounds__(256) void k_gather_records(
     const u8* __restrict__ recs, const u32* __restrict__ idx, u64 n,
     u32 rec_words_rt, u8* __restrict__ out) {
@@ -106,12 +106,46 @@ __global__ __launch_bounds__(256) void k_gather_records(
         for (int j = 0; j < 4; ++j) {
             u64 gj = g + (u64)j * stride;
             u64 rec = gj / rec_words;
-            v[j] = rin[(u64)idx[rec] * rec_words +
-                       (u32)(gj - rec * rec_words)];
+            const u32* src =
+                &rin[(u64)idx[rec] * rec_words + (u32)(gj - rec * rec_words)];
+            v[j] = NT ? __builtin_nontemporal_load(src) : *src;
         }
         for (int j = 0; j < 4; ++j) rout[g + (u64)j * stride] = v[j];
     }
     for (; g < total_words; g += stride) {
+        u64 rec = g / rec_words;
+        rout[g] = rin[(u64)idx[rec] * rec_words +
+                      (u32)(g - rec * rec_words)];
+    }
+}
+
+/* block-contiguous variant: each block owns one contiguous output span,
+ * iterated 256 words at a time — the idx window a block touches is a
+ * compact ~SPAN/RW range (L1-resident). */
+template <int RW, bool NT>
+__global__ __launch_bounds__(256) void k_gather_records_span(
+    const u8* __restrict__ recs, const u32* __restrict__ idx, u64 n,
+    u32 rec_words_rt, u8* __restrict__ out) {
+    const u32 rec_words = RW ? (u32)RW : rec_words_rt;
+    const u64 total_words = n * rec_words;
+    const u64 span = (total_words + gridDim.x - 1) / gridDim.x;
+    const u64 s0 = (u64)blockIdx.x * span;
+    const u64 s1 = (s0 + span < total_words) ? s0 + span : total_words;
+    const u32* rin = (const u32*)recs;
+    u32* rout = (u32*)out;
+    u64 g = s0 + threadIdx.x;
+    for (; g + 3 * 256 < s1; g += 4 * 256) {
+        u32 v[4];
+        for (int j = 0; j < 4; ++j) {
+            u64 gj = g + (u64)j * 256;
+            u64 rec = gj / rec_words;
+            const u32* src =
+                &rin[(u64)idx[rec] * rec_words + (u32)(gj - rec * rec_words)];
+            v[j] = NT ? __builtin_nontemporal_load(src) : *src;
+        }
+        for (int j = 0; j < 4; ++j) rout[g + (u64)j * 256] = v[j];
+    }
+    for (; g < s1; g += 256) {
         u64 rec = g / rec_words;
         rout[g] = rin[(u64)idx[rec] * rec_words +
                       (u32)(g - rec * rec_words)];
@@ -193,22 +227,44 @@ int t9_gather_records(t9_context* ctx, const u8* d_recs, const u32* d_idx,
     if (!d_recs || !d_idx || !d_out || rec_size % 4) return T9_EINVAL;
     if (n == 0) return T9_OK;
     const u32 rw = rec_size / 4;
-    const dim3 grid(grid_for(n * rw));
     hipStream_t s = (hipStream_t)stream;
+    const char* gv = getenv("T9_GATHER_VARIANT");
+    const int var = gv ? atoi(gv) : 1;   /* 1 strided, 2 +nt, 3 span,
+                                            4 span+nt */
+    const char* gg = getenv("T9_GATHER_GRID");
+    const u32 gcap = gg ? (u32)atoi(gg) : 4096;
+    u64 want = t9_ceil_div(n * rw, 256);
+    const dim3 grid((u32)((want < gcap) ? (want ? want : 1) : gcap));
     T9_PERF_WRAP(
         s, "gather",
-        if (rw == 25)
-            hipLaunchKernelGGL((k_gather_records<25>), grid, dim3(256), 0,
-                               s, d_recs, d_idx, n, rw, d_out);
+        if (rw == 25 && var == 1)
+            hipLaunchKernelGGL((k_gather_records<25, false>), grid,
+                               dim3(256), 0, s, d_recs, d_idx, n, rw,
+                               d_out);
+        else if (rw == 25 && var == 2)
+            hipLaunchKernelGGL((k_gather_records<25, true>), grid,
+                               dim3(256), 0, s, d_recs, d_idx, n, rw,
+                               d_out);
+        else if (rw == 25 && var == 3)
+            hipLaunchKernelGGL((k_gather_records_span<25, false>), grid,
+                               dim3(256), 0, s, d_recs, d_idx, n, rw,
+                               d_out);
+        else if (rw == 25 && var == 4)
+            hipLaunchKernelGGL((k_gather_records_span<25, true>), grid,
+                               dim3(256), 0, s, d_recs, d_idx, n, rw,
+                               d_out);
         else if (rw == 32)
-            hipLaunchKernelGGL((k_gather_records<32>), grid, dim3(256), 0,
-                               s, d_recs, d_idx, n, rw, d_out);
+            hipLaunchKernelGGL((k_gather_records<32, false>), grid,
+                               dim3(256), 0, s, d_recs, d_idx, n, rw,
+                               d_out);
         else if (rw == 2)
-            hipLaunchKernelGGL((k_gather_records<2>), grid, dim3(256), 0,
-                               s, d_recs, d_idx, n, rw, d_out);
+            hipLaunchKernelGGL((k_gather_records<2, false>), grid,
+                               dim3(256), 0, s, d_recs, d_idx, n, rw,
+                               d_out);
         else
-            hipLaunchKernelGGL((k_gather_records<0>), grid, dim3(256), 0,
-                               s, d_recs, d_idx, n, rw, d_out));
+            hipLaunchKernelGGL((k_gather_records<0, false>), grid,
+                               dim3(256), 0, s, d_recs, d_idx, n, rw,
+                               d_out));
     T9_LAUNCH_CHECK();
     return T9_OK;
 }
