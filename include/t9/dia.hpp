@@ -148,6 +148,14 @@ public:
     //! (INTEGRATION.md "Error behaviour").
     DIA Sort() const { return SortImpl(); }
 
+    //! SortStable (reference api/sort.hpp:900-937): the radix pipeline is
+    //! stable at every level, so Sort already implements the stable
+    //! contract (sort_node_test.cpp:291-404's "lower source index wins").
+    DIA SortStable() const { return SortImpl(); }
+
+    template <typename Compare>
+    DIA SortStable(const Compare& c) const { return Sort(c); }
+
     template <typename Compare>
     DIA Sort(const Compare&) const {
         static_assert(

@@ -258,3 +258,20 @@ def test_end_to_end_single_gpu_sample_sort(nat, oracle):
         nat.sort_u64(G.ptr(dpart), len(part), G.ptr(w), G.stream())
         out.append(G.host(dpart, np.uint64))
     assert np.array_equal(np.concatenate(out), np.sort(keys))
+
+
+def test_alltoall_loopback_world1(nat):
+    # t9_alltoall's world==1 self-exchange path (the RCCL grouped path is
+    # exercised by the driver's multi-GPU run; counts/displs marshalling
+    # is shared)
+    import ctypes
+    n = 10_000
+    send = G.dev(np.arange(n, dtype=np.uint64))
+    recv = G.empty(n, np.uint64)
+    counts = np.array([n], dtype=np.uint64)
+    displs = np.array([0], dtype=np.uint64)
+    cp = counts.ctypes.data_as(ctypes.c_void_p)
+    dp = displs.ctypes.data_as(ctypes.c_void_p)
+    nat.alltoall(G.ptr(send), cp, dp, G.ptr(recv), cp, dp, 8, G.stream())
+    assert np.array_equal(G.host(recv, np.uint64),
+                          np.arange(n, dtype=np.uint64))

@@ -229,10 +229,11 @@ int t9_gather_records(t9_context* ctx, const u8* d_recs, const u32* d_idx,
     const u32 rw = rec_size / 4;
     hipStream_t s = (hipStream_t)stream;
     const char* gv = getenv("T9_GATHER_VARIANT");
-    const int var = gv ? atoi(gv) : 1;   /* 1 strided, 2 +nt, 3 span,
-                                            4 span+nt */
+    const int var = gv ? atoi(gv) : 3;   /* 1 strided, 2 +nt, 3 span,
+                                            4 span+nt; measured best:
+                                            span, grid 16384 */
     const char* gg = getenv("T9_GATHER_GRID");
-    const u32 gcap = gg ? (u32)atoi(gg) : 4096;
+    const u32 gcap = gg ? (u32)atoi(gg) : 16384;
     u64 want = t9_ceil_div(n * rw, 256);
     const dim3 grid((u32)((want < gcap) ? (want ? want : 1) : gcap));
     T9_PERF_WRAP(
