@@ -212,11 +212,15 @@ int t9_extract_key64(t9_context* ctx, const u8* d_recs, u64 n, u32 rec_size,
     if (rec_size % 4 || key_off % 4 || key_off + 8 > rec_size)
         return T9_EINVAL;
     if (n == 0) return T9_OK;
+    const char* eg = getenv("T9_EXTRACT_GRID");
+    u64 ewant = t9_ceil_div(n, 256);
+    const u32 ecap = eg ? (u32)atoi(eg) : 16384;
+    const dim3 egrid((u32)((ewant < ecap) ? (ewant ? ewant : 1) : ecap));
     T9_PERF_WRAP((hipStream_t)stream, "extract",
-                 hipLaunchKernelGGL(k_extract_key64, dim3(grid_for(n)),
-                                    dim3(256), 0, (hipStream_t)stream,
-                                    d_recs, n, rec_size / 4, key_off,
-                                    d_keys, d_idx));
+                 hipLaunchKernelGGL(k_extract_key64, egrid, dim3(256), 0,
+                                    (hipStream_t)stream, d_recs, n,
+                                    rec_size / 4, key_off, d_keys,
+                                    d_idx));
     T9_LAUNCH_CHECK();
     return T9_OK;
 }
