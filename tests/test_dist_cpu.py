@@ -105,3 +105,29 @@ def test_distributed_terasort_protocol(n_total):
     got = np.concatenate([results[r][0] for r in range(WORLD)])
     assert got.shape == expect.shape
     assert np.array_equal(got, expect)
+
+
+def test_distributed_terasort_protocol_world3():
+    """world_size=3 with ragged shard sizes (n % 3 != 0)."""
+    global WORLD
+    seed, n_total = 0x55, 2003
+    ctxm = mp.get_context("spawn")
+    q = ctxm.Queue()
+    procs = [ctxm.Process(target=_worker,
+                          args=(r, 3, 29611, n_total, seed, q))
+             for r in range(3)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(3):
+        rank, data, cnt = q.get(timeout=180)
+        assert cnt >= 0, data
+        results[rank] = np.frombuffer(data, np.uint8).reshape(cnt, 100)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    from tests._oracle import Oracle
+    o = Oracle()
+    expect = o.sort_records(o.gen_records(n_total, seed=seed))
+    got = np.concatenate([results[r] for r in range(3)])
+    assert np.array_equal(got, expect)
