@@ -1,12 +1,14 @@
-/* t9_sort.hip — LSD radix sort + splitter classification kernels, gfx950.
+/* t9_sort.hip — radix sort dispatch, LSD pipeline + splitter
+ * classification kernels, gfx950.
  *
  * MI355X-native replacement for the reference's local sort path
  * (thrill/api/sort.hpp:665-786 SortAndWriteToFile/std::sort + PushData
  * loser-tree merge via core/multiway_merge.hpp:30-116): the whole per-GPU
- * partition is sorted in one LSD radix pipeline (8-bit digits, stable
- * within-tile ranking by wavefront ballot/popcount, LDS-staged reorder so
- * global writes are digit-run coalesced), so the run/merge split of the
- * reference — an artifact of bounded RAM — vanishes in 288 GB HBM3E.
+ * partition is sorted in one radix pipeline, so the run/merge split of
+ * the reference — an artifact of bounded RAM — vanishes in 288 GB HBM3E.
+ * Large n dispatches to the two-level MSB pipeline (t9_sort_msb.hip);
+ * this file holds the 8-pass LSD pipeline (small n + skew fallback), the
+ * shared hist/scan kernels and the classification/partition kernels.
  *
  * Classification (k_classify) replaces TransmitItems' tree-descent +
  * EqualSampleGreaterIndex walk (thrill/api/sort.hpp:434-535, :424-426) by
@@ -15,9 +17,9 @@
  * cross-checked against the literal tree restatement in the oracle tests.
  *
  * Roofline: integer/byte work, HBM-bound; no MFMA (BASELINE.json
- * north_star). Per pass the pipeline moves: hist read 8 B + scatter read
- * 8 B (L2-resident re-read inside the same block) + write 8 B per key,
- * plus ~3% hist/scan traffic. 8 passes for a full u64.
+ * north_star). LSD moves hist 8 B + scatter 24 B per pair per pass over
+ * 8 passes; the MSB pipeline replaces that with 2 passes + an LDS-only
+ * level.
  */
 
 #include "t9_common.h"
