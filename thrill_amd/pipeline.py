@@ -222,41 +222,34 @@ class TeraSort:
             zk = torch.zeros(1, dtype=torch.int64, device="cuda")
             zi = torch.zeros(1, dtype=torch.int64, device="cuda")
             return z, zk, zi
-        S = max(1, sample_size(self.n_total) // self.world)
+        # S is derived identically on every rank with no communication:
+        # capped by the minimum shard size (= base share), so the sample
+        # tensors are equal-sized and plain all_gather works (no object
+        # collectives — they cost milliseconds per step at N=8).
+        base = self.n_total // self.world
+        S = max(1, min(base, sample_size(self.n_total) // self.world))
         stride = max(1, self.n_local // S)
         pos = torch.arange(0, self.n_local, stride, device="cuda")[:S]
-        # gather sampled records (S x 100 bytes) via byte gather kernel:
-        # use t9_gather_records with the sample positions as indices
+        if len(pos) < S:   # only possible for tiny shards
+            pos = torch.cat([pos, pos[-1:].expand(S - len(pos))])
         dpos = pos.to(torch.int32)
-        d_samp = torch.empty(len(pos) * REC, dtype=torch.uint8,
-                             device="cuda")
-        self.nat.gather_records(_ptr(self.d_in), _ptr(dpos), len(pos), REC,
+        d_samp = torch.empty(S * REC, dtype=torch.uint8, device="cuda")
+        self.nat.gather_records(_ptr(self.d_in), _ptr(dpos), S, REC,
                                 _ptr(d_samp), _stream())
         gidx = (pos + self.gidx0).to(torch.int64)
-        # equal S per rank not guaranteed (remainders): pad to max S
-        sizes = [None] * self.world
-        dist.all_gather_object(sizes, int(len(pos)))
-        maxS = max(sizes)
-        pad_samp = torch.zeros(maxS * REC, dtype=torch.uint8, device="cuda")
-        pad_samp[:len(pos) * REC] = d_samp
-        pad_idx = torch.zeros(maxS, dtype=torch.int64, device="cuda")
-        pad_idx[:len(pos)] = gidx
-        gs = [torch.empty_like(pad_samp) for _ in range(self.world)]
-        gi = [torch.empty_like(pad_idx) for _ in range(self.world)]
-        dist.all_gather(gs, pad_samp)
-        dist.all_gather(gi, pad_idx)
+        gs = torch.empty(self.world * S * REC, dtype=torch.uint8,
+                         device="cuda")
+        gi = torch.empty(self.world * S, dtype=torch.int64, device="cuda")
+        dist.all_gather_into_tensor(gs, d_samp)
+        dist.all_gather_into_tensor(gi, gidx)
         p = self.world
         spl_recs_t = torch.empty(max(p - 1, 1) * REC, dtype=torch.uint8,
                                  device="cuda")
         spl_idx_t = torch.empty(max(p - 1, 1), dtype=torch.int64,
                                 device="cuda")
         if self.rank == 0:
-            all_recs = np.concatenate([
-                gs[r][:sizes[r] * REC].cpu().numpy().reshape(sizes[r], REC)
-                for r in range(self.world)])
-            all_idx = np.concatenate([
-                gi[r][:sizes[r]].cpu().numpy().astype(np.uint64)
-                for r in range(self.world)])
+            all_recs = gs.cpu().numpy().reshape(self.world * S, REC)
+            all_idx = gi.cpu().numpy().astype(np.uint64)
             spl_recs, spl_idx = select_splitters(all_recs, all_idx, p)
             spl_recs_t.copy_(torch.from_numpy(
                 spl_recs.reshape(-1).copy()).cuda())
