@@ -160,13 +160,36 @@ def main():
         per_launch_s = scat_ms / 1e3 / scat_n
         algo_bytes = 24.0 * ts.n_local
         achieved = algo_bytes / per_launch_s / 1e9
+        # traffic: PMC bytes per launch from the committed calibration run
+        # (profiles/pmc_traffic_terasort_r01.json — separate rocprofv3
+        # --pmc FETCH_SIZE / WRITE_SIZE passes; FETCH doubled per the
+        # gfx950 half-reporting of wide coalesced reads, absolute values
+        # carry the guide's per-pattern calibration caveat). Scaled to
+        # this run's n. None when the calibration file is absent or the
+        # workload size differs wildly.
+        traffic = None
+        cal_path = os.path.join(REPO, "profiles",
+                                "pmc_traffic_terasort_r01.json")
+        if os.path.exists(cal_path):
+            import json as _json
+            with open(cal_path) as f:
+                cal = _json.load(f)
+            tot = 0.0
+            cnt = 0
+            for k, v in cal.items():
+                if "scatter_wave512" in k or "scatter_seg" in k:
+                    tot += (v["fetch_bytes_x2_per_launch"] +
+                            v["write_bytes_per_launch"])
+                    cnt += 1
+            if cnt:
+                traffic = round(tot / cnt * (ts.n_local / 107_374_182))
         roofline = {
             "bound": "hbm",
             "achieved": round(achieved, 1),
             "peak": PEAK_HBM_GBS,
             "unit": "GB/s",
             "frac": round(achieved / PEAK_HBM_GBS, 4),
-            "traffic": None,
+            "traffic": traffic,
             "kernel": "pair_scatter (radix pass over u64+u32 pairs)",
             "avg_launch_ms": round(scat_ms / scat_n, 3),
         }
