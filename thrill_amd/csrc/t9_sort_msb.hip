@@ -276,6 +276,48 @@ __global__ __launch_bounds__(1024, 4) void k_scatter_seg(
     }
 }
 
+/* Greedy span packing: pack consecutive (b7,b6) sub-buckets into spans
+ * of <= SPANMAX elements. A span may cover several sub-buckets because
+ * their top-16 key bits are distinct and ordered, so sorting the span by
+ * the FULL 64-bit key both preserves the bucket grouping and sorts within
+ * — at the cost of 8 LDS passes instead of 6. One 256-thread block;
+ * thread t packs the 256 sub-buckets of top-byte bucket t (buckets are
+ * independent, so spans never cross a b7 boundary). A sub-bucket larger
+ * than SPANMAX becomes its own (oversize) span, handled by the ranged-LSD
+ * fallback. */
+__global__ __launch_bounds__(256) void k_span_pack(
+    const u32* __restrict__ sub_start, const u32* __restrict__ sub_n,
+    u32 spanmax, u32* __restrict__ span_count,
+    u32* __restrict__ span_start, u32* __restrict__ span_len) {
+    const u32 t = threadIdx.x;
+    u32 cur_start = 0, cur_len = 0;
+    for (u32 j = 0; j < 256; ++j) {
+        const u32 i = t * 256 + j;
+        const u32 ns = sub_n[i];
+        if (ns == 0) continue;
+        const u32 st = sub_start[i];
+        if (cur_len == 0) {
+            cur_start = st;
+            cur_len = ns;
+        }
+        else if (cur_len + ns <= spanmax) {
+            cur_len += ns;
+        }
+        else {
+            const u32 slot = atomicAdd(span_count, 1u);
+            span_start[slot] = cur_start;
+            span_len[slot] = cur_len;
+            cur_start = st;
+            cur_len = ns;
+        }
+    }
+    if (cur_len) {
+        const u32 slot = atomicAdd(span_count, 1u);
+        span_start[slot] = cur_start;
+        span_len[slot] = cur_len;
+    }
+}
+
 /* sub-bucket stats: info[0] = max size (atomicMax), info[1] = count of
  * sub-buckets above hardmax, list = their indices */
 __global__ __launch_bounds__(256) void k_subinfo(
@@ -399,6 +441,122 @@ __global__ __launch_bounds__(BLOCK, 4) void k_lds_sort_sub(
  * host
  * ------------------------------------------------------------------ */
 
+/* span LDS sort: up to SPANMAX pairs, 8 stable ballot-ranked passes over
+ * the FULL 64-bit key (spans cover several top-16 groups). in == out is
+ * allowed (in-place); reads go to LDS before any write-back. */
+template <int SPANMAX, int BLOCK, bool HAS_VAL>
+__global__ __launch_bounds__(BLOCK, 4) void k_lds_sort_span(
+    const u64* __restrict__ keys_in, const u32* __restrict__ vals_in,
+    u64* __restrict__ keys_out, u32* __restrict__ vals_out,
+    const u32* __restrict__ span_start, const u32* __restrict__ span_len) {
+    constexpr int NW = BLOCK / 64;
+    constexpr int SUBQ = SPANMAX / NW;
+    constexpr int GROUPS = SUBQ / 64;
+    __shared__ u64 s_k[2][SPANMAX];
+    __shared__ u32 s_v[2][HAS_VAL ? SPANMAX : 1];
+    __shared__ u16 s_rank[SPANMAX];
+    __shared__ u32 s_wcnt[NW * T9_RADIX];
+    __shared__ u32 s_woff[NW * T9_RADIX];
+    __shared__ u32 s_start[T9_RADIX];
+    __shared__ u32 s_differ;
+
+    const u32 sb = blockIdx.x;
+    const u32 ns = span_len[sb];
+    if (ns == 0) return;
+    const u32 gbase = span_start[sb];
+    const u32 tid = threadIdx.x, wave = tid >> 6, lane = tid & 63;
+
+    if (ns > (u32)SPANMAX) {
+        /* oversize span (single huge sub-bucket): copy through when
+         * out-of-place; the ranged-LSD fallback re-sorts it afterwards */
+        if (keys_out != keys_in)
+            for (u32 i = tid; i < ns; i += BLOCK) {
+                keys_out[gbase + i] = keys_in[gbase + i];
+                if (HAS_VAL) vals_out[gbase + i] = vals_in[gbase + i];
+            }
+        return;
+    }
+
+    if (tid == 0) s_differ = 0;
+    __syncthreads();
+    const u64 k0ref = keys_in[gbase];
+    for (u32 i = tid; i < ns; i += BLOCK) {
+        u64 k = keys_in[gbase + i];
+        s_k[0][i] = k;
+        if (HAS_VAL) s_v[0][i] = vals_in[gbase + i];
+        if (k != k0ref) s_differ = 1;
+    }
+    __syncthreads();
+    if (!s_differ) {
+        if (keys_out != keys_in)
+            for (u32 i = tid; i < ns; i += BLOCK) {
+                keys_out[gbase + i] = s_k[0][i];
+                if (HAS_VAL) vals_out[gbase + i] = s_v[0][i];
+            }
+        return;
+    }
+
+    int cur = 0;
+    for (int pass = 0; pass < 8; ++pass) {
+        const u32 shift = pass * 8;
+        for (u32 t = lane; t < T9_RADIX; t += 64)
+            s_wcnt[wave * T9_RADIX + t] = 0;
+        __syncthreads();
+        const u32 wbase = wave * SUBQ;
+        for (int g = 0; g < GROUPS; ++g) {
+            const u32 i = wbase + g * 64 + lane;
+            const bool valid = i < ns;
+            u32 d = 0;
+            if (valid) d = (u32)(s_k[cur][i] >> shift) & 255u;
+            u64 m = __ballot(valid);
+            for (int bit = 0; bit < 8; ++bit) {
+                u64 bb = __ballot((d >> bit) & 1u);
+                m &= ((d >> bit) & 1u) ? bb : ~bb;
+            }
+            const u32 wr = (u32)__popcll(m & ((1ull << lane) - 1ull));
+            const u32 before = valid ? s_wcnt[wave * T9_RADIX + d] : 0;
+            if (valid) {
+                s_rank[i] = (u16)(before + wr);
+                if (wr == 0)
+                    s_wcnt[wave * T9_RADIX + d] = before + (u32)__popcll(m);
+            }
+        }
+        __syncthreads();
+        if (tid < T9_RADIX) {
+            u32 run = 0;
+            for (int w = 0; w < NW; ++w) {
+                s_woff[w * T9_RADIX + tid] = run;
+                run += s_wcnt[w * T9_RADIX + tid];
+            }
+            s_start[tid] = run;
+        }
+        __syncthreads();
+        t9_scan256_onewave(s_start, tid);
+        __syncthreads();
+        if (tid < T9_RADIX) {
+            const u32 excl = s_start[tid];
+            for (int w = 0; w < NW; ++w) s_woff[w * T9_RADIX + tid] += excl;
+        }
+        __syncthreads();
+        for (int g = 0; g < GROUPS; ++g) {
+            const u32 i = wbase + g * 64 + lane;
+            if (i < ns) {
+                const u64 k = s_k[cur][i];
+                const u32 d = (u32)(k >> shift) & 255u;
+                const u32 pos = s_woff[wave * T9_RADIX + d] + s_rank[i];
+                s_k[cur ^ 1][pos] = k;
+                if (HAS_VAL) s_v[cur ^ 1][pos] = s_v[cur][i];
+            }
+        }
+        __syncthreads();
+        cur ^= 1;
+    }
+    for (u32 i = tid; i < ns; i += BLOCK) {
+        keys_out[gbase + i] = s_k[cur][i];
+        if (HAS_VAL) vals_out[gbase + i] = s_v[cur][i];
+    }
+}
+
 namespace {
 struct MsbWs {
     u64* alt_k;
@@ -411,7 +569,8 @@ struct MsbWs {
     u32* bucket_n;
     u32* sub_start;
     u32* sub_n;
-    u32* ovr;        /* [0] = count, [1..] = list */
+    u32* ovr;        /* [0] = max or count, [1] = count, [2..] = list */
+    u32* span;
     u64 B1, B2max;
 };
 
@@ -444,6 +603,8 @@ MsbWs carve_msb(char* p, u64 n) {
     p += t9_align256((u64)NSUB * 4);
     w.ovr = (u32*)p;
     p += t9_align256((u64)(NSUB + 2) * 4);
+    w.span = (u32*)p;   /* [0]=count, [1..CAP]=start, [1+CAP..]=len */
+    p += t9_align256((u64)(2 * (NSUB + 64) + 1) * 4);
     return w;
 }
 
@@ -455,7 +616,8 @@ u64 msb_ws_bytes(u64 n) {
            t9_align256(t9_ceil_div(B2max, T9_SCAN_CHUNK) * T9_RADIX * 4) +
            4 * t9_align256(T9_RADIX * 4) + t9_align256(257 * 4) +
            2 * t9_align256((u64)NSUB * 4) +
-           t9_align256((u64)(NSUB + 2) * 4);
+           t9_align256((u64)(NSUB + 2) * 4) +
+           t9_align256((u64)(2 * (NSUB + 64) + 1) * 4);
 }
 } // namespace
 
@@ -513,25 +675,63 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
         T9_LAUNCH_CHECK();
     }
 
-    /* ---- level 3: in-LDS sort of each (b7, b6) sub-bucket ---- */
-    HIP_TRY(hipMemsetAsync(w.ovr, 0, 8, s));
-    hipLaunchKernelGGL(k_subinfo, dim3(NSUB / 256), dim3(256), 0, s,
-                       w.sub_n, NSUB, T9_SUBMAX, w.ovr, w.ovr + 2);
-    u32 info[2] = { 0, 0 };
-    HIP_TRY(hipMemcpyAsync(info, w.ovr, 8, hipMemcpyDeviceToHost, s));
-    HIP_TRY(hipStreamSynchronize(s));
-    const u32 maxsub = info[0], novr = info[1];
-    T9_PERF_WRAP(
-        s, "lds_sort",
-        if (maxsub <= 2048)
-            hipLaunchKernelGGL((k_lds_sort_sub<2048, 512, HAS_VAL>),
-                               dim3(NSUB), dim3(512), 0, s, d_keys, d_vals,
-                               w.sub_start, w.sub_n);
-        else
-            hipLaunchKernelGGL((k_lds_sort_sub<T9_SUBMAX, 1024, HAS_VAL>),
-                               dim3(NSUB), dim3(1024), 0, s, d_keys,
-                               d_vals, w.sub_start, w.sub_n));
-    T9_LAUNCH_CHECK();
+    /* ---- level 3: in-LDS sort; spans pack consecutive sub-buckets to
+     * full 4096-element blocks sorted by the complete 64-bit key
+     * (T9_LDS_SPAN=0 falls back to one block per sub-bucket) ---- */
+    const char* se = getenv("T9_LDS_SPAN");
+    const bool use_span = !(se && atoi(se) == 0);
+    const u32* sel_start;
+    const u32* sel_n;
+    u32 novr = 0;
+    if (use_span) {
+        const u32 CAP = NSUB + 64;
+        u32* span_count = w.span;
+        u32* span_start = w.span + 1;
+        u32* span_len = w.span + 1 + CAP;
+        HIP_TRY(hipMemsetAsync(w.span, 0, (u64)(2 * CAP + 1) * 4, s));
+        hipLaunchKernelGGL(k_span_pack, dim3(1), dim3(256), 0, s,
+                           w.sub_start, w.sub_n, T9_SUBMAX, span_count,
+                           span_start, span_len);
+        HIP_TRY(hipMemsetAsync(w.ovr, 0, 8, s));
+        hipLaunchKernelGGL(k_subinfo, dim3(CAP / 256 + 1), dim3(256), 0, s,
+                           span_len, CAP, T9_SUBMAX, w.ovr, w.ovr + 2);
+        u32 info[2] = { 0, 0 };
+        HIP_TRY(hipMemcpyAsync(info, w.ovr, 8, hipMemcpyDeviceToHost, s));
+        HIP_TRY(hipStreamSynchronize(s));
+        novr = info[1];
+        T9_PERF_WRAP(
+            s, "lds_sort",
+            hipLaunchKernelGGL((k_lds_sort_span<T9_SUBMAX, 1024, HAS_VAL>),
+                               dim3(CAP), dim3(1024), 0, s, d_keys, d_vals,
+                               d_keys, d_vals, span_start, span_len));
+        T9_LAUNCH_CHECK();
+        sel_start = span_start;
+        sel_n = span_len;
+    }
+    else {
+        HIP_TRY(hipMemsetAsync(w.ovr, 0, 8, s));
+        hipLaunchKernelGGL(k_subinfo, dim3(NSUB / 256), dim3(256), 0, s,
+                           w.sub_n, NSUB, T9_SUBMAX, w.ovr, w.ovr + 2);
+        u32 info[2] = { 0, 0 };
+        HIP_TRY(hipMemcpyAsync(info, w.ovr, 8, hipMemcpyDeviceToHost, s));
+        HIP_TRY(hipStreamSynchronize(s));
+        const u32 maxsub = info[0];
+        novr = info[1];
+        T9_PERF_WRAP(
+            s, "lds_sort",
+            if (maxsub <= 2048)
+                hipLaunchKernelGGL((k_lds_sort_sub<2048, 512, HAS_VAL>),
+                                   dim3(NSUB), dim3(512), 0, s, d_keys,
+                                   d_vals, w.sub_start, w.sub_n);
+            else
+                hipLaunchKernelGGL(
+                    (k_lds_sort_sub<T9_SUBMAX, 1024, HAS_VAL>), dim3(NSUB),
+                    dim3(1024), 0, s, d_keys, d_vals, w.sub_start,
+                    w.sub_n));
+        T9_LAUNCH_CHECK();
+        sel_start = w.sub_start;
+        sel_n = w.sub_n;
+    }
     if (novr == 0) return T9_OK;
 
     if (novr > 64) {
@@ -550,9 +750,9 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
                       hipMemcpyDeviceToHost));
     std::vector<u32> starts(novr), counts(novr);
     for (u32 i = 0; i < novr; ++i) {
-        HIP_TRY(hipMemcpy(&starts[i], w.sub_start + list[i], 4,
+        HIP_TRY(hipMemcpy(&starts[i], sel_start + list[i], 4,
                           hipMemcpyDeviceToHost));
-        HIP_TRY(hipMemcpy(&counts[i], w.sub_n + list[i], 4,
+        HIP_TRY(hipMemcpy(&counts[i], sel_n + list[i], 4,
                           hipMemcpyDeviceToHost));
     }
     for (u32 i = 0; i < novr; ++i) {
