@@ -350,10 +350,7 @@ __global__ __launch_bounds__(BLOCK, 4) void k_lds_sort_sub(
     __shared__ u64 s_k[2][SUBMAX];
     __shared__ u32 s_v[2][HAS_VAL ? SUBMAX : 1];
     __shared__ u16 s_rank[SUBMAX];
-    /* per-(wave, group, digit) counts as u8 (<= 64 per group; within-wave
-     * group prefixes <= (GROUPS-1)*64 < 256): the groups' ballot rounds
-     * are fully independent — no per-group LDS read-modify-write chain */
-    __shared__ u8 s_gcnt[NW][GROUPS][T9_RADIX];
+    __shared__ u32 s_wcnt[NW * T9_RADIX];
     __shared__ u32 s_woff[NW * T9_RADIX];
     __shared__ u32 s_start[T9_RADIX];
     __shared__ u32 s_differ;
@@ -381,10 +378,9 @@ __global__ __launch_bounds__(BLOCK, 4) void k_lds_sort_sub(
     int cur = 0;
     for (int pass = 0; pass < 6; ++pass) {
         const u32 shift = pass * 8;
-        for (u32 t = lane; t < GROUPS * T9_RADIX / 4; t += 64)
-            ((u32*)s_gcnt[wave])[t] = 0;
-        /* no barrier: each wave touches only its own s_gcnt rows until
-         * the combine barrier below */
+        for (u32 t = lane; t < T9_RADIX; t += 64)
+            s_wcnt[wave * T9_RADIX + t] = 0;
+        __syncthreads();
         const u32 wbase = wave * SUBQ;
         for (int g = 0; g < GROUPS; ++g) {
             const u32 i = wbase + g * 64 + lane;
@@ -397,26 +393,20 @@ __global__ __launch_bounds__(BLOCK, 4) void k_lds_sort_sub(
                 m &= ((d >> bit) & 1u) ? bb : ~bb;
             }
             const u32 wr = (u32)__popcll(m & ((1ull << lane) - 1ull));
+            const u32 before = valid ? s_wcnt[wave * T9_RADIX + d] : 0;
             if (valid) {
-                s_rank[i] = (u16)wr;   /* within-group rank only */
+                s_rank[i] = (u16)(before + wr);
                 if (wr == 0)
-                    s_gcnt[wave][g][d] = (u8)__popcll(m);
+                    s_wcnt[wave * T9_RADIX + d] =
+                        before + (u32)__popcll(m);
             }
         }
         __syncthreads();
         if (tid < T9_RADIX) {
-            /* per-digit combine: wave bases into s_woff, within-wave
-             * group prefixes overwrite s_gcnt (each <= 192, fits u8) */
             u32 run = 0;
             for (int w = 0; w < NW; ++w) {
-                s_woff[w * T9_RADIX + tid] = run;
-                u32 wrun = 0;
-                for (int g = 0; g < GROUPS; ++g) {
-                    const u8 c = s_gcnt[w][g][tid];
-                    s_gcnt[w][g][tid] = (u8)wrun;
-                    wrun += c;
-                }
-                run += wrun;
+                s_woff[w * T9_RADIX + tid] = run;   /* wave-relative for now */
+                run += s_wcnt[w * T9_RADIX + tid];
             }
             s_start[tid] = run;
         }
@@ -433,8 +423,7 @@ __global__ __launch_bounds__(BLOCK, 4) void k_lds_sort_sub(
             if (i < ns) {
                 const u64 k = s_k[cur][i];
                 const u32 d = (u32)(k >> shift) & 255u;
-                const u32 pos = s_woff[wave * T9_RADIX + d] +
-                                s_gcnt[wave][g][d] + s_rank[i];
+                const u32 pos = s_woff[wave * T9_RADIX + d] + s_rank[i];
                 s_k[cur ^ 1][pos] = k;
                 if (HAS_VAL) s_v[cur ^ 1][pos] = s_v[cur][i];
             }
