@@ -188,3 +188,19 @@ def test_index_bucket_mapping(nat):
     got = G.host(db, np.uint32)
     expect = ((keys - begin) * p // size).astype(np.uint32)
     assert np.array_equal(got, expect)
+
+
+def test_reduce_overflow_sets_error(nat):
+    # more distinct keys than the table can hold: d_error must be set and
+    # the kernel must terminate (bounded probe loop), not hang.
+    n, cap = 10_000, 1024
+    keys = np.arange(1, n + 1, dtype=np.uint64)
+    vals = np.ones(n, dtype=np.uint64)
+    dk, dv = G.dev(keys), G.dev(vals)
+    tk, tv = G.empty(cap + 1, np.uint64), G.empty(cap + 1, np.uint64)
+    derr = G.empty(1, np.uint32)
+    s = G.stream()
+    nat.reduce_init(G.ptr(tk), G.ptr(tv), cap, s)
+    nat.reduce_build(G.ptr(dk), G.ptr(dv), n, G.ptr(tk), G.ptr(tv), cap,
+                     0, G.ptr(derr), s)
+    assert int(G.host(derr, np.uint32)[0]) == 1

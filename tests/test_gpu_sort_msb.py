@@ -155,3 +155,20 @@ def test_msb_records_128B_fused(nat):
     got = G.host(dout, np.uint8).reshape(n, 128)
     order = np.lexsort(tuple(recs[:, c] for c in range(127, -1, -1)))
     assert np.array_equal(got, recs[order])
+
+
+@pytest.mark.parametrize("n", [(1 << 22) - 1, (1 << 22), (1 << 22) + 1])
+def test_dispatch_boundary_sizes(nat, n):
+    # the LSD/MSB dispatch threshold (2^22) must be seamless
+    del os.environ["T9_SORT_ALGO"]   # default dispatch
+    try:
+        dk = G.empty(n, np.uint64)
+        nat.gen_u64(G.ptr(dk), 0, n, n, G.stream())
+        insum = int(dk.sum().item())
+        w = G.ws(nat.ws("sort_u64", n))
+        nat.sort_u64(G.ptr(dk), n, G.ptr(w), G.stream())
+        assert int(dk.sum().item()) == insum
+        signed = dk ^ (-2 ** 63)
+        assert bool((signed[1:] >= signed[:-1]).all().item())
+    finally:
+        os.environ["T9_SORT_ALGO"] = "msb"
