@@ -21,6 +21,7 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdint>
+#include <cstdio>
 #include <cstring>
 #include <functional>
 #include <memory>
@@ -180,6 +181,28 @@ public:
         return FromVector(*ctx_, out);
     }
 
+    //! WriteBinary — reference api/write_binary.hpp: the items as packed
+    //! binary (the POD raw-copy wire format, data/serialization.hpp:35-48),
+    //! one file per worker: pathbase + zero-padded rank. The on-disk bytes
+    //! are bit-compatible with the reference's files (terasort.cpp:184-200
+    //! file mode).
+    void WriteBinary(const std::string& pathbase) const {
+        char name[512];
+        std::snprintf(name, sizeof(name), "%s%010zu", pathbase.c_str(),
+                      ctx_->my_rank());
+        auto host = AllGather();
+        std::FILE* f = std::fopen(name, "wb");
+        if (!f) throw std::runtime_error("WriteBinary: cannot open " +
+                                         std::string(name));
+        if (!host.empty() &&
+            std::fwrite(host.data(), sizeof(ValueType), host.size(), f) !=
+                host.size()) {
+            std::fclose(f);
+            throw std::runtime_error("WriteBinary: short write");
+        }
+        std::fclose(f);
+    }
+
     Context& context() const { return *ctx_; }
     void* device_ptr() const { return buf_ ? buf_->ptr : nullptr; }
 
@@ -304,6 +327,92 @@ inline DIA<KeyValue> ReducePair(const DIA<KeyValue>& input,
     std::vector<KV> out(m);
     for (size_t i = 0; i < m; ++i) out[i] = KV{ rk[i], rv[i] };
     return FromVector(ctx, out);
+}
+
+//! ReadBinary — reference api/read_binary.hpp: read packed fixed-size
+//! items from files (the worker's share; world=1 here reads all).
+template <typename T>
+DIA<T> ReadBinary(Context& ctx, const std::vector<std::string>& files) {
+    std::vector<T> items;
+    for (const auto& path : files) {
+        std::FILE* f = std::fopen(path.c_str(), "rb");
+        if (!f) throw std::runtime_error("ReadBinary: cannot open " + path);
+        std::fseek(f, 0, SEEK_END);
+        long bytes = std::ftell(f);
+        std::fseek(f, 0, SEEK_SET);
+        if (bytes % (long)sizeof(T))
+            throw std::runtime_error("ReadBinary: size not a multiple of "
+                                     "the item size: " + path);
+        size_t n = (size_t)bytes / sizeof(T);
+        size_t old = items.size();
+        items.resize(old + n);
+        if (n && std::fread(items.data() + old, sizeof(T), n, f) != n) {
+            std::fclose(f);
+            throw std::runtime_error("ReadBinary: short read");
+        }
+        std::fclose(f);
+    }
+    return FromVector(ctx, items);
+}
+
+//! GroupByKey — reference api/group_by_key.hpp (sort-based): sort the
+//! pairs by key on the GPU, build the device group index
+//! (t9_group_index), then apply the user's group function on the host per
+//! group (user lambdas are host code in this surface). fn(key, vals_begin,
+//! vals_end) -> Result.
+template <typename Result, typename GroupFn>
+std::vector<Result> GroupByKey(const DIA<KeyValue>& input,
+                               const GroupFn& fn) {
+    Context& ctx = input.context();
+    auto host = input.AllGather();
+    const size_t n = host.size();
+    std::vector<uint64_t> hk(n), hv(n);
+    for (size_t i = 0; i < n; ++i) {
+        hk[i] = host[i].key;
+        hv[i] = host[i].value;
+    }
+    std::vector<Result> out;
+    if (n == 0) return out;
+    DeviceBuf dk(n * 8), dv(n * 8), didx(n * 4), dsv(n * 8);
+    DeviceBuf du(n * 8), doff(n * 8), dcnt(8);
+    T9_DIA_HIP(hipMemcpy(dk.ptr, hk.data(), n * 8, hipMemcpyHostToDevice));
+    T9_DIA_HIP(hipMemcpy(dv.ptr, hv.data(), n * 8, hipMemcpyHostToDevice));
+    hipStream_t s = ctx.stream();
+    {
+        /* iota payload, sort (key, idx), gather values by idx */
+        std::vector<uint32_t> iota(n);
+        for (size_t i = 0; i < n; ++i) iota[i] = (uint32_t)i;
+        T9_DIA_HIP(hipMemcpy(didx.ptr, iota.data(), n * 4,
+                             hipMemcpyHostToDevice));
+        DeviceBuf ws(t9_sort_pairs_workspace(n));
+        T9_DIA_TRY(t9_sort_pairs_u64_u32(ctx.native(), (uint64_t*)dk.ptr,
+                                         (uint32_t*)didx.ptr, n, ws.ptr,
+                                         s));
+        T9_DIA_TRY(t9_gather_records(ctx.native(), (const uint8_t*)dv.ptr,
+                                     (const uint32_t*)didx.ptr, n, 8,
+                                     (uint8_t*)dsv.ptr, s));
+        DeviceBuf gws(t9_group_index_workspace(n));
+        T9_DIA_TRY(t9_group_index(ctx.native(), (const uint64_t*)dk.ptr, n,
+                                  (uint64_t*)du.ptr, (uint64_t*)doff.ptr,
+                                  (uint64_t*)dcnt.ptr, gws.ptr, s));
+        T9_DIA_HIP(hipStreamSynchronize(s));
+    }
+    uint64_t groups = 0;
+    T9_DIA_HIP(hipMemcpy(&groups, dcnt.ptr, 8, hipMemcpyDeviceToHost));
+    std::vector<uint64_t> uk(groups), off(groups), sv(n);
+    T9_DIA_HIP(hipMemcpy(uk.data(), du.ptr, groups * 8,
+                         hipMemcpyDeviceToHost));
+    T9_DIA_HIP(hipMemcpy(off.data(), doff.ptr, groups * 8,
+                         hipMemcpyDeviceToHost));
+    T9_DIA_HIP(hipMemcpy(sv.data(), dsv.ptr, n * 8,
+                         hipMemcpyDeviceToHost));
+    out.reserve(groups);
+    for (uint64_t g = 0; g < groups; ++g) {
+        const uint64_t b = off[g];
+        const uint64_t e = (g + 1 < groups) ? off[g + 1] : n;
+        out.push_back(fn(uk[g], sv.data() + b, sv.data() + e));
+    }
+    return out;
 }
 
 //! Run — reference api/context.cpp:947: construct the context(s) and run

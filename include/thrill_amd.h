@@ -197,6 +197,15 @@ int t9_index_bucket(t9_context* ctx, const uint64_t* d_keys, uint64_t n,
                     uint64_t begin, uint64_t size, uint32_t p,
                     uint32_t* d_bucket, uint64_t* d_counts, void* stream);
 
+/* GroupByKey support (SURVEY.md §8f item 3 — thrill/api/group_by_key.hpp
+ * is sort-based): the group index of a key-sorted array. d_unique[g] /
+ * d_offsets[g] (ascending run starts) for g in [0, *d_count); d_count is
+ * a device u64. Output capacity n always suffices. */
+uint64_t t9_group_index_workspace(uint64_t n);
+int t9_group_index(t9_context* ctx, const uint64_t* d_sorted_keys,
+                   uint64_t n, uint64_t* d_unique, uint64_t* d_offsets,
+                   uint64_t* d_count, void* d_workspace, void* stream);
+
 /* Zipf(s, q, N) token sampling by inverse CDF (bit-identical to the
  * oracle's t9o_zipf_tokens given the same d_cdf table — the CDF itself is
  * computed once by the oracle/host and copied to the device). Restates

@@ -127,12 +127,61 @@ static void test_word_count(api::Context& ctx) {
     CHECK(got == expect);
 }
 
+/* WriteBinary/ReadBinary round trip (terasort.cpp:184-200 file mode;
+ * on-disk bytes = packed records, data/serialization.hpp:35-48) */
+static void test_binary_io(api::Context& ctx) {
+    const size_t n = 10000;
+    std::mt19937_64 rng(5);
+    std::vector<Record> input(n);
+    for (auto& r : input)
+        for (size_t j = 0; j < sizeof(Record); ++j)
+            ((uint8_t*)&r)[j] = (uint8_t)rng();
+    auto dia = api::FromVector(ctx, input);
+    dia.Sort(api::LexicographicLess<Record>())
+        .WriteBinary("/tmp/t9_dia_io-");
+    auto back = api::ReadBinary<Record>(
+        ctx, { "/tmp/t9_dia_io-0000000000" });
+    CHECK(back.Size() == n);
+    auto out = back.AllGather();
+    std::vector<Record> expect = input;
+    std::sort(expect.begin(), expect.end());
+    bool ok = true;
+    for (size_t i = 0; i < n; ++i) ok &= out[i] == expect[i];
+    CHECK(ok);
+    std::remove("/tmp/t9_dia_io-0000000000");
+}
+
+/* GroupByKey (api/group_by_key.hpp): per-key value collections */
+static void test_group_by_key(api::Context& ctx) {
+    const size_t n = 50000;
+    std::vector<api::KeyValue> pairs(n);
+    for (size_t i = 0; i < n; ++i)
+        pairs[i] = api::KeyValue{ i % 97, i };
+    auto sums = api::GroupByKey<std::pair<uint64_t, uint64_t> >(
+        api::FromVector(ctx, pairs),
+        [](uint64_t key, const uint64_t* b, const uint64_t* e) {
+            uint64_t s = 0;
+            for (const uint64_t* p = b; p != e; ++p) s += *p;
+            return std::make_pair(key, s);
+        });
+    CHECK(sums.size() == 97);
+    bool ok = true;
+    for (auto& kv : sums) {
+        uint64_t expect = 0;
+        for (size_t i = kv.first; i < n; i += 97) expect += i;
+        ok &= kv.second == expect;
+    }
+    CHECK(ok);
+}
+
 int main() {
     return api::Run([](api::Context& ctx) {
         test_sort_known_integers(ctx);
         test_sort_degenerate(ctx);
         test_sort_records(ctx);
         test_word_count(ctx);
+        test_binary_io(ctx);
+        test_group_by_key(ctx);
         if (failures == 0)
             std::printf("dia_test: all checks passed\n");
         else

@@ -204,3 +204,21 @@ def test_reduce_overflow_sets_error(nat):
     nat.reduce_build(G.ptr(dk), G.ptr(dv), n, G.ptr(tk), G.ptr(tv), cap,
                      0, G.ptr(derr), s)
     assert int(G.host(derr, np.uint32)[0]) == 1
+
+
+def test_group_index_parity(nat):
+    # group index over sorted keys vs numpy unique run starts
+    rng = np.random.default_rng(23)
+    keys = np.sort(rng.integers(0, 5000, 1 << 20).astype(np.uint64))
+    dk = G.dev(keys)
+    du = G.empty(len(keys), np.uint64)
+    do = G.empty(len(keys), np.uint64)
+    dc = G.empty(1, np.uint64)
+    w = G.ws(nat.ws("group_index", len(keys)))
+    nat.group_index(G.ptr(dk), len(keys), G.ptr(du), G.ptr(do), G.ptr(dc),
+                    G.ptr(w), G.stream())
+    m = int(G.host(dc, np.uint64)[0])
+    uk, idx = np.unique(keys, return_index=True)
+    assert m == len(uk)
+    assert np.array_equal(G.host(du, np.uint64)[:m], uk)
+    assert np.array_equal(G.host(do, np.uint64)[:m], idx.astype(np.uint64))
