@@ -415,6 +415,18 @@ std::vector<Result> GroupByKey(const DIA<KeyValue>& input,
     return out;
 }
 
+//! Merge — reference api/merge.hpp: merge sorted DIAs (u64, default
+//! order); equal keys keep source order (a's before b's).
+inline DIA<uint64_t> Merge(const DIA<uint64_t>& a, const DIA<uint64_t>& b) {
+    Context& ctx = a.context();
+    auto out = std::make_shared<DeviceBuf>((a.Size() + b.Size()) * 8);
+    T9_DIA_TRY(t9_merge_u64(ctx.native(), (const uint64_t*)a.device_ptr(),
+                            a.Size(), (const uint64_t*)b.device_ptr(),
+                            b.Size(), (uint64_t*)out->ptr, ctx.stream()));
+    T9_DIA_HIP(hipStreamSynchronize(ctx.stream()));
+    return DIA<uint64_t>(&ctx, out, a.Size() + b.Size());
+}
+
 //! Run — reference api/context.cpp:947: construct the context(s) and run
 //! the job. Round 1: one process, one GPU, rank 0.
 inline int Run(const std::function<void(Context&)>& job) {

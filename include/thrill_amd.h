@@ -206,6 +206,13 @@ int t9_group_index(t9_context* ctx, const uint64_t* d_sorted_keys,
                    uint64_t n, uint64_t* d_unique, uint64_t* d_offsets,
                    uint64_t* d_count, void* d_workspace, void* stream);
 
+/* Merge (SURVEY.md §8f item 4 — thrill/api/merge.hpp merges pre-sorted
+ * DIAs): merge two sorted u64 sequences into d_out (na+nb); equal keys
+ * from d_a precede those from d_b. One merge-path pass. */
+int t9_merge_u64(t9_context* ctx, const uint64_t* d_a, uint64_t na,
+                 const uint64_t* d_b, uint64_t nb, uint64_t* d_out,
+                 void* stream);
+
 /* Zipf(s, q, N) token sampling by inverse CDF (bit-identical to the
  * oracle's t9o_zipf_tokens given the same d_cdf table — the CDF itself is
  * computed once by the oracle/host and copied to the device). Restates

@@ -275,3 +275,29 @@ def test_alltoall_loopback_world1(nat):
     nat.alltoall(G.ptr(send), cp, dp, G.ptr(recv), cp, dp, 8, G.stream())
     assert np.array_equal(G.host(recv, np.uint64),
                           np.arange(n, dtype=np.uint64))
+
+
+@pytest.mark.parametrize("na,nb", [(0, 100), (100, 0), (1000, 1),
+                                   (100_000, 250_001), (1 << 20, 1 << 20)])
+def test_merge_u64_parity(nat, oracle, na, nb):
+    a = np.sort(oracle.gen_u64(na, seed=na + 1)) if na \
+        else np.empty(0, np.uint64)
+    b = np.sort(oracle.gen_u64(nb, seed=nb + 2)) if nb \
+        else np.empty(0, np.uint64)
+    da = G.dev(a) if na else G.empty(0, np.uint64)
+    db = G.dev(b) if nb else G.empty(0, np.uint64)
+    dout = G.empty(na + nb, np.uint64)
+    nat.merge_u64(G.ptr(da), na, G.ptr(db), nb, G.ptr(dout), G.stream())
+    got = G.host(dout, np.uint64)
+    assert np.array_equal(got, np.sort(np.concatenate([a, b]),
+                                       kind="stable"))
+
+
+def test_merge_u64_tie_source_order(nat):
+    # equal keys: all of a's precede b's — distinguish via known layout
+    a = np.array([5, 5, 7], dtype=np.uint64)
+    b = np.array([5, 6, 7, 7], dtype=np.uint64)
+    da, db = G.dev(a), G.dev(b)
+    dout = G.empty(7, np.uint64)
+    nat.merge_u64(G.ptr(da), 3, G.ptr(db), 4, G.ptr(dout), G.stream())
+    assert G.host(dout, np.uint64).tolist() == [5, 5, 5, 6, 7, 7, 7]

@@ -61,6 +61,7 @@ _SIGS = {
     "t9_reduce_build": (i32, [vp, vp, vp, u64, vp, vp, u64, u64, vp, vp]),
     "t9_reduce_drain": (i32, [vp, vp, vp, u64, vp, vp, vp, vp]),
     "t9_zipf_tokens": (i32, [vp, vp, vp, u64, u64, u64, u64, vp]),
+    "t9_merge_u64": (i32, [vp, vp, u64, vp, u64, vp, vp]),
     "t9_group_index_workspace": (u64, [u64]),
     "t9_group_index": (i32, [vp, vp, u64, vp, vp, vp, vp, vp]),
     "t9_perf_enable": (i32, [i32]),
