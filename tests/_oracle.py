@@ -34,6 +34,8 @@ class Oracle:
         lib.t9o_gen_records.argtypes = [u8p, u64, u64, u64]
         lib.t9o_sort_u64.argtypes = [u64p, u64]
         lib.t9o_sort_records.argtypes = [u8p, u64, u32]
+        lib.t9o_sort_records_parallel.restype = ctypes.c_int
+        lib.t9o_sort_records_parallel.argtypes = [u8p, u64, u32]
         lib.t9o_select_splitters_u64.argtypes = [
             u64p, u64p, u64, u32, u64p, u64p]
         lib.t9o_classify_u64.argtypes = [u64p, u64, u64, u64p, u64p, u32, u32p]
@@ -75,6 +77,13 @@ class Oracle:
         n, rec_size = recs.shape
         self._lib.t9o_sort_records(recs.reshape(-1), n, rec_size)
         return recs.reshape(n, rec_size)
+
+    def sort_records_parallel(self, recs):
+        recs = np.ascontiguousarray(recs, dtype=np.uint8).copy()
+        n, rec_size = recs.shape
+        threads = self._lib.t9o_sort_records_parallel(
+            recs.reshape(-1), n, rec_size)
+        return recs.reshape(n, rec_size), threads
 
     def select_splitters_u64(self, sample_keys, sample_idx, p):
         sample_keys = np.ascontiguousarray(sample_keys, dtype=np.uint64)

@@ -264,3 +264,10 @@ def test_reduce_by_index_oracle(oracle):
     expect = np.zeros(60, dtype=np.uint64)
     np.add.at(expect, (keys - 100).astype(np.int64), vals)
     assert np.array_equal(dense, expect)
+
+
+def test_sort_records_parallel_matches_serial(oracle):
+    recs = oracle.gen_records(30_000, seed=8)
+    par, threads = oracle.sort_records_parallel(recs)
+    assert threads >= 1
+    assert np.array_equal(par, oracle.sort_records(recs))
