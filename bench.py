@@ -88,7 +88,7 @@ def cpu_baseline_leg(seed):
     (~0.8 GiB), about 10-30 s of CPU work."""
     from tests._oracle import Oracle
     o = Oracle()
-    m = 8_000_000
+    m = 30_000_000
     recs = o.gen_records(m, seed=seed)
     t0 = time.perf_counter()
     _, cores = o.sort_records_parallel(recs)

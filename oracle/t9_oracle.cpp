@@ -50,6 +50,11 @@
 #include <cstring>
 #include <vector>
 
+#ifdef _OPENMP
+#include <omp.h>
+#include <parallel/algorithm>
+#endif
+
 extern "C" {
 
 /* ------------------------------------------------------------------ */
@@ -167,10 +172,6 @@ void t9o_sort_records(uint8_t* recs, uint64_t n, uint32_t rec_size) {
     std::memcpy(recs, tmp.data(), (size_t)n * rec_size);
 }
 
-#ifdef _OPENMP
-#include <omp.h>
-#endif
-
 int t9o_sort_records_parallel(uint8_t* recs, uint64_t n,
                               uint32_t rec_size) {
     int threads = 1;
@@ -183,36 +184,11 @@ int t9o_sort_records_parallel(uint8_t* recs, uint64_t n,
     };
     std::vector<uint64_t> idx(n);
     for (uint64_t i = 0; i < n; ++i) idx[i] = i;
-    const uint64_t nchunks = threads;
-    std::vector<uint64_t> bounds(nchunks + 1);
-    for (uint64_t c = 0; c <= nchunks; ++c) bounds[c] = n * c / nchunks;
 #ifdef _OPENMP
-#pragma omp parallel for schedule(static, 1)
+    __gnu_parallel::sort(idx.begin(), idx.end(), cmp);
+#else
+    std::sort(idx.begin(), idx.end(), cmp);
 #endif
-    for (uint64_t c = 0; c < nchunks; ++c)
-        std::sort(idx.begin() + bounds[c], idx.begin() + bounds[c + 1],
-                  cmp);
-    /* pairwise merge rounds */
-    std::vector<uint64_t> tmp(n);
-    std::vector<uint64_t> b = bounds;
-    while (b.size() > 2) {
-        std::vector<uint64_t> nb;
-        nb.push_back(0);
-        const uint64_t pairs = (b.size() - 1) / 2;
-#ifdef _OPENMP
-#pragma omp parallel for schedule(static, 1)
-#endif
-        for (uint64_t p2 = 0; p2 < pairs; ++p2) {
-            uint64_t lo = b[2 * p2], mid = b[2 * p2 + 1], hi = b[2 * p2 + 2];
-            std::merge(idx.begin() + lo, idx.begin() + mid,
-                       idx.begin() + mid, idx.begin() + hi,
-                       tmp.begin() + lo, cmp);
-            std::copy(tmp.begin() + lo, tmp.begin() + hi, idx.begin() + lo);
-        }
-        for (uint64_t p2 = 0; p2 < pairs; ++p2) nb.push_back(b[2 * p2 + 2]);
-        if ((b.size() - 1) % 2) nb.push_back(b.back());
-        b.swap(nb);
-    }
     std::vector<uint8_t> out((size_t)n * rec_size);
 #ifdef _OPENMP
 #pragma omp parallel for schedule(static)
