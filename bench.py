@@ -82,13 +82,13 @@ def validate(ts, d_out, n_out):
     return keysum, mono, first, last
 
 
-def cpu_baseline_leg(seed):
-    """Oracle (port of the reference CPU path) on a bounded sample,
-    OpenMP across the box's host cores: ~8M records
-    (~0.8 GiB), about 10-30 s of CPU work."""
+def cpu_baseline_leg(seed, n_total):
+    """Oracle (port of the reference CPU path: index sort under the
+    full-record comparator + permute) on the FULL 10 GiB workload, OpenMP
+    across the box's host cores."""
     from tests._oracle import Oracle
     o = Oracle()
-    m = 30_000_000
+    m = min(n_total, N_RECORDS_10GIB)
     recs = o.gen_records(m, seed=seed)
     t0 = time.perf_counter()
     _, cores = o.sort_records_parallel(recs)
@@ -98,9 +98,9 @@ def cpu_baseline_leg(seed):
         "unit": "keys/s",
         "cores": cores,
         "kind": "port",
-        "sample": f"{m} records ({m * REC / GIB:.2f} GiB) of the same "
-                  f"seeded workload, oracle chunk-sort + pairwise merges "
-                  f"(std::sort, full-record comparator), OpenMP over "
+        "sample": f"{m} records ({m * REC / GIB:.2f} GiB) = the whole "
+                  f"workload, oracle __gnu_parallel::sort index sort "
+                  f"(full-record comparator) + permute, OpenMP over "
                   f"{cores} threads; {dt:.1f}s",
     }
 
@@ -237,7 +237,7 @@ def main():
     cpu_baseline = None
     if rank == 0 and world == 1 and not args.no_cpu_baseline:
         log("timing CPU baseline (oracle, bounded sample)...")
-        cpu_baseline = cpu_baseline_leg(args.seed)
+        cpu_baseline = cpu_baseline_leg(args.seed, n_total)
 
     if rank == 0:
         result = {
