@@ -172,3 +172,48 @@ def test_dispatch_boundary_sizes(nat, n):
         assert bool((signed[1:] >= signed[:-1]).all().item())
     finally:
         os.environ["T9_SORT_ALGO"] = "msb"
+
+
+@pytest.mark.parametrize("case", ["uniform", "few_values", "all_equal",
+                                  "low_entropy_top"])
+def test_three_level_forced_parity(nat, oracle, case):
+    """3-level MSB (T9_MSB_LEVELS=3) at test size, vs stable argsort."""
+    os.environ["T9_MSB_LEVELS"] = "3"
+    try:
+        n = 1 << 18
+        rng = np.random.default_rng(len(case))
+        if case == "uniform":
+            keys = oracle.gen_u64(n, seed=2)
+        elif case == "few_values":
+            keys = rng.integers(0, 6, n).astype(np.uint64) * np.uint64(2**60)
+        elif case == "all_equal":
+            keys = np.full(n, 0x0102030405060708, dtype=np.uint64)
+        else:  # varies only below the top 24 bits
+            keys = rng.integers(0, 1 << 40, n).astype(np.uint64)
+        vals = np.arange(n, dtype=np.uint32)
+        dk, dv = G.dev(keys), G.dev(vals)
+        w = G.ws(nat.ws("sort_pairs", n))
+        nat.sort_pairs_u64_u32(G.ptr(dk), G.ptr(dv), n, G.ptr(w),
+                               G.stream())
+        gk, gv = G.host(dk, np.uint64), G.host(dv, np.uint32)
+        assert np.array_equal(gk, np.sort(keys))
+        order = np.argsort(keys, kind="stable").astype(np.uint32)
+        assert np.array_equal(gv, vals[order.astype(np.int64)])
+    finally:
+        del os.environ["T9_MSB_LEVELS"]
+
+
+def test_three_level_forced_records(nat, oracle):
+    os.environ["T9_MSB_LEVELS"] = "3"
+    try:
+        n = 200_000
+        recs = oracle.gen_records(n, seed=55)
+        din = G.dev(recs.reshape(-1))
+        dout = G.empty(n * 100, np.uint8)
+        w = G.ws(nat.ws("sort_records", n, 100))
+        nat.sort_records(G.ptr(din), G.ptr(dout), n, 100, 10, G.ptr(w),
+                         G.stream())
+        got = G.host(dout, np.uint8).reshape(n, 100)
+        assert np.array_equal(got, oracle.sort_records(recs))
+    finally:
+        del os.environ["T9_MSB_LEVELS"]

@@ -37,6 +37,10 @@
 __global__ void k_colsum(const u32*, u64, u32*);
 __global__ void k_chunkscan(u32*, u64, u32*);
 __global__ void k_finaloffs(u32*, u64, const u32*, const u32*);
+/* exclusive u32 scan + total (t9_group.hip) */
+__global__ void k_grp_scan(u32*, u64, u64*);
+
+#define T9_L3_MIN 240000000ull   /* 3 MSB levels above this n */
 
 extern "C" int t9i_sort_pairs_lsd(t9_context*, u64*, u32*, u64, void*,
                                   void*);
@@ -276,6 +280,184 @@ __global__ __launch_bounds__(1024, 4) void k_scatter_seg(
     }
 }
 
+/* nb[i] = blocks needed by sub-bucket i at T9_MSB_TILE granularity
+ * (input to the exclusive scan that builds the pass-3 block map) */
+__global__ __launch_bounds__(256) void k_nb_of(const u32* __restrict__ sub_n,
+                                               u32 nsub,
+                                               u32* __restrict__ nb) {
+    const u32 i = blockIdx.x * 256 + threadIdx.x;
+    if (i < nsub) nb[i] = (sub_n[i] + T9_MSB_TILE - 1) / T9_MSB_TILE;
+}
+
+/* find the segment owning block b via the exclusive blocks-prefix bp */
+__device__ inline u32 seg_of_block(const u32* __restrict__ bp, u32 nsub,
+                                   u32 b) {
+    u32 lo = 0, hi = nsub - 1;
+    while (lo < hi) {
+        u32 mid = (lo + hi + 1) >> 1;
+        if (bp[mid] <= b) lo = mid; else hi = mid - 1;
+    }
+    return lo;
+}
+
+/* pass-3 histogram: tiles mapped to (b7,b6) sub-buckets via bp */
+__global__ __launch_bounds__(256) void k_hist_seg3(
+    const u64* __restrict__ keys, const u32* __restrict__ bp, u32 nsub,
+    const u64* __restrict__ d_total, const u32* __restrict__ sub_start,
+    const u32* __restrict__ sub_n, u32 shift, u32* __restrict__ hist) {
+    __shared__ u32 s_cnt[T9_RADIX];
+    const u32 tid = threadIdx.x;
+    s_cnt[tid] = 0;
+    __syncthreads();
+    const u32 b = blockIdx.x;
+    if (b < (u32)*d_total) {
+        const u32 s = seg_of_block(bp, nsub, b);
+        const u32 toff = (b - bp[s]) * T9_MSB_TILE;
+        const u32 base = sub_start[s] + toff;
+        const u32 rem = sub_n[s] - toff;
+        const u32 tn = (rem < (u32)T9_MSB_TILE) ? rem : (u32)T9_MSB_TILE;
+        for (u32 i = tid; i < tn; i += 256)
+            atomicAdd(&s_cnt[(u32)(keys[(u64)base + i] >> shift) & 255u],
+                      1u);
+    }
+    __syncthreads();
+    hist[(u64)blockIdx.x * T9_RADIX + tid] = s_cnt[tid];
+}
+
+/* pass-3 per-sub2 segmented scan -> final offsets + sub3 index */
+__global__ __launch_bounds__(256) void k_seg_scan3(
+    u32* __restrict__ hist, const u32* __restrict__ bp, u32 nsub,
+    const u32* __restrict__ sub_start, const u32* __restrict__ sub_n,
+    u32* __restrict__ sub3_start, u32* __restrict__ sub3_n) {
+    const u32 s = blockIdx.x;
+    const u32 tid = threadIdx.x;
+    const u32 r0 = bp[s];
+    const u32 rows = (sub_n[s] + T9_MSB_TILE - 1) / T9_MSB_TILE;
+    u32 total = 0;
+    for (u32 r = 0; r < rows; ++r)
+        total += hist[(u64)(r0 + r) * T9_RADIX + tid];
+    __shared__ u32 sh[T9_RADIX];
+    sh[tid] = total;
+    __syncthreads();
+    for (int off = 1; off < T9_RADIX; off <<= 1) {
+        u32 y = (tid >= (u32)off) ? sh[tid - off] : 0;
+        __syncthreads();
+        sh[tid] += y;
+        __syncthreads();
+    }
+    const u32 excl = sh[tid] - total;
+    const u32 gstart = sub_start[s] + excl;
+    sub3_start[(u64)s * T9_RADIX + tid] = gstart;
+    sub3_n[(u64)s * T9_RADIX + tid] = total;
+    u32 run = gstart;
+    for (u32 r = 0; r < rows; ++r) {
+        u32 v = hist[(u64)(r0 + r) * T9_RADIX + tid];
+        hist[(u64)(r0 + r) * T9_RADIX + tid] = run;
+        run += v;
+    }
+}
+
+/* pass-3 scatter (1024 threads, wave-autonomous, bp-mapped tiles) */
+template <bool HAS_VAL>
+__global__ __launch_bounds__(1024, 4) void k_scatter_seg3(
+    const u64* __restrict__ in_keys, const u32* __restrict__ in_vals,
+    const u32* __restrict__ bp, u32 nsub, const u64* __restrict__ d_total,
+    const u32* __restrict__ sub_start, const u32* __restrict__ sub_n,
+    u64* __restrict__ out_keys, u32* __restrict__ out_vals,
+    const u32* __restrict__ offs, u32 shift) {
+    constexpr int TILE = T9_MSB_TILE;
+    constexpr int NW = 16;
+    constexpr int SUB = TILE / NW;
+    constexpr int GROUPS = SUB / 64;
+    __shared__ u64 s_okeys[TILE];
+    __shared__ u32 s_ovals[HAS_VAL ? TILE : 1];
+    __shared__ u16 s_rank[TILE];
+    __shared__ u8 s_digof[TILE];
+    __shared__ u32 s_wcnt[NW * T9_RADIX];
+    __shared__ u32 s_woff[NW * T9_RADIX];
+    __shared__ u32 s_start[T9_RADIX];
+    __shared__ u32 s_goff[T9_RADIX];
+    __shared__ u32 s_range[2];   /* base, tn */
+
+    const u32 tid = threadIdx.x, wave = tid >> 6, lane = tid & 63;
+    const u32 b = blockIdx.x;
+    if (b >= (u32)*d_total) return;
+    if (tid == 0) {
+        const u32 s = seg_of_block(bp, nsub, b);
+        const u32 toff = (b - bp[s]) * TILE;
+        s_range[0] = sub_start[s] + toff;
+        const u32 rem = sub_n[s] - toff;
+        s_range[1] = (rem < (u32)TILE) ? rem : (u32)TILE;
+    }
+    if (tid < T9_RADIX)
+        s_goff[tid] = offs[(u64)b * T9_RADIX + tid];
+    for (u32 t = lane; t < T9_RADIX; t += 64) s_wcnt[wave * T9_RADIX + t] = 0;
+    __syncthreads();
+    const u32 base = s_range[0], tn = s_range[1];
+
+    const u32 wbase = wave * SUB;
+    for (int g = 0; g < GROUPS; ++g) {
+        const u32 i = wbase + g * 64 + lane;
+        const bool valid = i < tn;
+        u32 d = 0;
+        if (valid) d = (u32)(in_keys[(u64)base + i] >> shift) & 255u;
+        u64 m = __ballot(valid);
+        for (int bit = 0; bit < 8; ++bit) {
+            u64 bb = __ballot((d >> bit) & 1u);
+            m &= ((d >> bit) & 1u) ? bb : ~bb;
+        }
+        const u32 wr = (u32)__popcll(m & ((1ull << lane) - 1ull));
+        const u32 before = valid ? s_wcnt[wave * T9_RADIX + d] : 0;
+        if (valid) {
+            s_rank[i] = (u16)(before + wr);
+            if (wr == 0)
+                s_wcnt[wave * T9_RADIX + d] = before + (u32)__popcll(m);
+        }
+    }
+    __syncthreads();
+
+    if (tid < T9_RADIX) {
+        u32 run = 0;
+        for (int w = 0; w < NW; ++w) {
+            s_woff[w * T9_RADIX + tid] = run;
+            run += s_wcnt[w * T9_RADIX + tid];
+        }
+        s_start[tid] = run;
+    }
+    __syncthreads();
+    t9_scan256_onewave(s_start, tid);
+    __syncthreads();
+    if (tid < T9_RADIX) {
+        const u32 excl = s_start[tid];
+        for (int w = 0; w < NW; ++w) s_woff[w * T9_RADIX + tid] += excl;
+    }
+    __syncthreads();
+
+    for (int g = 0; g < GROUPS; ++g) {
+        const u32 i = wbase + g * 64 + lane;
+        if (i < tn) {
+            const u64 k = in_keys[(u64)base + i];
+            const u32 d = (u32)(k >> shift) & 255u;
+            const u32 pos = s_woff[wave * T9_RADIX + d] + s_rank[i];
+            s_okeys[pos] = k;
+            if (HAS_VAL) s_ovals[pos] = in_vals[(u64)base + i];
+            s_digof[pos] = (u8)d;
+        }
+    }
+    __syncthreads();
+
+    constexpr int CHUNKS = TILE / 1024;
+    for (int c = 0; c < CHUNKS; ++c) {
+        const u32 j = c * 1024 + tid;
+        if (j < tn) {
+            const u32 d = s_digof[j];
+            const u64 gpos = (u64)s_goff[d] + (j - s_start[d]);
+            out_keys[gpos] = s_okeys[j];
+            if (HAS_VAL) out_vals[gpos] = s_ovals[j];
+        }
+    }
+}
+
 /* Greedy span packing: pack consecutive (b7,b6) sub-buckets into spans
  * of <= SPANMAX elements. A span may cover several sub-buckets because
  * their top-16 key bits are distinct and ordered, so sorting the span by
@@ -289,7 +471,7 @@ __global__ __launch_bounds__(256) void k_span_pack(
     const u32* __restrict__ sub_start, const u32* __restrict__ sub_n,
     u32 spanmax, u32* __restrict__ span_count,
     u32* __restrict__ span_start, u32* __restrict__ span_len) {
-    const u32 t = threadIdx.x;
+    const u32 t = blockIdx.x * 256 + threadIdx.x;
     u32 cur_start = 0, cur_len = 0;
     for (u32 j = 0; j < 256; ++j) {
         const u32 i = t * 256 + j;
@@ -321,7 +503,7 @@ __global__ __launch_bounds__(256) void k_span_pack(
 /* sub-bucket stats: info[0] = max size (atomicMax), info[1] = count of
  * sub-buckets above hardmax, list = their indices */
 __global__ __launch_bounds__(256) void k_subinfo(
-    const u32* __restrict__ sub_n, u32 nsub, u32 hardmax,
+    const u32* __restrict__ sub_n, u32 nsub, u32 hardmax, u32 list_cap,
     u32* __restrict__ info, u32* __restrict__ list) {
     const u32 i = blockIdx.x * 256 + threadIdx.x;
     if (i < nsub) {
@@ -329,7 +511,7 @@ __global__ __launch_bounds__(256) void k_subinfo(
         atomicMax(&info[0], v);
         if (v > hardmax) {
             u32 pos = atomicAdd(&info[1], 1u);
-            list[pos] = i;
+            if (pos < list_cap) list[pos] = i;
         }
     }
 }
@@ -444,7 +626,7 @@ __global__ __launch_bounds__(BLOCK, 4) void k_lds_sort_sub(
 /* span LDS sort: up to SPANMAX pairs, 8 stable ballot-ranked passes over
  * the FULL 64-bit key (spans cover several top-16 groups). in == out is
  * allowed (in-place); reads go to LDS before any write-back. */
-template <int SPANMAX, int BLOCK, bool HAS_VAL>
+template <int SPANMAX, int BLOCK, bool HAS_VAL, int PASSES>
 __global__ __launch_bounds__(BLOCK, 4) void k_lds_sort_span(
     const u64* __restrict__ keys_in, const u32* __restrict__ vals_in,
     u64* __restrict__ keys_out, u32* __restrict__ vals_out,
@@ -497,7 +679,7 @@ __global__ __launch_bounds__(BLOCK, 4) void k_lds_sort_span(
     }
 
     int cur = 0;
-    for (int pass = 0; pass < 8; ++pass) {
+    for (int pass = 0; pass < PASSES; ++pass) {
         const u32 shift = pass * 8;
         for (u32 t = lane; t < T9_RADIX; t += 64)
             s_wcnt[wave * T9_RADIX + t] = 0;
@@ -571,8 +753,16 @@ struct MsbWs {
     u32* sub_n;
     u32* ovr;        /* [0] = max or count, [1] = count, [2..] = list */
     u32* span;
+    /* 3-level extras (allocated only for n >= T9_L3_MIN) */
+    u32* bp;         /* NSUB blocks-prefix */
+    u64* d_total;
+    u32* sub3_start;
+    u32* sub3_n;
+    u32* span3;      /* [0]=count, starts, lens */
     u64 B1, B2max;
 };
+
+constexpr u64 NSUB3 = 256ull * 256 * 256;
 
 constexpr u32 NSUB = 256 * 256;
 
@@ -581,6 +771,11 @@ MsbWs carve_msb(char* p, u64 n) {
     const u64 npad = n + 256ull * T9_MSB_TILE;
     w.B1 = t9_ceil_div(n, T9_MSB_TILE);
     w.B2max = t9_ceil_div(n, T9_MSB_TILE) + 256;
+    {
+        const char* l3c = getenv("T9_MSB_LEVELS");
+        if (n >= T9_L3_MIN || (l3c && atoi(l3c) == 3))
+            w.B2max = t9_ceil_div(n, T9_MSB_TILE) + NSUB;
+    }
     w.alt_k = (u64*)p;
     p += t9_align256(npad * 8);
     w.alt_v = (u32*)p;
@@ -605,13 +800,34 @@ MsbWs carve_msb(char* p, u64 n) {
     p += t9_align256((u64)(NSUB + 2) * 4);
     w.span = (u32*)p;   /* [0]=count, [1..CAP]=start, [1+CAP..]=len */
     p += t9_align256((u64)(2 * (NSUB + 64) + 1) * 4);
+    const char* l3 = getenv("T9_MSB_LEVELS");
+    if (n >= T9_L3_MIN || (l3 && atoi(l3) == 3)) {
+        w.bp = (u32*)p;
+        p += t9_align256((u64)(NSUB + 1) * 4);
+        w.d_total = (u64*)p;
+        p += 256;
+        w.sub3_start = (u32*)p;
+        p += t9_align256(NSUB3 * 4);
+        w.sub3_n = (u32*)p;
+        p += t9_align256(NSUB3 * 4);
+        w.span3 = (u32*)p;
+        p += t9_align256((2 * (NSUB3 + 64) + 1) * 4);
+    }
     return w;
 }
 
 u64 msb_ws_bytes(u64 n) {
     const u64 npad = n + 256ull * T9_MSB_TILE;
-    const u64 B2max = t9_ceil_div(n, T9_MSB_TILE) + 256;
-    return t9_align256(npad * 8) + t9_align256(npad * 4) +
+    const char* l3e = getenv("T9_MSB_LEVELS");
+    u64 B2max = t9_ceil_div(n, T9_MSB_TILE) + 256;
+    if (n >= T9_L3_MIN || (l3e && atoi(l3e) == 3))
+        B2max = t9_ceil_div(n, T9_MSB_TILE) + NSUB;   /* pass-3 rows */
+    u64 l3 = 0;
+    if (n >= T9_L3_MIN || (l3e && atoi(l3e) == 3))
+        l3 = t9_align256((u64)(NSUB + 1) * 4) + 256 +
+             2 * t9_align256(NSUB3 * 4) +
+             t9_align256((2 * (NSUB3 + 64) + 1) * 4);
+    return l3 + t9_align256(npad * 8) + t9_align256(npad * 4) +
            t9_align256(B2max * T9_RADIX * 4) +
            t9_align256(t9_ceil_div(B2max, T9_SCAN_CHUNK) * T9_RADIX * 4) +
            4 * t9_align256(T9_RADIX * 4) + t9_align256(257 * 4) +
@@ -675,6 +891,90 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
         T9_LAUNCH_CHECK();
     }
 
+    const char* l3e = getenv("T9_MSB_LEVELS");
+    const bool use3 = (n >= T9_L3_MIN) ||
+                      (l3e && atoi(l3e) == 3 && n >= (1ull << 14));
+    if (use3) {
+        /* ---- pass 3: byte 5, segmented per (b7,b6) sub-bucket
+         * (d_keys -> alt), then level 4: span-packed LDS sort over the
+         * (b7,b6,b5) sub-buckets (low 48 bits; spans stay inside one
+         * (b7,b6) bucket), alt -> d_keys ---- */
+        hipLaunchKernelGGL(k_nb_of, dim3(NSUB / 256), dim3(256), 0, s,
+                           w.sub_n, NSUB, w.bp);
+        hipLaunchKernelGGL(k_grp_scan, dim3(1), dim3(256), 0, s, w.bp,
+                           (u64)NSUB, w.d_total);
+        const u64 B3 = t9_ceil_div(n, T9_MSB_TILE) + NSUB;
+        T9_PERF_WRAP(s, "hist_pairs",
+                     hipLaunchKernelGGL(k_hist_seg3, dim3((u32)B3),
+                                        dim3(256), 0, s, d_keys, w.bp,
+                                        NSUB, w.d_total, w.sub_start,
+                                        w.sub_n, 40, w.hist));
+        hipLaunchKernelGGL(k_seg_scan3, dim3(NSUB), dim3(256), 0, s,
+                           w.hist, w.bp, NSUB, w.sub_start, w.sub_n,
+                           w.sub3_start, w.sub3_n);
+        T9_PERF_WRAP(
+            s, "pair_scatter",
+            hipLaunchKernelGGL((k_scatter_seg3<HAS_VAL>), dim3((u32)B3),
+                               dim3(1024), 0, s, d_keys, d_vals, w.bp,
+                               NSUB, w.d_total, w.sub_start, w.sub_n,
+                               w.alt_k, w.alt_v, w.hist, 40));
+        T9_LAUNCH_CHECK();
+
+        const u64 CAP3 = NSUB3 + 64;
+        u32* s3_count = w.span3;
+        u32* s3_start = w.span3 + 1;
+        u32* s3_len = w.span3 + 1 + CAP3;
+        HIP_TRY(hipMemsetAsync(w.span3, 0, (2 * CAP3 + 1) * 4, s));
+        hipLaunchKernelGGL(k_span_pack, dim3(NSUB / 256), dim3(256), 0, s,
+                           w.sub3_start, w.sub3_n, T9_SUBMAX, s3_count,
+                           s3_start, s3_len);
+        HIP_TRY(hipMemsetAsync(w.ovr, 0, 8, s));
+        hipLaunchKernelGGL(k_subinfo, dim3((u32)(CAP3 / 256 + 1)),
+                           dim3(256), 0, s, s3_len, (u32)CAP3, T9_SUBMAX,
+                           NSUB, w.ovr, w.ovr + 2);
+        u32 hdr[2] = { 0, 0 };
+        u32 nspan = 0;
+        HIP_TRY(hipMemcpyAsync(&nspan, s3_count, 4,
+                               hipMemcpyDeviceToHost, s));
+        HIP_TRY(hipMemcpyAsync(hdr, w.ovr, 8, hipMemcpyDeviceToHost, s));
+        HIP_TRY(hipStreamSynchronize(s));
+        const u32 novr3 = hdr[1];
+        T9_PERF_WRAP(
+            s, "lds_sort",
+            hipLaunchKernelGGL(
+                (k_lds_sort_span<T9_SUBMAX, 1024, HAS_VAL, 6>),
+                dim3(nspan ? nspan : 1), dim3(1024), 0, s, w.alt_k,
+                w.alt_v, d_keys, d_vals, s3_start, s3_len));
+        T9_LAUNCH_CHECK();
+        if (novr3 == 0) return T9_OK;
+        if (novr3 > 64)
+            return HAS_VAL ? t9i_sort_pairs_lsd(ctx, d_keys, d_vals, n,
+                                                d_workspace, stream)
+                           : t9i_sort_keys_lsd(ctx, d_keys, n, d_workspace,
+                                               stream);
+        HIP_TRY(hipStreamSynchronize(s));
+        std::vector<u32> list3(novr3);
+        HIP_TRY(hipMemcpy(list3.data(), w.ovr + 2, novr3 * 4,
+                          hipMemcpyDeviceToHost));
+        std::vector<u32> st3(novr3), cn3(novr3);
+        for (u32 i = 0; i < novr3; ++i) {
+            HIP_TRY(hipMemcpy(&st3[i], s3_start + list3[i], 4,
+                              hipMemcpyDeviceToHost));
+            HIP_TRY(hipMemcpy(&cn3[i], s3_len + list3[i], 4,
+                              hipMemcpyDeviceToHost));
+        }
+        for (u32 i = 0; i < novr3; ++i) {
+            int rc = HAS_VAL
+                         ? t9i_sort_pairs_lsd(ctx, d_keys + st3[i],
+                                              d_vals + st3[i], cn3[i],
+                                              d_workspace, stream)
+                         : t9i_sort_keys_lsd(ctx, d_keys + st3[i], cn3[i],
+                                             d_workspace, stream);
+            if (rc) return rc;
+        }
+        return T9_OK;
+    }
+
     /* ---- level 3: in-LDS sort; spans pack consecutive sub-buckets to
      * full 4096-element blocks sorted by the complete 64-bit key
      * (T9_LDS_SPAN=0 falls back to one block per sub-bucket) ---- */
@@ -698,16 +998,18 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
                            span_start, span_len);
         HIP_TRY(hipMemsetAsync(w.ovr, 0, 8, s));
         hipLaunchKernelGGL(k_subinfo, dim3(CAP / 256 + 1), dim3(256), 0, s,
-                           span_len, CAP, T9_SUBMAX, w.ovr, w.ovr + 2);
+                           span_len, CAP, T9_SUBMAX, NSUB, w.ovr,
+                           w.ovr + 2);
         u32 info[2] = { 0, 0 };
         HIP_TRY(hipMemcpyAsync(info, w.ovr, 8, hipMemcpyDeviceToHost, s));
         HIP_TRY(hipStreamSynchronize(s));
         novr = info[1];
         T9_PERF_WRAP(
             s, "lds_sort",
-            hipLaunchKernelGGL((k_lds_sort_span<T9_SUBMAX, 1024, HAS_VAL>),
-                               dim3(CAP), dim3(1024), 0, s, d_keys, d_vals,
-                               d_keys, d_vals, span_start, span_len));
+            hipLaunchKernelGGL(
+                (k_lds_sort_span<T9_SUBMAX, 1024, HAS_VAL, 7>),
+                dim3(CAP), dim3(1024), 0, s, d_keys, d_vals, d_keys,
+                d_vals, span_start, span_len));
         T9_LAUNCH_CHECK();
         sel_start = span_start;
         sel_n = span_len;
@@ -715,7 +1017,8 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
     else {
         HIP_TRY(hipMemsetAsync(w.ovr, 0, 8, s));
         hipLaunchKernelGGL(k_subinfo, dim3(NSUB / 256), dim3(256), 0, s,
-                           w.sub_n, NSUB, T9_SUBMAX, w.ovr, w.ovr + 2);
+                           w.sub_n, NSUB, T9_SUBMAX, NSUB, w.ovr,
+                           w.ovr + 2);
         u32 info[2] = { 0, 0 };
         HIP_TRY(hipMemcpyAsync(info, w.ovr, 8, hipMemcpyDeviceToHost, s));
         HIP_TRY(hipStreamSynchronize(s));
