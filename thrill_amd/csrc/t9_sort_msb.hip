@@ -920,17 +920,29 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
                                w.alt_k, w.alt_v, w.hist, 40));
         T9_LAUNCH_CHECK();
 
+        /* max (b7,b6,b5) group size decides the span geometry: tiny
+         * groups (the common big-n case) pack into 2048-element spans at
+         * 2 blocks/CU — same lever that won on the level-3 sub sort */
+        HIP_TRY(hipMemsetAsync(w.ovr, 0, 8, s));
+        hipLaunchKernelGGL(k_subinfo, dim3((u32)(NSUB3 / 256)), dim3(256),
+                           0, s, w.sub3_n, (u32)NSUB3, T9_SUBMAX, NSUB,
+                           w.ovr, w.ovr + 2);
+        u32 mx[1] = { 0 };
+        HIP_TRY(hipMemcpyAsync(mx, w.ovr, 4, hipMemcpyDeviceToHost, s));
+        HIP_TRY(hipStreamSynchronize(s));
+        const u32 spanmax = (mx[0] <= 2048) ? 2048 : T9_SUBMAX;
+
         const u64 CAP3 = NSUB3 + 64;
         u32* s3_count = w.span3;
         u32* s3_start = w.span3 + 1;
         u32* s3_len = w.span3 + 1 + CAP3;
         HIP_TRY(hipMemsetAsync(w.span3, 0, (2 * CAP3 + 1) * 4, s));
         hipLaunchKernelGGL(k_span_pack, dim3(NSUB / 256), dim3(256), 0, s,
-                           w.sub3_start, w.sub3_n, T9_SUBMAX, s3_count,
+                           w.sub3_start, w.sub3_n, spanmax, s3_count,
                            s3_start, s3_len);
         HIP_TRY(hipMemsetAsync(w.ovr, 0, 8, s));
         hipLaunchKernelGGL(k_subinfo, dim3((u32)(CAP3 / 256 + 1)),
-                           dim3(256), 0, s, s3_len, (u32)CAP3, T9_SUBMAX,
+                           dim3(256), 0, s, s3_len, (u32)CAP3, spanmax,
                            NSUB, w.ovr, w.ovr + 2);
         u32 hdr[2] = { 0, 0 };
         u32 nspan = 0;
@@ -941,10 +953,16 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
         const u32 novr3 = hdr[1];
         T9_PERF_WRAP(
             s, "lds_sort",
-            hipLaunchKernelGGL(
-                (k_lds_sort_span<T9_SUBMAX, 1024, HAS_VAL, 6>),
-                dim3(nspan ? nspan : 1), dim3(1024), 0, s, w.alt_k,
-                w.alt_v, d_keys, d_vals, s3_start, s3_len));
+            if (spanmax == 2048)
+                hipLaunchKernelGGL(
+                    (k_lds_sort_span<2048, 512, HAS_VAL, 6>),
+                    dim3(nspan ? nspan : 1), dim3(512), 0, s, w.alt_k,
+                    w.alt_v, d_keys, d_vals, s3_start, s3_len);
+            else
+                hipLaunchKernelGGL(
+                    (k_lds_sort_span<T9_SUBMAX, 1024, HAS_VAL, 6>),
+                    dim3(nspan ? nspan : 1), dim3(1024), 0, s, w.alt_k,
+                    w.alt_v, d_keys, d_vals, s3_start, s3_len));
         T9_LAUNCH_CHECK();
         if (novr3 == 0) return T9_OK;
         if (novr3 > 64)
