@@ -72,6 +72,12 @@ int t9_extract_key64(t9_context* ctx, const uint8_t* d_recs, uint64_t n,
                      uint32_t rec_size, uint32_t key_off, uint64_t* d_keys,
                      uint32_t* d_idx, void* stream);
 
+/* Little-endian variant: the key is a native uint64_t field (numeric
+ * order) — BASELINE config 5's struct{u64 key; u8 payload[120]}. */
+int t9_extract_key64_le(t9_context* ctx, const uint8_t* d_recs, uint64_t n,
+                        uint32_t rec_size, uint32_t key_off,
+                        uint64_t* d_keys, uint32_t* d_idx, void* stream);
+
 /* out[i] = recs[idx[i]] for fixed-size records (rec_size % 4 == 0). */
 int t9_gather_records(t9_context* ctx, const uint8_t* d_recs,
                       const uint32_t* d_idx, uint64_t n, uint32_t rec_size,
@@ -88,6 +94,12 @@ uint64_t t9_sort_records_workspace(uint64_t n, uint32_t rec_size);
 int t9_sort_records(t9_context* ctx, const uint8_t* d_in, uint8_t* d_out,
                     uint64_t n, uint32_t rec_size, uint32_t key_len,
                     void* d_workspace, void* stream);
+
+/* config-5 variant: sort records whose key is a native little-endian
+ * uint64_t at offset 0 (numeric order; payload-byte tiebreak). */
+int t9_sort_records_keyle(t9_context* ctx, const uint8_t* d_in,
+                          uint8_t* d_out, uint64_t n, uint32_t rec_size,
+                          void* d_workspace, void* stream);
 
 /* ------------------------------------------------------------------ *
  * Classification + partition — replaces TransmitItems' tree-descent loop

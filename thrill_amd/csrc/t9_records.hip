@@ -69,7 +69,11 @@ __global__ __launch_bounds__(256) void k_gen_records(u8* __restrict__ out,
     for (u32 w = tid; w * 4 < nb; w += 256) gout[w] = gin[w];
 }
 
-/* big-endian u64 key prefix at key_off + record index iota */
+/* u64 key prefix at key_off + record index iota. BE (default): the key
+ * is 8 lexicographic bytes (TeraSort, sorts like memcmp); LE: a native
+ * uint64_t field (BASELINE config 5's struct{u64 key; u8 payload[120]}),
+ * sorted numerically. */
+template <bool LE>
 __global__ __launch_bounds__(256) void k_extract_key64(
     const u8* __restrict__ recs, u64 n, u32 rec_words, u32 key_off,
     u64* __restrict__ keys, u32* __restrict__ idx) {
@@ -78,7 +82,9 @@ __global__ __launch_bounds__(256) void k_extract_key64(
     for (u64 i = (u64)blockIdx.x * 256 + threadIdx.x; i < n; i += stride) {
         u64 w = (u64)i * rec_words + key_off / 4;
         u32 w0 = r32[w], w1 = r32[w + 1];
-        keys[i] = ((u64)__builtin_bswap32(w0) << 32) | __builtin_bswap32(w1);
+        keys[i] = LE ? (((u64)w1 << 32) | w0)
+                     : (((u64)__builtin_bswap32(w0) << 32) |
+                        __builtin_bswap32(w1));
         idx[i] = (u32)i;
     }
 }
@@ -217,10 +223,29 @@ int t9_extract_key64(t9_context* ctx, const u8* d_recs, u64 n, u32 rec_size,
     const u32 ecap = eg ? (u32)atoi(eg) : 16384;
     const dim3 egrid((u32)((ewant < ecap) ? (ewant ? ewant : 1) : ecap));
     T9_PERF_WRAP((hipStream_t)stream, "extract",
-                 hipLaunchKernelGGL(k_extract_key64, egrid, dim3(256), 0,
-                                    (hipStream_t)stream, d_recs, n,
-                                    rec_size / 4, key_off, d_keys,
-                                    d_idx));
+                 hipLaunchKernelGGL((k_extract_key64<false>), egrid,
+                                    dim3(256), 0, (hipStream_t)stream,
+                                    d_recs, n, rec_size / 4, key_off,
+                                    d_keys, d_idx));
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+int t9_extract_key64_le(t9_context* ctx, const u8* d_recs, u64 n,
+                        u32 rec_size, u32 key_off, u64* d_keys, u32* d_idx,
+                        void* stream) {
+    (void)ctx;
+    if (!d_recs || !d_keys || !d_idx) return T9_EINVAL;
+    if (rec_size % 4 || key_off % 4 || key_off + 8 > rec_size)
+        return T9_EINVAL;
+    if (n == 0) return T9_OK;
+    u64 ewant = t9_ceil_div(n, 256);
+    const dim3 egrid((u32)((ewant < 16384) ? (ewant ? ewant : 1) : 16384));
+    T9_PERF_WRAP((hipStream_t)stream, "extract",
+                 hipLaunchKernelGGL((k_extract_key64<true>), egrid,
+                                    dim3(256), 0, (hipStream_t)stream,
+                                    d_recs, n, rec_size / 4, key_off,
+                                    d_keys, d_idx));
     T9_LAUNCH_CHECK();
     return T9_OK;
 }
@@ -286,6 +311,10 @@ int t9_zipf_tokens(t9_context* ctx, u64* d_out, const double* d_cdf, u64 N,
     return T9_OK;
 }
 
+int t9_sort_records_keyle(t9_context* ctx, const u8* d_in, u8* d_out,
+                          u64 n, u32 rec_size, void* d_workspace,
+                          void* stream);
+
 u64 t9_sort_records_workspace(u64 n, u32 rec_size) {
     (void)rec_size;
     if (n < 2) return 256;
@@ -300,9 +329,29 @@ u64 t9_sort_records_workspace(u64 n, u32 rec_size) {
  * host pass, since at the benchmark's uniform-key sizes the expected number
  * of colliding 8-byte prefixes is < 1 (SURVEY.md §7 hard part (a));
  * adversarial all-equal inputs take the slow path but stay correct). */
+static int sort_records_impl(t9_context* ctx, const u8* d_in, u8* d_out,
+                             u64 n, u32 rec_size, u32 key_len,
+                             void* d_workspace, void* stream, bool le);
+
 int t9_sort_records(t9_context* ctx, const u8* d_in, u8* d_out, u64 n,
                     u32 rec_size, u32 key_len, void* d_workspace,
                     void* stream) {
+    return sort_records_impl(ctx, d_in, d_out, n, rec_size, key_len,
+                             d_workspace, stream, false);
+}
+
+/* config-5 variant: the record key is a native little-endian uint64_t at
+ * offset 0, ordered numerically; payload ties by byte order. */
+int t9_sort_records_keyle(t9_context* ctx, const u8* d_in, u8* d_out,
+                          u64 n, u32 rec_size, void* d_workspace,
+                          void* stream) {
+    return sort_records_impl(ctx, d_in, d_out, n, rec_size, 8, d_workspace,
+                             stream, true);
+}
+
+static int sort_records_impl(t9_context* ctx, const u8* d_in, u8* d_out,
+                             u64 n, u32 rec_size, u32 key_len,
+                             void* d_workspace, void* stream, bool le) {
     if (rec_size % 4 || key_len > rec_size || n >= (1ull << 32))
         return T9_EINVAL;
     hipStream_t s = (hipStream_t)stream;
@@ -330,13 +379,15 @@ int t9_sort_records(t9_context* ctx, const u8* d_in, u8* d_out, u64 n,
     bool fused = fe && atoi(fe) && n >= (1ull << 14) &&
                  (rec_size == 100 || rec_size == 128);
     int rc;
-    if (fused) {
+    if (fused && !le) {
         rc = t9i_sort_recs_msb(ctx, d_in, rec_size, d_keys, d_idx, n,
                                pair_ws, stream);
     }
     else {
-        rc = t9_extract_key64(ctx, d_in, n, rec_size, 0, d_keys, d_idx,
-                              stream);
+        rc = le ? t9_extract_key64_le(ctx, d_in, n, rec_size, 0, d_keys,
+                                      d_idx, stream)
+                : t9_extract_key64(ctx, d_in, n, rec_size, 0, d_keys,
+                                   d_idx, stream);
         if (rc) return rc;
         rc = t9_sort_pairs_u64_u32(ctx, d_keys, d_idx, n, pair_ws, stream);
     }

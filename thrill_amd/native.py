@@ -48,6 +48,8 @@ _SIGS = {
     "t9_gather_records": (i32, [vp, vp, vp, u64, u32, vp, vp]),
     "t9_sort_records_workspace": (u64, [u64, u32]),
     "t9_sort_records": (i32, [vp, vp, vp, u64, u32, u32, vp, vp]),
+    "t9_sort_records_keyle": (i32, [vp, vp, vp, u64, u32, vp, vp]),
+    "t9_extract_key64_le": (i32, [vp, vp, u64, u32, u32, vp, vp, vp]),
     "t9_classify_u64": (i32, [vp, vp, u64, u64, vp, vp, u32, vp, vp, vp]),
     "t9_classify_rec": (i32, [vp, vp, vp, u64, u64, vp, vp, vp, u32, u32,
                               vp, vp, vp]),
