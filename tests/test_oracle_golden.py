@@ -271,3 +271,35 @@ def test_sort_records_parallel_matches_serial(oracle):
     par, threads = oracle.sort_records_parallel(recs)
     assert threads >= 1
     assert np.array_equal(par, oracle.sort_records(recs))
+
+
+# ------------------------------------------------------- property fuzzing
+
+try:
+    from hypothesis import given, settings, strategies as st
+
+    @settings(max_examples=25, deadline=None)
+    @given(st.integers(2, 16),
+           st.lists(st.integers(0, 30), min_size=16, max_size=400),
+           st.integers(0, 2**32))
+    def test_fuzz_partition_concat_equals_sort(p, key_list, gidx0):
+        """For ANY splitter choice drawn from the data, classify ->
+        per-bucket sort -> concat == total sort (the invariant SURVEY §8c
+        rests on), with heavy duplicates and arbitrary global offsets."""
+        from tests._oracle import Oracle
+        o = Oracle()
+        keys = np.array(key_list, dtype=np.uint64)
+        n = len(keys)
+        rng = np.random.default_rng(p)
+        pos = np.sort(rng.choice(n, min(n, 8), replace=False)).astype(
+            np.uint64) + gidx0
+        sk = keys[(pos - gidx0).astype(np.int64)]
+        spl_k, spl_i = o.select_splitters_u64(sk, pos, p)
+        buckets = o.classify_u64(keys, gidx0, spl_k, spl_i, p)
+        parts = [np.sort(keys[buckets == b]) for b in range(p)]
+        assert np.array_equal(np.concatenate(parts), np.sort(keys))
+        # tree restatement must agree element-wise
+        tree = o.classify_u64(keys, gidx0, spl_k, spl_i, p, tree=True)
+        assert np.array_equal(buckets, tree)
+except ImportError:  # pragma: no cover
+    pass
