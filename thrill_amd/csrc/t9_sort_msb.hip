@@ -505,15 +505,20 @@ __global__ __launch_bounds__(256) void k_span_pack(
 __global__ __launch_bounds__(256) void k_subinfo(
     const u32* __restrict__ sub_n, u32 nsub, u32 hardmax, u32 list_cap,
     u32* __restrict__ info, u32* __restrict__ list) {
+    __shared__ u32 s_max;
+    if (threadIdx.x == 0) s_max = 0;
+    __syncthreads();
     const u32 i = blockIdx.x * 256 + threadIdx.x;
     if (i < nsub) {
         const u32 v = sub_n[i];
-        atomicMax(&info[0], v);
+        atomicMax(&s_max, v);   /* block-local; one global atomic below */
         if (v > hardmax) {
             u32 pos = atomicAdd(&info[1], 1u);
             if (pos < list_cap) list[pos] = i;
         }
     }
+    __syncthreads();
+    if (threadIdx.x == 0 && s_max) atomicMax(&info[0], s_max);
 }
 
 /* level 3: sort one sub-bucket (<= SUBMAX pairs) in LDS over the low
