@@ -79,12 +79,21 @@ __global__ __launch_bounds__(256) void k_extract_key64(
     u64* __restrict__ keys, u32* __restrict__ idx) {
     const u64 stride = (u64)gridDim.x * 256;
     const u32* r32 = (const u32*)recs;
+    const u64* r64 = (const u64*)recs;
+    const bool aligned8 = LE && (rec_words % 2 == 0) && (key_off % 8 == 0);
     for (u64 i = (u64)blockIdx.x * 256 + threadIdx.x; i < n; i += stride) {
-        u64 w = (u64)i * rec_words + key_off / 4;
-        u32 w0 = r32[w], w1 = r32[w + 1];
-        keys[i] = LE ? (((u64)w1 << 32) | w0)
-                     : (((u64)__builtin_bswap32(w0) << 32) |
-                        __builtin_bswap32(w1));
+        u64 k;
+        if (aligned8) {
+            k = r64[(u64)i * (rec_words / 2) + key_off / 8];
+        }
+        else {
+            u64 w = (u64)i * rec_words + key_off / 4;
+            u32 w0 = r32[w], w1 = r32[w + 1];
+            k = LE ? (((u64)w1 << 32) | w0)
+                   : (((u64)__builtin_bswap32(w0) << 32) |
+                      __builtin_bswap32(w1));
+        }
+        keys[i] = k;
         idx[i] = (u32)i;
     }
 }
