@@ -117,6 +117,13 @@ def main():
         torch.cuda.set_device(local_rank)
         dist.init_process_group("nccl")
         assert world == args.gpus, (world, args.gpus)
+    elif os.environ.get("T9_FORCE_DIST"):
+        # validation mode: run the distributed branch at world=1
+        # (self-exchange) so the exact multi-rank code path is exercised
+        import torch.distributed as dist_mod
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29712")
+        dist_mod.init_process_group("nccl", rank=0, world_size=1)
     assert torch.cuda.is_available(), "bench.py needs a GPU (no CPU path)"
 
     from thrill_amd.pipeline import TeraSort
