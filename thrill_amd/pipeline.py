@@ -152,12 +152,16 @@ class WordCount:
         n_recv = int(recv_counts.sum())
         rk = torch.empty(max(n_recv, 1), dtype=torch.int64, device="cuda")
         rv = torch.empty(max(n_recv, 1), dtype=torch.int64, device="cuda")
-        dist.all_to_all_single(rk[:n_recv], ks[:m],
-                               output_split_sizes=recv_counts.tolist(),
-                               input_split_sizes=send_counts.tolist())
-        dist.all_to_all_single(rv[:n_recv], vs[:m],
-                               output_split_sizes=recv_counts.tolist(),
-                               input_split_sizes=send_counts.tolist())
+        if self.world == 1:
+            rk[:n_recv].copy_(ks[:m])
+            rv[:n_recv].copy_(vs[:m])
+        else:
+            dist.all_to_all_single(rk[:n_recv], ks[:m],
+                                   output_split_sizes=recv_counts.tolist(),
+                                   input_split_sizes=send_counts.tolist())
+            dist.all_to_all_single(rv[:n_recv], vs[:m],
+                                   output_split_sizes=recv_counts.tolist(),
+                                   input_split_sizes=send_counts.tolist())
         ok2, ov2, m2 = self._reduce(rk, rv, n_recv)
         return ok2.clone(), ov2.clone(), m2
 
@@ -300,10 +304,16 @@ class TeraSort:
         n_recv = int(recv_counts.sum())
         d_recv = torch.empty(max(n_recv, 1) * REC, dtype=torch.uint8,
                              device="cuda")
-        dist.all_to_all_single(
-            d_recv[:n_recv * REC], self.d_send,
-            output_split_sizes=(recv_counts * REC).tolist(),
-            input_split_sizes=(send_counts * REC).tolist())
+        if self.world == 1:
+            # loopback: direct device copy (the same shortcut
+            # t9_alltoall takes at world==1; a 10.7 GB NCCL self-exchange
+            # measured as a hang)
+            d_recv[:n_recv * REC].copy_(self.d_send[:n_recv * REC])
+        else:
+            dist.all_to_all_single(
+                d_recv[:n_recv * REC], self.d_send,
+                output_split_sizes=(recv_counts * REC).tolist(),
+                input_split_sizes=(send_counts * REC).tolist())
         d_sorted = torch.empty(max(n_recv, 1) * REC, dtype=torch.uint8,
                                device="cuda")
         ws_need = nat.ws("sort_records", n_recv, REC)
