@@ -10,6 +10,7 @@ No CPU fallback: everything data-sized runs through the C ABI on the GPU.
 """
 import ctypes
 import math
+import os
 
 import numpy as np
 import torch
@@ -119,7 +120,6 @@ class WordCount:
     def step(self):
         """One full ReduceByKey of the (distributed) token stream. Returns
         (keys tensor, vals tensor, m) of this rank's final pairs."""
-        import os
         nat, s = self.nat, _stream()
         ok, ov, m = self._reduce(self.d_toks, self.d_ones, self.n_local)
         if self.world == 1 and not os.environ.get("T9_FORCE_DIST"):
@@ -197,7 +197,6 @@ class TeraSort:
                 "sort_records", self.n_local + self.n_local // 3 + 16, REC))
         self.d_ws = torch.empty(int(ws_bytes), dtype=torch.uint8,
                                 device="cuda")
-        import os
         if world > 1 or os.environ.get("T9_FORCE_DIST"):
             self.d_keys = torch.empty(self.n_local, dtype=torch.int64,
                                       device="cuda")
@@ -273,7 +272,6 @@ class TeraSort:
         T9_FORCE_DIST=1 routes world==1 through the distributed branch
         (self-exchange) so the exact multi-rank code path is testable on
         one GPU."""
-        import os
         nat, s = self.nat, _stream()
         if self.world == 1 and not os.environ.get("T9_FORCE_DIST"):
             nat.sort_records(_ptr(self.d_in), _ptr(self.d_out),
@@ -296,7 +294,6 @@ class TeraSort:
         nat.gather_records(_ptr(self.d_in), _ptr(self.d_perm), self.n_local,
                            REC, _ptr(self.d_send), s)
         send_counts = self.d_counts.cpu().numpy().astype(np.int64)
-        recv_counts = np.empty(p, dtype=np.int64)
         sc_t = torch.from_numpy(send_counts).cuda()
         rc_t = torch.empty(p, dtype=torch.int64, device="cuda")
         dist.all_to_all_single(rc_t, sc_t)
