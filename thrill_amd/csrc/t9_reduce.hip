@@ -234,16 +234,42 @@ __global__ __launch_bounds__(256) void k_index_bucket(
                   (unsigned long long)scnt[tid]);
 }
 
+/* Drain with block-local aggregation: each block counts its occupied
+ * slots, claims one contiguous output range with a single atomicAdd (a
+ * per-element add on one counter measured ~6 ms at 2^25 slots), then
+ * writes at block-local scanned positions. Output order is arbitrary, as
+ * the reference documents for reducing (word_count_test.cpp:73-74). */
 __global__ __launch_bounds__(256) void k_reduce_drain(
     const u64* __restrict__ tk, const u64* __restrict__ tv, u64 cap,
     u64* __restrict__ ok, u64* __restrict__ ov, u64* __restrict__ out_n) {
+    __shared__ u32 s_pre[256];
+    __shared__ u64 s_base;
+    const u32 tid = threadIdx.x;
     const u64 stride = (u64)gridDim.x * 256;
-    const u64 gid = (u64)blockIdx.x * 256 + threadIdx.x;
-    for (u64 i = gid; i < cap; i += stride) {
-        if (tk[i] != T9_EMPTY) {
-            u64 pos = atomicAdd((unsigned long long*)out_n, 1ull);
-            ok[pos] = tk[i];
-            ov[pos] = tv[i];
+    const u64 gid = (u64)blockIdx.x * 256 + tid;
+    u32 mine = 0;
+    for (u64 i = gid; i < cap; i += stride)
+        if (tk[i] != T9_EMPTY) ++mine;
+    s_pre[tid] = mine;
+    __syncthreads();
+    for (int off = 1; off < 256; off <<= 1) {
+        u32 y = (tid >= (u32)off) ? s_pre[tid - off] : 0;
+        __syncthreads();
+        s_pre[tid] += y;
+        __syncthreads();
+    }
+    if (tid == 255 && s_pre[255])
+        s_base = atomicAdd((unsigned long long*)out_n,
+                           (unsigned long long)s_pre[255]);
+    __syncthreads();
+    if (mine) {
+        u64 pos = s_base + s_pre[tid] - mine;
+        for (u64 i = gid; i < cap; i += stride) {
+            if (tk[i] != T9_EMPTY) {
+                ok[pos] = tk[i];
+                ov[pos] = tv[i];
+                ++pos;
+            }
         }
     }
     if (gid == 0 && tk[cap] > 0) {
