@@ -472,7 +472,26 @@ __global__ __launch_bounds__(256) void k_span_pack(
     u32 spanmax, u32* __restrict__ span_count,
     u32* __restrict__ span_start, u32* __restrict__ span_len) {
     const u32 t = blockIdx.x * 256 + threadIdx.x;
-    u32 cur_start = 0, cur_len = 0;
+    /* walk 1: count this walker's spans; one atomicAdd claims the range
+     * (a per-span add on one counter serializes at 3-level span counts) */
+    u32 nspans = 0, cur_len = 0;
+    for (u32 j = 0; j < 256; ++j) {
+        const u32 ns = sub_n[t * 256 + j];
+        if (ns == 0) continue;
+        if (cur_len == 0 || cur_len + ns > spanmax) {
+            if (cur_len) ++nspans;
+            cur_len = ns;
+        }
+        else {
+            cur_len += ns;
+        }
+    }
+    if (cur_len) ++nspans;
+    if (nspans == 0) return;
+    u32 slot = atomicAdd(span_count, nspans);
+    /* walk 2: emit */
+    u32 cur_start = 0;
+    cur_len = 0;
     for (u32 j = 0; j < 256; ++j) {
         const u32 i = t * 256 + j;
         const u32 ns = sub_n[i];
@@ -486,18 +505,15 @@ __global__ __launch_bounds__(256) void k_span_pack(
             cur_len += ns;
         }
         else {
-            const u32 slot = atomicAdd(span_count, 1u);
             span_start[slot] = cur_start;
             span_len[slot] = cur_len;
+            ++slot;
             cur_start = st;
             cur_len = ns;
         }
     }
-    if (cur_len) {
-        const u32 slot = atomicAdd(span_count, 1u);
-        span_start[slot] = cur_start;
-        span_len[slot] = cur_len;
-    }
+    span_start[slot] = cur_start;
+    span_len[slot] = cur_len;
 }
 
 /* sub-bucket stats: info[0] = max size (atomicMax), info[1] = count of
