@@ -301,3 +301,16 @@ def test_merge_u64_tie_source_order(nat):
     dout = G.empty(7, np.uint64)
     nat.merge_u64(G.ptr(da), 3, G.ptr(db), 4, G.ptr(dout), G.stream())
     assert G.host(dout, np.uint64).tolist() == [5, 5, 5, 6, 7, 7, 7]
+
+
+@pytest.mark.parametrize("seed", [11, 222, 3333, 44444, 0xDEAD])
+def test_sort_records_parity_seed_sweep(nat, oracle, seed):
+    n = 50_000
+    recs = oracle.gen_records(n, seed=seed)
+    din = G.dev(recs.reshape(-1))
+    dout = G.empty(n * 100, np.uint8)
+    w = G.ws(nat.ws("sort_records", n, 100))
+    nat.sort_records(G.ptr(din), G.ptr(dout), n, 100, 10, G.ptr(w),
+                     G.stream())
+    got = G.host(dout, np.uint8).reshape(n, 100)
+    assert np.array_equal(got, oracle.sort_records(recs))
