@@ -237,3 +237,29 @@ def test_sort_records_keyle_config5_shape(nat):
     order = np.lexsort(tuple(recs[:, c] for c in range(127, 7, -1))
                        + (keys,))
     assert np.array_equal(got, recs[order])
+
+
+@pytest.mark.parametrize("case", ["uniform", "few_values", "skew_top"])
+def test_pass2_9bit_parity(nat, oracle, case):
+    """experimental 9-bit level-2 digit (T9_PASS2_BITS=9)."""
+    os.environ["T9_PASS2_BITS"] = "9"
+    try:
+        n = 1 << 18
+        rng = np.random.default_rng(len(case) + 7)
+        if case == "uniform":
+            keys = oracle.gen_u64(n, seed=77)
+        elif case == "few_values":
+            keys = rng.integers(0, 9, n).astype(np.uint64) * np.uint64(2**59)
+        else:
+            keys = rng.integers(0, 1 << 30, n).astype(np.uint64)
+        vals = np.arange(n, dtype=np.uint32)
+        dk, dv = G.dev(keys), G.dev(vals)
+        w = G.ws(nat.ws("sort_pairs", n))
+        nat.sort_pairs_u64_u32(G.ptr(dk), G.ptr(dv), n, G.ptr(w),
+                               G.stream())
+        gk, gv = G.host(dk, np.uint64), G.host(dv, np.uint32)
+        assert np.array_equal(gk, np.sort(keys))
+        order = np.argsort(keys, kind="stable").astype(np.uint32)
+        assert np.array_equal(gv, vals[order.astype(np.int64)])
+    finally:
+        del os.environ["T9_PASS2_BITS"]

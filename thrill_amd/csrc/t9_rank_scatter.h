@@ -4,6 +4,27 @@
 #pragma once
 #include "t9_common.h"
 
+/* exclusive scan of NDIG LDS entries by wave 0 alone (generalized form
+ * below; the 256 wrapper keeps existing call sites) */
+template <int NDIG>
+__device__ inline void t9_scan_onewave(u32* s_vals, u32 tid) {
+    if (tid < 64) {
+        const u32 lane = tid;
+        u32 carry = 0;
+        for (int c = 0; c < NDIG / 64; ++c) {
+            const u32 v0 = s_vals[c * 64 + lane];
+            u32 v = v0;
+            for (int off = 1; off < 64; off <<= 1) {
+                u32 y = __shfl_up(v, off);
+                if (lane >= (u32)off) v += y;
+            }
+            const u32 total = __shfl(v, 63);
+            s_vals[c * 64 + lane] = (v - v0) + carry;
+            carry += total;
+        }
+    }
+}
+
 /* exclusive scan of 256 LDS entries by wave 0 alone (4 shfl-scanned
  * 64-lane chunks with a running carry) — replaces the 256-thread
  * Hillis-Steele scan and its 16 block barriers with one. Caller barriers

@@ -40,6 +40,16 @@ __global__ void k_finaloffs(u32*, u64, const u32*, const u32*);
 /* exclusive u32 scan + total (t9_group.hip) */
 __global__ void k_grp_scan(u32*, u64, u64*);
 
+/* 9-bit level-2 experiment (t9_sort_msb9.hip, T9_PASS2_BITS=9) */
+extern "C" void t9i_launch_hist_seg9(u32, void*, const u64*, const u32*,
+                                     const u32*, u32*);
+extern "C" void t9i_launch_seg_scan9(u32, void*, u32*, const u32*,
+                                     const u32*, const u32*, u32*, u32*);
+extern "C" void t9i_launch_scatter_seg9(u32, int, void*, const u64*,
+                                        const u32*, const u32*, const u32*,
+                                        u64*, u32*, const u32*);
+constexpr u64 NSUB9 = 256ull * 512;
+
 #define T9_L3_MIN 240000000ull   /* 3 MSB levels above this n */
 
 extern "C" int t9i_sort_pairs_lsd(t9_context*, u64*, u32*, u64, void*,
@@ -796,6 +806,8 @@ MsbWs carve_msb(char* p, u64 n) {
         const char* l3c = getenv("T9_MSB_LEVELS");
         if (n >= T9_L3_MIN || (l3c && atoi(l3c) == 3))
             w.B2max = t9_ceil_div(n, T9_MSB_TILE) + NSUB;
+        const char* p2c = getenv("T9_PASS2_BITS");
+        if (p2c && atoi(p2c) == 9) w.B2max *= 2;
     }
     w.alt_k = (u64*)p;
     p += t9_align256(npad * 8);
@@ -814,9 +826,9 @@ MsbWs carve_msb(char* p, u64 n) {
     w.bucket_n = (u32*)p;
     p += t9_align256(T9_RADIX * 4);
     w.sub_start = (u32*)p;
-    p += t9_align256((u64)NSUB * 4);
+    p += t9_align256(NSUB9 * 4);
     w.sub_n = (u32*)p;
-    p += t9_align256((u64)NSUB * 4);
+    p += t9_align256(NSUB9 * 4);
     w.ovr = (u32*)p;
     p += t9_align256((u64)(NSUB + 2) * 4);
     w.span = (u32*)p;   /* [0]=count, [1..CAP]=start, [1+CAP..]=len */
@@ -843,6 +855,9 @@ u64 msb_ws_bytes(u64 n) {
     u64 B2max = t9_ceil_div(n, T9_MSB_TILE) + 256;
     if (n >= T9_L3_MIN || (l3e && atoi(l3e) == 3))
         B2max = t9_ceil_div(n, T9_MSB_TILE) + NSUB;   /* pass-3 rows */
+    const char* p2e = getenv("T9_PASS2_BITS");
+    if (p2e && atoi(p2e) == 9)
+        B2max *= 2;   /* 512-bin hist rows */
     u64 l3 = 0;
     if (n >= T9_L3_MIN || (l3e && atoi(l3e) == 3))
         l3 = t9_align256((u64)(NSUB + 1) * 4) + 256 +
@@ -852,7 +867,7 @@ u64 msb_ws_bytes(u64 n) {
            t9_align256(B2max * T9_RADIX * 4) +
            t9_align256(t9_ceil_div(B2max, T9_SCAN_CHUNK) * T9_RADIX * 4) +
            4 * t9_align256(T9_RADIX * 4) + t9_align256(257 * 4) +
-           2 * t9_align256((u64)NSUB * 4) +
+           2 * t9_align256(NSUB9 * 4) +
            t9_align256((u64)(NSUB + 2) * 4) +
            t9_align256((u64)(2 * (NSUB + 64) + 1) * 4);
 }
@@ -891,6 +906,82 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
                 dim3((u32)B), dim3(1024), 0, s, pass1_src, d_vals, w.alt_k,
                 w.alt_v, w.hist, n, 56));
         T9_LAUNCH_CHECK();
+    }
+
+    /* ---- optional 9-bit level 2 (experiment, T9_PASS2_BITS=9) ---- */
+    {
+        const char* p2e = getenv("T9_PASS2_BITS");
+        const char* l3x = getenv("T9_MSB_LEVELS");
+        const bool blocked3 = (n >= T9_L3_MIN) ||
+                              (l3x && atoi(l3x) == 3 && n >= (1ull << 14));
+        if (p2e && atoi(p2e) == 9 && !blocked3) {
+            const u64 B2 = t9_ceil_div(n, T9_MSB_TILE) + 256;
+            T9_PERF_WRAP(s, "hist_pairs",
+                         t9i_launch_hist_seg9((u32)B2, stream, w.alt_k,
+                                              w.abase, w.bucket_n,
+                                              w.hist));
+            t9i_launch_seg_scan9(256, stream, w.hist, w.abase, w.bucket_n,
+                                 w.true_base, w.sub_start, w.sub_n);
+            T9_PERF_WRAP(s, "pair_scatter",
+                         t9i_launch_scatter_seg9((u32)B2, HAS_VAL, stream,
+                                                 w.alt_k, w.alt_v, w.abase,
+                                                 w.bucket_n, d_keys,
+                                                 d_vals, w.hist));
+            T9_LAUNCH_CHECK();
+            HIP_TRY(hipMemsetAsync(w.ovr, 0, 8, s));
+            hipLaunchKernelGGL(k_subinfo, dim3((u32)(NSUB9 / 256)),
+                               dim3(256), 0, s, w.sub_n, (u32)NSUB9,
+                               T9_SUBMAX, NSUB, w.ovr, w.ovr + 2);
+            u32 info[2] = { 0, 0 };
+            HIP_TRY(hipMemcpyAsync(info, w.ovr, 8, hipMemcpyDeviceToHost,
+                                   s));
+            HIP_TRY(hipStreamSynchronize(s));
+            const u32 maxsub = info[0], novr9 = info[1];
+            T9_PERF_WRAP(
+                s, "lds_sort",
+                if (maxsub <= 1024)
+                    hipLaunchKernelGGL(
+                        (k_lds_sort_sub<1024, 256, HAS_VAL>), dim3(NSUB9),
+                        dim3(256), 0, s, d_keys, d_vals, w.sub_start,
+                        w.sub_n);
+                else if (maxsub <= 2048)
+                    hipLaunchKernelGGL(
+                        (k_lds_sort_sub<2048, 512, HAS_VAL>), dim3(NSUB9),
+                        dim3(512), 0, s, d_keys, d_vals, w.sub_start,
+                        w.sub_n);
+                else
+                    hipLaunchKernelGGL(
+                        (k_lds_sort_sub<T9_SUBMAX, 1024, HAS_VAL>),
+                        dim3(NSUB9), dim3(1024), 0, s, d_keys, d_vals,
+                        w.sub_start, w.sub_n));
+            T9_LAUNCH_CHECK();
+            if (novr9 == 0) return T9_OK;
+            if (novr9 > 64)
+                return HAS_VAL ? t9i_sort_pairs_lsd(ctx, d_keys, d_vals, n,
+                                                    d_workspace, stream)
+                               : t9i_sort_keys_lsd(ctx, d_keys, n,
+                                                   d_workspace, stream);
+            std::vector<u32> list9(novr9), st9(novr9), cn9(novr9);
+            HIP_TRY(hipMemcpy(list9.data(), w.ovr + 2, novr9 * 4,
+                              hipMemcpyDeviceToHost));
+            for (u32 i = 0; i < novr9; ++i) {
+                HIP_TRY(hipMemcpy(&st9[i], w.sub_start + list9[i], 4,
+                                  hipMemcpyDeviceToHost));
+                HIP_TRY(hipMemcpy(&cn9[i], w.sub_n + list9[i], 4,
+                                  hipMemcpyDeviceToHost));
+            }
+            for (u32 i = 0; i < novr9; ++i) {
+                int rc = HAS_VAL
+                             ? t9i_sort_pairs_lsd(ctx, d_keys + st9[i],
+                                                  d_vals + st9[i], cn9[i],
+                                                  d_workspace, stream)
+                             : t9i_sort_keys_lsd(ctx, d_keys + st9[i],
+                                                 cn9[i], d_workspace,
+                                                 stream);
+                if (rc) return rc;
+            }
+            return T9_OK;
+        }
     }
 
     /* ---- pass 2: byte 6, segmented per bucket (alt -> input, compact
