@@ -1,0 +1,154 @@
+/* t9_bitonic.h — register/shfl bitonic alternative to k_lds_sort_sub
+ * (included by t9_sort_msb.hip; selected with T9_LDS_BITONIC=1).
+ *
+ * Motivation (profiles/r01_pmc_wavecycle_decomposition.txt): the 6-pass
+ * LDS counting sort is 58% wave-parked on its ~38 block barriers with
+ * only 4% LDS-conflict stall, so the fix is structural: a bitonic
+ * network runs 52 of its 55 compare-exchange stages entirely in
+ * registers/crosslane (no barrier, no LDS), leaving only the j>=256
+ * stages (3/6/10 for SUBMAX 1024/2048/4096) to stage through LDS.
+ *
+ * Stability: bitonic networks are not stable, so the network sorts the
+ * composite c = (key & mask48) << 12 | pos (pos = load index < 4096).
+ * Composites are unique, ties in low-48 resolve to input order — the
+ * exact order the stable radix passes produce. The high 16 key bits are
+ * shared by every element of a sub-bucket (byte-7 + level-2 digit), so
+ * the key is reconstructed as shared_high | (c >> 12) and need not be
+ * carried through the network.
+ *
+ * Element mapping: e = tid*4 + r (4 registers per thread, wave w owns
+ * the contiguous run [w*256, (w+1)*256)). CE distance j=1,2: register
+ * pair inside the lane; j=4..128: __shfl_xor at lane distance j/4;
+ * j>=256: LDS exchange at wave distance j/256.
+ * Padding: slots >= ns carry c = ~0 (sinks to the end, never written
+ * back); pad-vs-pad CEs may duplicate pad payloads, which is harmless
+ * because the composite multiset above ns stays all-max. */
+#pragma once
+#include "t9_common.h"
+
+/* one compare-exchange against a crosslane partner: both sides compute;
+ * take the partner's element iff (mine > partner) == (I am the lower
+ * position in an ascending pair). Composites are unique so c == oc
+ * cannot occur between live elements. */
+__device__ inline void t9_ce_xor(u64& c, u32& v, u32 lanemask, bool lower,
+                                 bool asc) {
+    const u64 oc = __shfl_xor(c, lanemask);
+    const u32 ov = __shfl_xor(v, lanemask);
+    if ((c > oc) == (lower == asc)) { c = oc; v = ov; }
+}
+
+template <int SUBMAX, int BLOCK, bool HAS_VAL>
+__global__ __launch_bounds__(BLOCK, 2) void k_bitonic_sort_sub(
+    u64* __restrict__ keys, u32* __restrict__ vals,
+    const u32* __restrict__ sub_start, const u32* __restrict__ sub_n) {
+    constexpr int E = SUBMAX / BLOCK;        /* 4 */
+    static_assert(E == 4, "mapping assumes 4 elements per thread");
+    __shared__ u64 s_c[SUBMAX];
+    __shared__ u32 s_v[HAS_VAL ? SUBMAX : 1];
+    __shared__ u32 s_differ;
+
+    const u32 sb = blockIdx.x;
+    const u32 ns = sub_n[sb];
+    if (ns <= 1 || ns > (u32)SUBMAX) return;
+    const u32 gbase = sub_start[sb];
+    const u32 tid = threadIdx.x, lane = tid & 63;
+
+    if (tid == 0) s_differ = 0;
+    __syncthreads();
+
+    const u64 mask48 = 0x0000FFFFFFFFFFFFull;
+    const u64 high16 = keys[gbase] & ~mask48;
+    const u64 ref48 = keys[gbase] & mask48;
+
+    /* load: 4 contiguous elements per thread -> composite registers */
+    u64 c[E];
+    u32 v[E];
+    u32 differ = 0;
+    const u32 e0 = tid * E;
+#pragma unroll
+    for (int r = 0; r < E; ++r) {
+        const u32 i = e0 + r;
+        if (i < ns) {
+            const u64 k = keys[gbase + i];
+            c[r] = ((k & mask48) << 12) | (u64)i;
+            if (HAS_VAL) v[r] = vals[gbase + i];
+            differ |= ((k & mask48) != ref48);
+        } else {
+            c[r] = ~0ull;
+            if (HAS_VAL) v[r] = 0;
+        }
+    }
+    if (differ) s_differ = 1;
+    __syncthreads();
+    if (!s_differ) return;   /* all-equal low-48: stable order is in place */
+
+    /* bitonic network over SUBMAX elements — fully unrolled (lk/lj are
+     * compile-time constants, so lane masks become immediates and the
+     * per-stage branch selection disappears) */
+    constexpr int LOG = (SUBMAX == 1024) ? 10 : (SUBMAX == 2048) ? 11 : 12;
+#pragma unroll
+    for (int lk = 1; lk <= LOG; ++lk) {
+        const u32 k = 1u << lk;
+#pragma unroll
+        for (int lj = lk - 1; lj >= 0; --lj) {
+            const u32 j = 1u << lj;
+            if (j >= 256) {
+                /* cross-wave: stage through LDS */
+                __syncthreads();   /* WAR: prior reads of s_c complete */
+#pragma unroll
+                for (int r = 0; r < E; ++r) {
+                    s_c[e0 + r] = c[r];
+                    if (HAS_VAL) s_v[e0 + r] = v[r];
+                }
+                __syncthreads();
+#pragma unroll
+                for (int r = 0; r < E; ++r) {
+                    const u32 e = e0 + r;
+                    const u32 p = e ^ j;
+                    const u64 oc = s_c[p];
+                    const bool lower = (e & j) == 0;
+                    const bool asc = (e & k) == 0;
+                    if ((c[r] > oc) == (lower == asc)) {
+                        c[r] = oc;
+                        if (HAS_VAL) v[r] = s_v[p];
+                    }
+                }
+            } else if (j >= 4) {
+                /* crosslane within the wave: lane distance j/4 */
+                const u32 lm = j >> 2;
+#pragma unroll
+                for (int r = 0; r < E; ++r) {
+                    const u32 e = e0 + r;
+                    t9_ce_xor(c[r], v[r], lm, (lane & lm) == 0,
+                              (e & k) == 0);
+                }
+            } else {
+                /* register pair inside the lane: r ^ j (j = 1 or 2) */
+#pragma unroll
+                for (int r = 0; r < E; ++r) {
+                    const int q = r ^ (int)j;
+                    if (q > r) {
+                        const bool asc = ((e0 + r) & k) == 0;
+                        if ((c[r] > c[q]) == asc) {
+                            u64 tc = c[r]; c[r] = c[q]; c[q] = tc;
+                            if (HAS_VAL) {
+                                u32 tv = v[r]; v[r] = v[q]; v[q] = tv;
+                            }
+                        }
+                    }
+                }
+            }
+        }
+    }
+
+    /* write back: reconstruct keys from composites; element e0+r of the
+     * sorted order lives in register r of thread tid */
+#pragma unroll
+    for (int r = 0; r < E; ++r) {
+        const u32 i = e0 + r;
+        if (i < ns) {
+            keys[gbase + i] = high16 | (c[r] >> 12);
+            if (HAS_VAL) vals[gbase + i] = v[r];
+        }
+    }
+}

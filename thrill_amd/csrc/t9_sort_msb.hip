@@ -25,6 +25,7 @@
 
 #include "t9_common.h"
 #include "t9_rank_scatter.h"
+#include "t9_bitonic.h"
 
 #include <cstdlib>
 #include <vector>
@@ -555,6 +556,32 @@ __global__ __launch_bounds__(256) void k_subinfo(
     if (threadIdx.x == 0 && s_max) atomicMax(&info[0], s_max);
 }
 
+static bool t9i_lds_bitonic() {
+    static const bool on = [] {
+        const char* e = getenv("T9_LDS_BITONIC");
+        return e && e[0] == '1';
+    }();
+    return on;
+}
+
+template <bool HAS_VAL>
+static void t9i_launch_bitonic_sub(u32 grid, u32 maxsub, hipStream_t s,
+                                   u64* d_keys, u32* d_vals,
+                                   const u32* sub_start, const u32* sub_n) {
+    if (maxsub <= 1024)
+        hipLaunchKernelGGL((k_bitonic_sort_sub<1024, 256, HAS_VAL>),
+                           dim3(grid), dim3(256), 0, s, d_keys, d_vals,
+                           sub_start, sub_n);
+    else if (maxsub <= 2048)
+        hipLaunchKernelGGL((k_bitonic_sort_sub<2048, 512, HAS_VAL>),
+                           dim3(grid), dim3(512), 0, s, d_keys, d_vals,
+                           sub_start, sub_n);
+    else
+        hipLaunchKernelGGL((k_bitonic_sort_sub<4096, 1024, HAS_VAL>),
+                           dim3(grid), dim3(1024), 0, s, d_keys, d_vals,
+                           sub_start, sub_n);
+}
+
 /* level 3: sort one sub-bucket (<= SUBMAX pairs) in LDS over the low
  * 48 key bits — 6 stable ballot-ranked passes, then write back.
  * <2048,512>: 70 KB LDS, 2 blocks/CU (cross-block overlap hides the
@@ -943,7 +970,11 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
             const u32 maxsub = info[0], novr9 = info[1];
             T9_PERF_WRAP(
                 s, "lds_sort",
-                if (maxsub <= 1024)
+                if (t9i_lds_bitonic())
+                    t9i_launch_bitonic_sub<HAS_VAL>((u32)NSUB9, maxsub, s,
+                                                    d_keys, d_vals,
+                                                    w.sub_start, w.sub_n);
+                else if (maxsub <= 1024)
                     hipLaunchKernelGGL(
                         (k_lds_sort_sub<1024, 256, HAS_VAL>), dim3(NSUB9),
                         dim3(256), 0, s, d_keys, d_vals, w.sub_start,
@@ -1160,7 +1191,11 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
         novr = info[1];
         T9_PERF_WRAP(
             s, "lds_sort",
-            if (maxsub <= 2048)
+            if (t9i_lds_bitonic())
+                t9i_launch_bitonic_sub<HAS_VAL>((u32)NSUB, maxsub, s,
+                                                d_keys, d_vals,
+                                                w.sub_start, w.sub_n);
+            else if (maxsub <= 2048)
                 hipLaunchKernelGGL((k_lds_sort_sub<2048, 512, HAS_VAL>),
                                    dim3(NSUB), dim3(512), 0, s, d_keys,
                                    d_vals, w.sub_start, w.sub_n);
