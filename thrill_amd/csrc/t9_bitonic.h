@@ -152,3 +152,123 @@ __global__ __launch_bounds__(BLOCK, 2) void k_bitonic_sort_sub(
         }
     }
 }
+
+/* ------------------------------------------------------------------ *
+ * wave16 blocksort (T9_LDS_WAVE16=1): single-wave block, 16 elements
+ * per lane. The in-lane sort is a pure-VALU bitonic network over
+ * registers (zero LDS traffic where the radix sort spends 4 of its 6
+ * passes), then 6 merge rounds (16->1024) via merge-path splits into
+ * LDS: ~2 LDS writes + ~3 LDS reads per element per round. With a
+ * 64-thread block every __syncthreads() compiles to a wave-level
+ * scheduling barrier (no s_barrier cost). Same stable-composite scheme
+ * as k_bitonic_sort_sub above.
+ * ------------------------------------------------------------------ */
+template <bool HAS_VAL>
+__global__ __launch_bounds__(64, 2) void k_wave16_sort_sub(
+    u64* __restrict__ keys, u32* __restrict__ vals,
+    const u32* __restrict__ sub_start, const u32* __restrict__ sub_n) {
+    constexpr int E = 16;
+    constexpr int NSORT = 1024;
+    __shared__ u64 lc[NSORT];
+    __shared__ u32 lv[HAS_VAL ? NSORT : 1];
+    __shared__ u32 s_differ;
+
+    const u32 sb = blockIdx.x;
+    const u32 ns = sub_n[sb];
+    if (ns <= 1 || ns > (u32)NSORT) return;
+    const u32 gbase = sub_start[sb];
+    const u32 lane = threadIdx.x;
+    const u32 e0 = lane * E;
+
+    if (lane == 0) s_differ = 0;
+    __syncthreads();
+
+    const u64 mask48 = 0x0000FFFFFFFFFFFFull;
+    const u64 high16 = keys[gbase] & ~mask48;
+    const u64 ref48 = keys[gbase] & mask48;
+
+    u64 c[E];
+    u32 v[E];
+    u32 differ = 0;
+#pragma unroll
+    for (int r = 0; r < E; ++r) {
+        const u32 i = e0 + r;
+        if (i < ns) {
+            const u64 k = keys[gbase + i];
+            c[r] = ((k & mask48) << 12) | (u64)i;
+            if (HAS_VAL) v[r] = vals[gbase + i];
+            differ |= ((k & mask48) != ref48);
+        } else {
+            c[r] = ~0ull;
+            if (HAS_VAL) v[r] = 0;
+        }
+    }
+    if (differ) s_differ = 1;
+    __syncthreads();
+    if (!s_differ) return;
+
+    /* in-register bitonic sort of the lane's 16 elements (pure VALU) */
+#pragma unroll
+    for (int lk = 1; lk <= 4; ++lk) {
+        const int k = 1 << lk;
+#pragma unroll
+        for (int lj = lk - 1; lj >= 0; --lj) {
+            const int j = 1 << lj;
+#pragma unroll
+            for (int r = 0; r < E; ++r) {
+                const int q = r ^ j;
+                if (q > r) {
+                    const bool asc = (r & k) == 0;
+                    if ((c[r] > c[q]) == asc) {
+                        u64 tc = c[r]; c[r] = c[q]; c[q] = tc;
+                        if (HAS_VAL) { u32 tv = v[r]; v[r] = v[q]; v[q] = tv; }
+                    }
+                }
+            }
+        }
+    }
+
+    /* 6 merge rounds: sorted runs of L pairwise -> 2L via merge-path.
+     * Composites are unique, so `<=` against the B run is a stable merge
+     * (A-run elements precede equal... equality cannot occur). */
+#pragma unroll
+    for (int lm = 0; lm < 6; ++lm) {
+        const u32 L = (u32)E << lm;
+        __syncthreads();
+#pragma unroll
+        for (int r = 0; r < E; ++r) {
+            lc[e0 + r] = c[r];
+            if (HAS_VAL) lv[e0 + r] = v[r];
+        }
+        __syncthreads();
+        const u32 pairbase = e0 & ~(2 * L - 1);
+        const u32 d = e0 - pairbase;          /* my first output diagonal */
+        const u64* A = lc + pairbase;
+        const u64* B = lc + pairbase + L;
+        u32 lo = (d > L) ? d - L : 0;
+        u32 hi = (d < L) ? d : L;
+        while (lo < hi) {
+            const u32 m = (lo + hi) >> 1;
+            if (A[m] <= B[d - m - 1]) lo = m + 1;
+            else hi = m;
+        }
+        u32 i = lo, j = d - lo;
+#pragma unroll
+        for (int r = 0; r < E; ++r) {
+            const bool ta = (j >= L) || (i < L && A[i] <= B[j]);
+            const u32 src = ta ? i : L + j;
+            c[r] = (ta ? A[i] : B[j]);
+            if (HAS_VAL) v[r] = lv[pairbase + src];
+            if (ta) ++i; else ++j;
+        }
+    }
+
+#pragma unroll
+    for (int r = 0; r < E; ++r) {
+        const u32 i = e0 + r;
+        if (i < ns) {
+            keys[gbase + i] = high16 | (c[r] >> 12);
+            if (HAS_VAL) vals[gbase + i] = v[r];
+        }
+    }
+}

@@ -556,6 +556,17 @@ __global__ __launch_bounds__(256) void k_subinfo(
     if (threadIdx.x == 0 && s_max) atomicMax(&info[0], s_max);
 }
 
+/* wave16 blocksort is the default for sub-buckets <= 1024 (measured
+ * 2.44 vs 2.64 ms radix on the bench workload); T9_LDS_WAVE16=0
+ * restores the 6-pass radix LDS sort. */
+static bool t9i_lds_wave16() {
+    static const bool on = [] {
+        const char* e = getenv("T9_LDS_WAVE16");
+        return !(e && e[0] == '0');
+    }();
+    return on;
+}
+
 static bool t9i_lds_bitonic() {
     static const bool on = [] {
         const char* e = getenv("T9_LDS_BITONIC");
@@ -970,7 +981,11 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
             const u32 maxsub = info[0], novr9 = info[1];
             T9_PERF_WRAP(
                 s, "lds_sort",
-                if (t9i_lds_bitonic())
+                if (t9i_lds_wave16() && maxsub <= 1024)
+                    hipLaunchKernelGGL((k_wave16_sort_sub<HAS_VAL>),
+                                       dim3(NSUB9), dim3(64), 0, s, d_keys,
+                                       d_vals, w.sub_start, w.sub_n);
+                else if (t9i_lds_bitonic())
                     t9i_launch_bitonic_sub<HAS_VAL>((u32)NSUB9, maxsub, s,
                                                     d_keys, d_vals,
                                                     w.sub_start, w.sub_n);
@@ -1191,7 +1206,11 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
         novr = info[1];
         T9_PERF_WRAP(
             s, "lds_sort",
-            if (t9i_lds_bitonic())
+            if (t9i_lds_wave16() && maxsub <= 1024)
+                hipLaunchKernelGGL((k_wave16_sort_sub<HAS_VAL>), dim3(NSUB),
+                                   dim3(64), 0, s, d_keys, d_vals,
+                                   w.sub_start, w.sub_n);
+            else if (t9i_lds_bitonic())
                 t9i_launch_bitonic_sub<HAS_VAL>((u32)NSUB, maxsub, s,
                                                 d_keys, d_vals,
                                                 w.sub_start, w.sub_n);
