@@ -163,12 +163,13 @@ __global__ __launch_bounds__(BLOCK, 2) void k_bitonic_sort_sub(
  * scheduling barrier (no s_barrier cost). Same stable-composite scheme
  * as k_bitonic_sort_sub above.
  * ------------------------------------------------------------------ */
-template <bool HAS_VAL>
-__global__ __launch_bounds__(64, 2) void k_wave16_sort_sub(
+template <int NSORT, int BLOCK, bool HAS_VAL>
+__global__ __launch_bounds__(BLOCK, 2) void k_wave16_sort_sub(
     u64* __restrict__ keys, u32* __restrict__ vals,
     const u32* __restrict__ sub_start, const u32* __restrict__ sub_n) {
-    constexpr int E = 16;
-    constexpr int NSORT = 1024;
+    constexpr int E = NSORT / BLOCK;
+    static_assert(E == 16, "wave16 mapping: 16 elements per lane");
+    constexpr int ROUNDS = (NSORT == 1024) ? 6 : (NSORT == 2048) ? 7 : 8;
     __shared__ u64 lc[NSORT];
     __shared__ u32 lv[HAS_VAL ? NSORT : 1];
     __shared__ u32 s_differ;
@@ -177,7 +178,7 @@ __global__ __launch_bounds__(64, 2) void k_wave16_sort_sub(
     const u32 ns = sub_n[sb];
     if (ns <= 1 || ns > (u32)NSORT) return;
     const u32 gbase = sub_start[sb];
-    const u32 lane = threadIdx.x;
+    const u32 lane = threadIdx.x;   /* block-level thread id */
     const u32 e0 = lane * E;
 
     if (lane == 0) s_differ = 0;
@@ -232,7 +233,7 @@ __global__ __launch_bounds__(64, 2) void k_wave16_sort_sub(
      * Composites are unique, so `<=` against the B run is a stable merge
      * (A-run elements precede equal... equality cannot occur). */
 #pragma unroll
-    for (int lm = 0; lm < 6; ++lm) {
+    for (int lm = 0; lm < ROUNDS; ++lm) {
         const u32 L = (u32)E << lm;
         __syncthreads();
 #pragma unroll
@@ -269,6 +270,142 @@ __global__ __launch_bounds__(64, 2) void k_wave16_sort_sub(
         if (i < ns) {
             keys[gbase + i] = high16 | (c[r] >> 12);
             if (HAS_VAL) vals[gbase + i] = v[r];
+        }
+    }
+}
+
+/* wave16 span sort (3-level path, PASSES=6 spans): same blocksort as
+ * k_wave16_sort_sub, but out-of-place over span descriptors. Valid for
+ * the 6-pass span case only: spans pack (b7,b6,b5) groups that share
+ * their top 16 key bits, so the 60-bit stable composite applies.
+ * Oversize spans copy through (the ranged-LSD fallback re-sorts them),
+ * mirroring k_lds_sort_span. in == out is safe: a block barrier
+ * separates every global read from the first global write. */
+template <int NSORT, int BLOCK, bool HAS_VAL>
+__global__ __launch_bounds__(BLOCK, 2) void k_wave16_sort_span(
+    const u64* __restrict__ keys_in, const u32* __restrict__ vals_in,
+    u64* __restrict__ keys_out, u32* __restrict__ vals_out,
+    const u32* __restrict__ span_start, const u32* __restrict__ span_len) {
+    constexpr int E = NSORT / BLOCK;
+    static_assert(E == 16, "wave16 mapping: 16 elements per lane");
+    constexpr int ROUNDS = (NSORT == 1024) ? 6 : (NSORT == 2048) ? 7 : 8;
+    __shared__ u64 lc[NSORT];
+    __shared__ u32 lv[HAS_VAL ? NSORT : 1];
+    __shared__ u32 s_differ;
+
+    const u32 sb = blockIdx.x;
+    const u32 ns = span_len[sb];
+    if (ns == 0) return;
+    const u32 gbase = span_start[sb];
+    const u32 lane = threadIdx.x;
+    const u32 e0 = lane * E;
+
+    if (ns > (u32)NSORT) {
+        if (keys_out != keys_in)
+            for (u32 i = lane; i < ns; i += BLOCK) {
+                keys_out[gbase + i] = keys_in[gbase + i];
+                if (HAS_VAL) vals_out[gbase + i] = vals_in[gbase + i];
+            }
+        return;
+    }
+
+    if (lane == 0) s_differ = 0;
+    __syncthreads();
+
+    const u64 mask48 = 0x0000FFFFFFFFFFFFull;
+    const u64 high16 = keys_in[gbase] & ~mask48;
+    const u64 ref48 = keys_in[gbase] & mask48;
+
+    u64 c[E];
+    u32 v[E];
+    u32 differ = 0;
+#pragma unroll
+    for (int r = 0; r < E; ++r) {
+        const u32 i = e0 + r;
+        if (i < ns) {
+            const u64 k = keys_in[gbase + i];
+            c[r] = ((k & mask48) << 12) | (u64)i;
+            if (HAS_VAL) v[r] = vals_in[gbase + i];
+            differ |= ((k & mask48) != ref48);
+        } else {
+            c[r] = ~0ull;
+            if (HAS_VAL) v[r] = 0;
+        }
+    }
+    if (differ) s_differ = 1;
+    __syncthreads();
+    if (!s_differ) {
+        if (keys_out != keys_in) {
+#pragma unroll
+            for (int r = 0; r < E; ++r) {
+                const u32 i = e0 + r;
+                if (i < ns) {
+                    keys_out[gbase + i] = high16 | ((c[r] >> 12) & mask48);
+                    if (HAS_VAL) vals_out[gbase + i] = v[r];
+                }
+            }
+        }
+        return;
+    }
+
+#pragma unroll
+    for (int lk = 1; lk <= 4; ++lk) {
+        const int k = 1 << lk;
+#pragma unroll
+        for (int lj = lk - 1; lj >= 0; --lj) {
+            const int j = 1 << lj;
+#pragma unroll
+            for (int r = 0; r < E; ++r) {
+                const int q = r ^ j;
+                if (q > r) {
+                    const bool asc = (r & k) == 0;
+                    if ((c[r] > c[q]) == asc) {
+                        u64 tc = c[r]; c[r] = c[q]; c[q] = tc;
+                        if (HAS_VAL) { u32 tv = v[r]; v[r] = v[q]; v[q] = tv; }
+                    }
+                }
+            }
+        }
+    }
+
+#pragma unroll
+    for (int lm = 0; lm < ROUNDS; ++lm) {
+        const u32 L = (u32)E << lm;
+        __syncthreads();
+#pragma unroll
+        for (int r = 0; r < E; ++r) {
+            lc[e0 + r] = c[r];
+            if (HAS_VAL) lv[e0 + r] = v[r];
+        }
+        __syncthreads();
+        const u32 pairbase = e0 & ~(2 * L - 1);
+        const u32 d = e0 - pairbase;
+        const u64* A = lc + pairbase;
+        const u64* B = lc + pairbase + L;
+        u32 lo = (d > L) ? d - L : 0;
+        u32 hi = (d < L) ? d : L;
+        while (lo < hi) {
+            const u32 m = (lo + hi) >> 1;
+            if (A[m] <= B[d - m - 1]) lo = m + 1;
+            else hi = m;
+        }
+        u32 i = lo, j = d - lo;
+#pragma unroll
+        for (int r = 0; r < E; ++r) {
+            const bool ta = (j >= L) || (i < L && A[i] <= B[j]);
+            const u32 src = ta ? i : L + j;
+            c[r] = (ta ? A[i] : B[j]);
+            if (HAS_VAL) v[r] = lv[pairbase + src];
+            if (ta) ++i; else ++j;
+        }
+    }
+
+#pragma unroll
+    for (int r = 0; r < E; ++r) {
+        const u32 i = e0 + r;
+        if (i < ns) {
+            keys_out[gbase + i] = high16 | (c[r] >> 12);
+            if (HAS_VAL) vals_out[gbase + i] = v[r];
         }
     }
 }

@@ -567,6 +567,45 @@ static bool t9i_lds_wave16() {
     return on;
 }
 
+/* largest sub-bucket size routed to wave16 (tiers 1024/2048/4096);
+ * measured tier by tier before raising the default */
+static u32 t9i_wave16_max() {
+    static const u32 v = [] {
+        const char* e = getenv("T9_WAVE16_MAX");
+        return e ? (u32)atoi(e) : 1024u;
+    }();
+    return v;
+}
+
+/* span wave16 is default-on (68.3 vs 69.5 ms at 2^30): spans are packed
+ * nearly full so the fixed-size network wastes no padding work.
+ * T9_SPAN_WAVE16=0 restores the 6-pass radix span sort. */
+static bool t9i_span_wave16() {
+    static const bool on = [] {
+        const char* e = getenv("T9_SPAN_WAVE16");
+        return !(e && e[0] == '0');
+    }();
+    return on;
+}
+
+template <bool HAS_VAL>
+static void t9i_launch_wave16_sub(u32 grid, u32 maxsub, hipStream_t s,
+                                  u64* d_keys, u32* d_vals,
+                                  const u32* sub_start, const u32* sub_n) {
+    if (maxsub <= 1024)
+        hipLaunchKernelGGL((k_wave16_sort_sub<1024, 64, HAS_VAL>),
+                           dim3(grid), dim3(64), 0, s, d_keys, d_vals,
+                           sub_start, sub_n);
+    else if (maxsub <= 2048)
+        hipLaunchKernelGGL((k_wave16_sort_sub<2048, 128, HAS_VAL>),
+                           dim3(grid), dim3(128), 0, s, d_keys, d_vals,
+                           sub_start, sub_n);
+    else
+        hipLaunchKernelGGL((k_wave16_sort_sub<4096, 256, HAS_VAL>),
+                           dim3(grid), dim3(256), 0, s, d_keys, d_vals,
+                           sub_start, sub_n);
+}
+
 static bool t9i_lds_bitonic() {
     static const bool on = [] {
         const char* e = getenv("T9_LDS_BITONIC");
@@ -981,10 +1020,10 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
             const u32 maxsub = info[0], novr9 = info[1];
             T9_PERF_WRAP(
                 s, "lds_sort",
-                if (t9i_lds_wave16() && maxsub <= 1024)
-                    hipLaunchKernelGGL((k_wave16_sort_sub<HAS_VAL>),
-                                       dim3(NSUB9), dim3(64), 0, s, d_keys,
-                                       d_vals, w.sub_start, w.sub_n);
+                if (t9i_lds_wave16() && maxsub <= t9i_wave16_max())
+                    t9i_launch_wave16_sub<HAS_VAL>((u32)NSUB9, maxsub, s,
+                                                   d_keys, d_vals,
+                                                   w.sub_start, w.sub_n);
                 else if (t9i_lds_bitonic())
                     t9i_launch_bitonic_sub<HAS_VAL>((u32)NSUB9, maxsub, s,
                                                     d_keys, d_vals,
@@ -1115,7 +1154,17 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
         const u32 novr3 = hdr[1];
         T9_PERF_WRAP(
             s, "lds_sort",
-            if (spanmax == 2048)
+            if (t9i_span_wave16() && spanmax == 2048)
+                hipLaunchKernelGGL(
+                    (k_wave16_sort_span<2048, 128, HAS_VAL>),
+                    dim3(nspan ? nspan : 1), dim3(128), 0, s, w.alt_k,
+                    w.alt_v, d_keys, d_vals, s3_start, s3_len);
+            else if (t9i_span_wave16())
+                hipLaunchKernelGGL(
+                    (k_wave16_sort_span<4096, 256, HAS_VAL>),
+                    dim3(nspan ? nspan : 1), dim3(256), 0, s, w.alt_k,
+                    w.alt_v, d_keys, d_vals, s3_start, s3_len);
+            else if (spanmax == 2048)
                 hipLaunchKernelGGL(
                     (k_lds_sort_span<2048, 512, HAS_VAL, 6>),
                     dim3(nspan ? nspan : 1), dim3(512), 0, s, w.alt_k,
@@ -1206,10 +1255,10 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
         novr = info[1];
         T9_PERF_WRAP(
             s, "lds_sort",
-            if (t9i_lds_wave16() && maxsub <= 1024)
-                hipLaunchKernelGGL((k_wave16_sort_sub<HAS_VAL>), dim3(NSUB),
-                                   dim3(64), 0, s, d_keys, d_vals,
-                                   w.sub_start, w.sub_n);
+            if (t9i_lds_wave16() && maxsub <= t9i_wave16_max())
+                t9i_launch_wave16_sub<HAS_VAL>((u32)NSUB, maxsub, s,
+                                               d_keys, d_vals,
+                                               w.sub_start, w.sub_n);
             else if (t9i_lds_bitonic())
                 t9i_launch_bitonic_sub<HAS_VAL>((u32)NSUB, maxsub, s,
                                                 d_keys, d_vals,
