@@ -109,10 +109,16 @@ __global__ __launch_bounds__(256) void k_reduce_build(
 __global__ __launch_bounds__(256) void k_reduce_build_lds(
     const u64* __restrict__ keys, const u64* __restrict__ vals, u64 n,
     u64* __restrict__ tk, u64* __restrict__ tv, u64 cap, u64 salt,
-    u32* __restrict__ err) {
+    u32* __restrict__ err, int wavecomb) {
     __shared__ u64 lk[T9_LDS_SLOTS];
     __shared__ u64 lv[T9_LDS_SLOTS];
+    /* wave-combine scratch: per-lane (key, val) visible to the wave's
+     * group leaders via plain LDS reads (per-wave-disjoint index ranges,
+     * no barrier needed) */
+    __shared__ u64 sc_k[256];
+    __shared__ u64 sc_v[256];
     const u32 tid = threadIdx.x;
+    const u32 lane = tid & 63, wbase = tid & ~63u;
     for (u32 s = tid; s < T9_LDS_SLOTS; s += 256) {
         lk[s] = T9_EMPTY;
         lv[s] = 0;
@@ -120,17 +126,55 @@ __global__ __launch_bounds__(256) void k_reduce_build_lds(
     __syncthreads();
 
     const u64 gsz = (u64)gridDim.x * 256;
-    for (u64 i = (u64)blockIdx.x * 256 + tid; i < n; i += gsz) {
-        const u64 k = keys[i];
-        const u64 v = vals[i];
-        if (k == T9_EMPTY) {
+    for (u64 base = (u64)blockIdx.x * 256; base < n; base += gsz) {
+        const u64 i = base + tid;
+        const bool valid = i < n;
+        const u64 k = valid ? keys[i] : 0;
+        const u64 v = valid ? vals[i] : 0;
+        const bool sent = valid && k == T9_EMPTY;
+        if (sent) {
             atomicAdd((unsigned long long*)&tk[cap], 1ull);
             atomicAdd((unsigned long long*)&tv[cap],
                       (unsigned long long)v);
-            continue;
         }
+        const bool live = valid && !sent;
         const u64 h = t9_hash128to64(salt, k);
         u32 ls = (u32)(h >> 48) & (T9_LDS_SLOTS - 1);
+
+        /* wave pre-combine (the Zipf hot path): ballot-match lanes on
+         * the 11-bit LDS slot (VALU only), then the lowest matched lane
+         * absorbs equal-key members through plain-LDS scratch — ONE
+         * atomic round-trip per wave-group instead of one per lane.
+         * Without this the same-slot LDS atomics of a skewed stream
+         * serialize: measured 94.8% SQ_WAIT_ANY
+         * (profiles/r01_pmc_wavecycle_decomposition.txt appendix). */
+        u64 gsum = v;
+        bool lead = live;
+        if (wavecomb) {
+        u64 m = __ballot(live);
+        for (int b = 0; b < 11; ++b) {
+            const u64 bb = __ballot((ls >> b) & 1u);
+            m &= ((ls >> b) & 1u) ? bb : ~bb;
+        }
+        if (live && __popcll(m) > 1) {
+            sc_k[tid] = k;
+            sc_v[tid] = v;
+            const u32 l0 = (u32)__ffsll((unsigned long long)m) - 1;
+            if (lane == l0) {
+                u64 mm = m & ~(1ull << lane);
+                while (mm) {
+                    const u32 b = (u32)__ffsll((unsigned long long)mm) - 1;
+                    mm &= mm - 1;
+                    if (sc_k[wbase + b] == k) gsum += sc_v[wbase + b];
+                }
+            } else if (sc_k[wbase + l0] == k) {
+                lead = false;   /* absorbed by the group leader */
+            }
+        }
+        }
+        if (!lead) continue;
+        if (!live) continue;
+
         bool done = false;
         for (int p = 0; p < 4; ++p) {
             u64 prev = atomicCAS((unsigned long long*)&lk[ls],
@@ -138,7 +182,7 @@ __global__ __launch_bounds__(256) void k_reduce_build_lds(
                                  (unsigned long long)k);
             if (prev == T9_EMPTY || prev == k) {
                 atomicAdd((unsigned long long*)&lv[ls],
-                          (unsigned long long)v);
+                          (unsigned long long)gsum);
                 done = true;
                 break;
             }
@@ -154,7 +198,7 @@ __global__ __launch_bounds__(256) void k_reduce_build_lds(
                                      (unsigned long long)k);
                 if (prev == T9_EMPTY || prev == k) {
                     atomicAdd((unsigned long long*)&tv[slot],
-                              (unsigned long long)v);
+                              (unsigned long long)gsum);
                     break;
                 }
                 slot = (slot + 1) & (cap - 1);
@@ -378,12 +422,19 @@ int t9_reduce_build(t9_context* ctx, const u64* d_keys, const u64* d_vals,
     if (!d_keys || !d_vals) return T9_EINVAL;
     const char* ce = getenv("T9_REDUCE_COMBINE");
     const int mode = ce ? atoi(ce) : 2;   /* 2 = LDS table (default) */
+    /* wave pre-combine is OFF by default: measured SLOWER at every
+     * vocab (1e3: 2.38->3.70 ms; 1e7: 11.06->11.75 ms at 2^28 tokens)
+     * — CDNA4 LDS atomics absorb same-address contention better than
+     * the ballot/scratch/leader overhead costs. Kept opt-in
+     * (T9_REDUCE_WAVECOMB=1) as a recorded negative result. */
+    const char* we = getenv("T9_REDUCE_WAVECOMB");
+    const int wavecomb = (we && we[0] == '1') ? 1 : 0;
     T9_PERF_WRAP(
         s, "reduce_build",
         if (mode == 2)
             hipLaunchKernelGGL(k_reduce_build_lds, dim3(grid_for(n)),
                                dim3(256), 0, s, d_keys, d_vals, n, d_tk,
-                               d_tv, cap, salt, d_error);
+                               d_tv, cap, salt, d_error, wavecomb);
         else
             hipLaunchKernelGGL(k_reduce_build, dim3(grid_for(n)),
                                dim3(256), 0, s, d_keys, d_vals, n, d_tk,
