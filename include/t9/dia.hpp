@@ -297,18 +297,17 @@ inline DIA<KeyValue> ReducePair(const DIA<KeyValue>& input,
     }
     uint64_t cap = 1024;
     while (cap < 2 * n + 2) cap <<= 1;   // no grow/spill: size for 2x
-    DeviceBuf tk((cap + 1) * 8), tv((cap + 1) * 8);
+    DeviceBuf tbl(2 * (cap + 1) * 8);   // interleaved (key, sum) slots
     DeviceBuf ok((cap + 1) * 8), ov((cap + 1) * 8);
     DeviceBuf derr(4), dn(8);
     hipStream_t s = ctx.stream();
-    T9_DIA_TRY(t9_reduce_init(ctx.native(), (uint64_t*)tk.ptr,
-                              (uint64_t*)tv.ptr, cap, s));
+    T9_DIA_TRY(t9_reduce_init(ctx.native(), (uint64_t*)tbl.ptr, cap, s));
     T9_DIA_TRY(t9_reduce_build(ctx.native(), (const uint64_t*)dk.ptr,
                                (const uint64_t*)dv.ptr, n,
-                               (uint64_t*)tk.ptr, (uint64_t*)tv.ptr, cap,
+                               (uint64_t*)tbl.ptr, cap,
                                salt, (uint32_t*)derr.ptr, s));
-    T9_DIA_TRY(t9_reduce_drain(ctx.native(), (const uint64_t*)tk.ptr,
-                               (const uint64_t*)tv.ptr, cap,
+    T9_DIA_TRY(t9_reduce_drain(ctx.native(), (const uint64_t*)tbl.ptr,
+                               cap,
                                (uint64_t*)ok.ptr, (uint64_t*)ov.ptr,
                                (uint64_t*)dn.ptr, s));
     uint64_t m = 0, err = 0;

@@ -174,22 +174,25 @@ int t9_hash_bucket(t9_context* ctx, const uint64_t* d_keys, uint64_t n,
                    uint64_t salt, uint32_t p, uint32_t* d_bucket,
                    uint64_t* d_counts, void* stream);
 
-int t9_reduce_init(t9_context* ctx, uint64_t* d_table_keys,
-                   uint64_t* d_table_vals, uint64_t capacity, void* stream);
+/* The reduce table is ONE interleaved u64 array of 2*(capacity+1)
+ * elements: slot i = (d_table[2i] key, d_table[2i+1] sum), aux sentinel
+ * slot at [2*capacity .. 2*capacity+1]. Interleaving keeps each probing
+ * CAS + add inside one cache line (the build of a big-vocab stream is
+ * random-line bound). capacity must be a power of two. */
+int t9_reduce_init(t9_context* ctx, uint64_t* d_table, uint64_t capacity,
+                   void* stream);
 /* Accumulate n (key, value) pairs into the table; value reduce = u64 add.
  * d_error (device u32, zeroed by the call) is set nonzero if the table
  * overflows. */
 int t9_reduce_build(t9_context* ctx, const uint64_t* d_keys,
                     const uint64_t* d_vals, uint64_t n,
-                    uint64_t* d_table_keys, uint64_t* d_table_vals,
-                    uint64_t capacity, uint64_t salt, uint32_t* d_error,
-                    void* stream);
+                    uint64_t* d_table, uint64_t capacity, uint64_t salt,
+                    uint32_t* d_error, void* stream);
 /* Compact occupied slots to (key, value) arrays (unordered);
  * d_out_n (device u64) receives the count. */
-int t9_reduce_drain(t9_context* ctx, const uint64_t* d_table_keys,
-                    const uint64_t* d_table_vals, uint64_t capacity,
-                    uint64_t* d_out_keys, uint64_t* d_out_vals,
-                    uint64_t* d_out_n, void* stream);
+int t9_reduce_drain(t9_context* ctx, const uint64_t* d_table,
+                    uint64_t capacity, uint64_t* d_out_keys,
+                    uint64_t* d_out_vals, uint64_t* d_out_n, void* stream);
 
 /* ReduceToIndex — reference
  * api/reduce_to_index.hpp + core/reduce_by_index_post_phase.hpp with the

@@ -25,15 +25,15 @@ def nat():
 def run_reduce(nat, keys, vals, cap, salt=0):
     n = len(keys)
     dk, dv = G.dev(keys), G.dev(vals)
-    tk, tv = G.empty(cap + 1, np.uint64), G.empty(cap + 1, np.uint64)
+    tbl = G.empty(2 * (cap + 1), np.uint64)
     derr = G.empty(1, np.uint32)
     dn = G.empty(1, np.uint64)
     ok, ov = G.empty(cap + 1, np.uint64), G.empty(cap + 1, np.uint64)
     s = G.stream()
-    nat.reduce_init(G.ptr(tk), G.ptr(tv), cap, s)
-    nat.reduce_build(G.ptr(dk), G.ptr(dv), n, G.ptr(tk), G.ptr(tv), cap,
+    nat.reduce_init(G.ptr(tbl), cap, s)
+    nat.reduce_build(G.ptr(dk), G.ptr(dv), n, G.ptr(tbl), cap,
                      salt, G.ptr(derr), s)
-    nat.reduce_drain(G.ptr(tk), G.ptr(tv), cap, G.ptr(ok), G.ptr(ov),
+    nat.reduce_drain(G.ptr(tbl), cap, G.ptr(ok), G.ptr(ov),
                      G.ptr(dn), s)
     assert int(G.host(derr, np.uint32)[0]) == 0, "table overflow"
     m = int(G.host(dn, np.uint64)[0])
@@ -197,11 +197,11 @@ def test_reduce_overflow_sets_error(nat):
     keys = np.arange(1, n + 1, dtype=np.uint64)
     vals = np.ones(n, dtype=np.uint64)
     dk, dv = G.dev(keys), G.dev(vals)
-    tk, tv = G.empty(cap + 1, np.uint64), G.empty(cap + 1, np.uint64)
+    tbl = G.empty(2 * (cap + 1), np.uint64)
     derr = G.empty(1, np.uint32)
     s = G.stream()
-    nat.reduce_init(G.ptr(tk), G.ptr(tv), cap, s)
-    nat.reduce_build(G.ptr(dk), G.ptr(dv), n, G.ptr(tk), G.ptr(tv), cap,
+    nat.reduce_init(G.ptr(tbl), cap, s)
+    nat.reduce_build(G.ptr(dk), G.ptr(dv), n, G.ptr(tbl), cap,
                      0, G.ptr(derr), s)
     assert int(G.host(derr, np.uint32)[0]) == 1
 

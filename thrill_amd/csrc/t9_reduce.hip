@@ -23,20 +23,25 @@
 
 #define T9_EMPTY 0xFFFFFFFFFFFFFFFFull
 
-__global__ __launch_bounds__(256) void k_reduce_init(u64* __restrict__ tk,
-                                                     u64* __restrict__ tv,
+/* The table is ONE interleaved array t[2*(cap+1)]: slot i = (t[2i] key,
+ * t[2i+1] sum), aux sentinel slot at t[2cap..2cap+1]. Interleaving keeps
+ * each cold-path CAS+add inside ONE 128-B line: with split key/val
+ * arrays every cold token of a big-vocab stream paid two random HBM
+ * lines, and the build kernel parked 94.8% of its wave cycles on them
+ * (profiles/r01_pmc_wavecycle_decomposition.txt appendix). */
+__global__ __launch_bounds__(256) void k_reduce_init(u64* __restrict__ t,
                                                      u64 cap) {
     const u64 stride = (u64)gridDim.x * 256;
     for (u64 i = (u64)blockIdx.x * 256 + threadIdx.x; i <= cap; i += stride) {
-        tk[i] = (i == cap) ? 0 : T9_EMPTY;  /* aux slot counts sentinel-key
-                                               occurrences */
-        tv[i] = 0;
+        t[2 * i] = (i == cap) ? 0 : T9_EMPTY;  /* aux slot counts
+                                                  sentinel-key occurrences */
+        t[2 * i + 1] = 0;
     }
 }
 
 __global__ __launch_bounds__(256) void k_reduce_build(
     const u64* __restrict__ keys, const u64* __restrict__ vals, u64 n,
-    u64* __restrict__ tk, u64* __restrict__ tv, u64 cap, u64 salt,
+    u64* __restrict__ t, u64 cap, u64 salt,
     u32* __restrict__ err, int combine) {
     const u64 gsz = (u64)gridDim.x * 256;
     const u32 lane = threadIdx.x & 63;
@@ -70,20 +75,20 @@ __global__ __launch_bounds__(256) void k_reduce_build(
         }
         if (leader) {
             if (k == T9_EMPTY) {
-                atomicAdd((unsigned long long*)&tk[cap],
+                atomicAdd((unsigned long long*)&t[2 * cap],
                           (unsigned long long)cnt_same);
-                atomicAdd((unsigned long long*)&tv[cap],
+                atomicAdd((unsigned long long*)&t[2 * cap + 1],
                           (unsigned long long)gsum);
             }
             else {
                 u64 slot = t9_hash128to64(salt, k) & (cap - 1);
                 u64 probes = 0;
                 for (;;) {
-                    u64 prev = atomicCAS((unsigned long long*)&tk[slot],
+                    u64 prev = atomicCAS((unsigned long long*)&t[2 * slot],
                                          (unsigned long long)T9_EMPTY,
                                          (unsigned long long)k);
                     if (prev == T9_EMPTY || prev == k) {
-                        atomicAdd((unsigned long long*)&tv[slot],
+                        atomicAdd((unsigned long long*)&t[2 * slot + 1],
                                   (unsigned long long)gsum);
                         break;
                     }
@@ -108,7 +113,7 @@ __global__ __launch_bounds__(256) void k_reduce_build(
 #define T9_LDS_SLOTS 2048
 __global__ __launch_bounds__(256) void k_reduce_build_lds(
     const u64* __restrict__ keys, const u64* __restrict__ vals, u64 n,
-    u64* __restrict__ tk, u64* __restrict__ tv, u64 cap, u64 salt,
+    u64* __restrict__ t, u64 cap, u64 salt,
     u32* __restrict__ err, int wavecomb) {
     __shared__ u64 lk[T9_LDS_SLOTS];
     __shared__ u64 lv[T9_LDS_SLOTS];
@@ -133,8 +138,8 @@ __global__ __launch_bounds__(256) void k_reduce_build_lds(
         const u64 v = valid ? vals[i] : 0;
         const bool sent = valid && k == T9_EMPTY;
         if (sent) {
-            atomicAdd((unsigned long long*)&tk[cap], 1ull);
-            atomicAdd((unsigned long long*)&tv[cap],
+            atomicAdd((unsigned long long*)&t[2 * cap], 1ull);
+            atomicAdd((unsigned long long*)&t[2 * cap + 1],
                       (unsigned long long)v);
         }
         const bool live = valid && !sent;
@@ -189,15 +194,16 @@ __global__ __launch_bounds__(256) void k_reduce_build_lds(
             ls = (ls + 1) & (T9_LDS_SLOTS - 1);
         }
         if (!done) {
-            /* cold path: straight to the global table */
+            /* cold path: straight to the global table — CAS and add
+             * land in the SAME 128-B line (interleaved layout) */
             u64 slot = h & (cap - 1);
             u64 probes = 0;
             for (;;) {
-                u64 prev = atomicCAS((unsigned long long*)&tk[slot],
+                u64 prev = atomicCAS((unsigned long long*)&t[2 * slot],
                                      (unsigned long long)T9_EMPTY,
                                      (unsigned long long)k);
                 if (prev == T9_EMPTY || prev == k) {
-                    atomicAdd((unsigned long long*)&tv[slot],
+                    atomicAdd((unsigned long long*)&t[2 * slot + 1],
                               (unsigned long long)gsum);
                     break;
                 }
@@ -219,11 +225,11 @@ __global__ __launch_bounds__(256) void k_reduce_build_lds(
         u64 slot = t9_hash128to64(salt, k) & (cap - 1);
         u64 probes = 0;
         for (;;) {
-            u64 prev = atomicCAS((unsigned long long*)&tk[slot],
+            u64 prev = atomicCAS((unsigned long long*)&t[2 * slot],
                                  (unsigned long long)T9_EMPTY,
                                  (unsigned long long)k);
             if (prev == T9_EMPTY || prev == k) {
-                atomicAdd((unsigned long long*)&tv[slot],
+                atomicAdd((unsigned long long*)&t[2 * slot + 1],
                           (unsigned long long)v);
                 break;
             }
@@ -284,7 +290,7 @@ __global__ __launch_bounds__(256) void k_index_bucket(
  * writes at block-local scanned positions. Output order is arbitrary, as
  * the reference documents for reducing (word_count_test.cpp:73-74). */
 __global__ __launch_bounds__(256) void k_reduce_drain(
-    const u64* __restrict__ tk, const u64* __restrict__ tv, u64 cap,
+    const u64* __restrict__ t, u64 cap,
     u64* __restrict__ ok, u64* __restrict__ ov, u64* __restrict__ out_n) {
     __shared__ u32 s_pre[256];
     __shared__ u64 s_base;
@@ -293,7 +299,7 @@ __global__ __launch_bounds__(256) void k_reduce_drain(
     const u64 gid = (u64)blockIdx.x * 256 + tid;
     u32 mine = 0;
     for (u64 i = gid; i < cap; i += stride)
-        if (tk[i] != T9_EMPTY) ++mine;
+        if (t[2 * i] != T9_EMPTY) ++mine;
     s_pre[tid] = mine;
     __syncthreads();
     for (int off = 1; off < 256; off <<= 1) {
@@ -309,17 +315,17 @@ __global__ __launch_bounds__(256) void k_reduce_drain(
     if (mine) {
         u64 pos = s_base + s_pre[tid] - mine;
         for (u64 i = gid; i < cap; i += stride) {
-            if (tk[i] != T9_EMPTY) {
-                ok[pos] = tk[i];
-                ov[pos] = tv[i];
+            if (t[2 * i] != T9_EMPTY) {
+                ok[pos] = t[2 * i];
+                ov[pos] = t[2 * i + 1];
                 ++pos;
             }
         }
     }
-    if (gid == 0 && tk[cap] > 0) {
+    if (gid == 0 && t[2 * cap] > 0) {
         u64 pos = atomicAdd((unsigned long long*)out_n, 1ull);
         ok[pos] = T9_EMPTY;
-        ov[pos] = tv[cap];
+        ov[pos] = t[2 * cap + 1];
     }
 }
 
@@ -401,21 +407,20 @@ int t9_index_bucket(t9_context* ctx, const u64* d_keys, u64 n, u64 begin,
     return T9_OK;
 }
 
-int t9_reduce_init(t9_context* ctx, u64* d_tk, u64* d_tv, u64 cap,
-                   void* stream) {
+int t9_reduce_init(t9_context* ctx, u64* d_table, u64 cap, void* stream) {
     (void)ctx;
-    if (!d_tk || !d_tv || !is_pow2(cap)) return T9_EINVAL;
+    if (!d_table || !is_pow2(cap)) return T9_EINVAL;
     hipLaunchKernelGGL(k_reduce_init, dim3(grid_for(cap + 1)), dim3(256), 0,
-                       (hipStream_t)stream, d_tk, d_tv, cap);
+                       (hipStream_t)stream, d_table, cap);
     T9_LAUNCH_CHECK();
     return T9_OK;
 }
 
 int t9_reduce_build(t9_context* ctx, const u64* d_keys, const u64* d_vals,
-                    u64 n, u64* d_tk, u64* d_tv, u64 cap, u64 salt,
+                    u64 n, u64* d_table, u64 cap, u64 salt,
                     u32* d_error, void* stream) {
     (void)ctx;
-    if (!d_tk || !d_tv || !d_error || !is_pow2(cap)) return T9_EINVAL;
+    if (!d_table || !d_error || !is_pow2(cap)) return T9_EINVAL;
     hipStream_t s = (hipStream_t)stream;
     HIP_TRY(hipMemsetAsync(d_error, 0, 4, s));
     if (n == 0) return T9_OK;
@@ -433,26 +438,26 @@ int t9_reduce_build(t9_context* ctx, const u64* d_keys, const u64* d_vals,
         s, "reduce_build",
         if (mode == 2)
             hipLaunchKernelGGL(k_reduce_build_lds, dim3(grid_for(n)),
-                               dim3(256), 0, s, d_keys, d_vals, n, d_tk,
-                               d_tv, cap, salt, d_error, wavecomb);
+                               dim3(256), 0, s, d_keys, d_vals, n, d_table,
+                               cap, salt, d_error, wavecomb);
         else
             hipLaunchKernelGGL(k_reduce_build, dim3(grid_for(n)),
-                               dim3(256), 0, s, d_keys, d_vals, n, d_tk,
-                               d_tv, cap, salt, d_error, mode));
+                               dim3(256), 0, s, d_keys, d_vals, n, d_table,
+                               cap, salt, d_error, mode));
     T9_LAUNCH_CHECK();
     return T9_OK;
 }
 
-int t9_reduce_drain(t9_context* ctx, const u64* d_tk, const u64* d_tv,
+int t9_reduce_drain(t9_context* ctx, const u64* d_table,
                     u64 cap, u64* d_ok, u64* d_ov, u64* d_out_n,
                     void* stream) {
     (void)ctx;
-    if (!d_tk || !d_tv || !d_ok || !d_ov || !d_out_n || !is_pow2(cap))
+    if (!d_table || !d_ok || !d_ov || !d_out_n || !is_pow2(cap))
         return T9_EINVAL;
     hipStream_t s = (hipStream_t)stream;
     HIP_TRY(hipMemsetAsync(d_out_n, 0, 8, s));
     hipLaunchKernelGGL(k_reduce_drain, dim3(grid_for(cap)), dim3(256), 0, s,
-                       d_tk, d_tv, cap, d_ok, d_ov, d_out_n);
+                       d_table, cap, d_ok, d_ov, d_out_n);
     T9_LAUNCH_CHECK();
     return T9_OK;
 }

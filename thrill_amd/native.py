@@ -59,9 +59,9 @@ _SIGS = {
     "t9_hash_bucket": (i32, [vp, vp, u64, u64, u32, vp, vp, vp]),
     "t9_reduce_by_index": (i32, [vp, vp, vp, u64, u64, u64, vp, vp, vp]),
     "t9_index_bucket": (i32, [vp, vp, u64, u64, u64, u32, vp, vp, vp]),
-    "t9_reduce_init": (i32, [vp, vp, vp, u64, vp]),
-    "t9_reduce_build": (i32, [vp, vp, vp, u64, vp, vp, u64, u64, vp, vp]),
-    "t9_reduce_drain": (i32, [vp, vp, vp, u64, vp, vp, vp, vp]),
+    "t9_reduce_init": (i32, [vp, vp, u64, vp]),
+    "t9_reduce_build": (i32, [vp, vp, vp, u64, vp, u64, u64, vp, vp]),
+    "t9_reduce_drain": (i32, [vp, vp, u64, vp, vp, vp, vp]),
     "t9_zipf_tokens": (i32, [vp, vp, vp, u64, u64, u64, u64, vp]),
     "t9_merge_u64": (i32, [vp, vp, u64, vp, u64, vp, vp]),
     "t9_group_index_workspace": (u64, [u64]),

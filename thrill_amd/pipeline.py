@@ -92,8 +92,9 @@ class WordCount:
                                  device="cuda")
         cap = 1 << max(10, int(math.ceil(math.log2(2 * vocab + 2))))
         self.cap = cap
-        self.d_tk = torch.empty(cap + 1, dtype=torch.int64, device="cuda")
-        self.d_tv = torch.empty(cap + 1, dtype=torch.int64, device="cuda")
+        # interleaved (key, sum) table: u64[2*(cap+1)]
+        self.d_tbl = torch.empty(2 * (cap + 1), dtype=torch.int64,
+                                 device="cuda")
         self.d_ok = torch.empty(cap + 1, dtype=torch.int64, device="cuda")
         self.d_ov = torch.empty(cap + 1, dtype=torch.int64, device="cuda")
         self.d_err = torch.empty(1, dtype=torch.int32, device="cuda")
@@ -106,11 +107,10 @@ class WordCount:
 
     def _reduce(self, d_keys, d_vals, n, salt=0):
         nat, s = self.nat, _stream()
-        nat.reduce_init(_ptr(self.d_tk), _ptr(self.d_tv), self.cap, s)
-        nat.reduce_build(_ptr(d_keys), _ptr(d_vals), n, _ptr(self.d_tk),
-                         _ptr(self.d_tv), self.cap, salt, _ptr(self.d_err),
-                         s)
-        nat.reduce_drain(_ptr(self.d_tk), _ptr(self.d_tv), self.cap,
+        nat.reduce_init(_ptr(self.d_tbl), self.cap, s)
+        nat.reduce_build(_ptr(d_keys), _ptr(d_vals), n, _ptr(self.d_tbl),
+                         self.cap, salt, _ptr(self.d_err), s)
+        nat.reduce_drain(_ptr(self.d_tbl), self.cap,
                          _ptr(self.d_ok), _ptr(self.d_ov), _ptr(self.d_n),
                          s)
         m = int(self.d_n.cpu().item())
