@@ -110,11 +110,15 @@ __global__ __launch_bounds__(256) void k_reduce_build(
  * its LDS table once at the end. Plays the role the reference's
  * in-cache pre-table plays for its CPU cache
  * (core/reduce_pre_phase.hpp). */
-#define T9_LDS_SLOTS 2048
+/* SLOTS trades hot-key coverage against occupancy (LDS bytes/block):
+ * 1024 -> 8 blocks/CU, 2048 -> 4 (default), 4096 -> 2. Selected with
+ * T9_LDS_SLOTS. */
+template <int SLOTS>
 __global__ __launch_bounds__(256) void k_reduce_build_lds(
     const u64* __restrict__ keys, const u64* __restrict__ vals, u64 n,
     u64* __restrict__ t, u64 cap, u64 salt,
     u32* __restrict__ err, int wavecomb) {
+    constexpr int T9_LDS_SLOTS = SLOTS;
     __shared__ u64 lk[T9_LDS_SLOTS];
     __shared__ u64 lv[T9_LDS_SLOTS];
     /* wave-combine scratch: per-lane (key, val) visible to the wave's
@@ -436,10 +440,37 @@ int t9_reduce_build(t9_context* ctx, const u64* d_keys, const u64* d_vals,
     const int wavecomb = (we && we[0] == '1') ? 1 : 0;
     T9_PERF_WRAP(
         s, "reduce_build",
-        if (mode == 2)
-            hipLaunchKernelGGL(k_reduce_build_lds, dim3(grid_for(n)),
-                               dim3(256), 0, s, d_keys, d_vals, n, d_table,
-                               cap, salt, d_error, wavecomb);
+        if (mode == 2) {
+            /* measured optimum at 10M-vocab Zipf(1.1), 2^29 tokens:
+             * slots=4096 (2 blocks/CU) beats 2048 (21.6 -> 19.1 ms,
+             * coverage over concurrency) and 8192 (1 block/CU, 22.0);
+             * a 1024 grid shaves flush duplication (19.1 -> 18.8). */
+            const char* se = getenv("T9_LDS_SLOTS");
+            const int slots = se ? atoi(se) : 4096;
+            const char* ge = getenv("T9_REDUCE_GRID");
+            u32 grid = ge ? (u32)atoi(ge) : grid_for(n);
+            if (!ge && grid > 1024) grid = 1024;
+            if (slots >= 8192)
+                hipLaunchKernelGGL(k_reduce_build_lds<8192>,
+                                   dim3(grid), dim3(256), 0, s,
+                                   d_keys, d_vals, n, d_table, cap, salt,
+                                   d_error, wavecomb);
+            else if (slots <= 1024)
+                hipLaunchKernelGGL(k_reduce_build_lds<1024>,
+                                   dim3(grid), dim3(256), 0, s,
+                                   d_keys, d_vals, n, d_table, cap, salt,
+                                   d_error, wavecomb);
+            else if (slots >= 4096)
+                hipLaunchKernelGGL(k_reduce_build_lds<4096>,
+                                   dim3(grid), dim3(256), 0, s,
+                                   d_keys, d_vals, n, d_table, cap, salt,
+                                   d_error, wavecomb);
+            else
+                hipLaunchKernelGGL(k_reduce_build_lds<2048>,
+                                   dim3(grid), dim3(256), 0, s,
+                                   d_keys, d_vals, n, d_table, cap, salt,
+                                   d_error, wavecomb);
+        }
         else
             hipLaunchKernelGGL(k_reduce_build, dim3(grid_for(n)),
                                dim3(256), 0, s, d_keys, d_vals, n, d_table,
