@@ -14,12 +14,12 @@ timed region starts; the input buffer is read-only to the pipeline, so
 every step redoes identical work). value = records sorted per second,
 whole-job over all N ranks.
 
-The roofline object tracks the dominant kernel (pair_scatter — the radix
-scatter moving (u64 key, u32 idx) pairs): achieved = algorithmic bytes per
-launch (24 B per record per pass: 12 read + 12 written)
-/ HIP-event-measured average launch duration, against the 8 TB/s HBM3E
-peak. traffic (PMC bytes) is collected by the committed rocprofv3 runs
-under profiles/, not live — reported null here.
+The roofline object tracks the dominant kernel (the payload gather —
+out[i] = rec[idx[i]], ~45% of the step): achieved = algorithmic bytes per
+launch (2R + 4 = 204 B per record) / HIP-event-
+measured average launch duration, against the 8 TB/s HBM3E peak. traffic
+is the PMC byte count per launch from the committed rocprofv3 calibration
+under profiles/ (not measured live — PMC needs its own rocprofv3 pass).
 
 cpu_baseline: the CPU oracle (reference-semantics restatement,
 oracle/t9_oracle.cpp — kind "port") timed on a bounded sample of the same
@@ -110,12 +110,17 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    # control-plane backend: gloo (CPU scalars) by default, so that ALL
+    # device traffic goes through the C-ABI t9_alltoall RCCL communicator
+    # (the product data plane). T9_PG_BACKEND=nccl restores a torch RCCL
+    # process group (needed for T9_EXCHANGE=torch).
+    backend = os.environ.get("T9_PG_BACKEND", "gloo")
     dist = None
     if world > 1:
         import torch.distributed as dist_mod
         dist = dist_mod
         torch.cuda.set_device(local_rank)
-        dist.init_process_group("nccl")
+        dist.init_process_group(backend)
         assert world == args.gpus, (world, args.gpus)
     elif os.environ.get("T9_FORCE_DIST"):
         # validation mode: run the distributed branch at world=1
@@ -123,7 +128,7 @@ def main():
         import torch.distributed as dist_mod
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29712")
-        dist_mod.init_process_group("nccl", rank=0, world_size=1)
+        dist_mod.init_process_group(backend, rank=0, world_size=1)
     assert torch.cuda.is_available(), "bench.py needs a GPU (no CPU path)"
 
     from thrill_amd.pipeline import TeraSort
@@ -156,26 +161,34 @@ def main():
     out, n_out = ts.step()
     torch.cuda.synchronize()
     ts.nat.perf_enable(False)
-    scat_ms, scat_n = ts.nat.perf_read("pair_scatter")
     perf_breakdown = {}
     for cls in ["pair_scatter", "hist_pairs", "lds_sort", "extract", "gather"]:
         ms, cnt = ts.nat.perf_read(cls)
         perf_breakdown[cls] = {"total_ms": round(ms, 3), "launches": cnt}
     ts.nat.perf_reset()
-    # algorithmic bytes per pair_scatter launch: read (8+4) + write (8+4)
-    # B per record = 24 B x n_local records
+    # The roofline tracks the DOMINANT kernel of the step: the payload
+    # gather (out[i] = rec[idx[i]], k_gather_records_span — ~45% of the
+    # step). Algorithmic bytes per launch: 100 B read
+    # + 100 B written + 4 B index per record (SURVEY.md §8d "payload
+    # gather = 2R"). Per-kernel shares for the rest are in
+    # perf_breakdown.
     roofline = None
-    if scat_n:
-        per_launch_s = scat_ms / 1e3 / scat_n
-        algo_bytes = 24.0 * ts.n_local
+    g_ms, g_n = perf_breakdown["gather"]["total_ms"], \
+        perf_breakdown["gather"]["launches"]
+    if g_n:
+        per_launch_s = g_ms / 1e3 / g_n
+        algo_bytes = (2.0 * REC + 4.0) * ts.n_local
         achieved = algo_bytes / per_launch_s / 1e9
         # traffic: PMC bytes per launch from the committed calibration run
         # (profiles/pmc_traffic_terasort_r01.json — separate rocprofv3
         # --pmc FETCH_SIZE / WRITE_SIZE passes; FETCH doubled per the
         # gfx950 half-reporting of wide coalesced reads, absolute values
         # carry the guide's per-pattern calibration caveat). Scaled to
-        # this run's n. None when the calibration file is absent or the
-        # workload size differs wildly.
+        # this run's n. None when the calibration file is absent. The
+        # gather's fetch traffic is ~2.4x the algorithmic read: random
+        # 100-B record spans fetch whole cache lines (the measured
+        # random-line wall — scripts/probe_gather_wall.py puts this
+        # kernel at 82% of its own streaming rate).
         traffic = None
         cal_path = os.path.join(REPO, "profiles",
                                 "pmc_traffic_terasort_r01.json")
@@ -186,7 +199,7 @@ def main():
             tot = 0.0
             cnt = 0
             for k, v in cal.items():
-                if "scatter_wave512" in k or "scatter_seg" in k:
+                if "gather_records" in k:
                     tot += (v["fetch_bytes_x2_per_launch"] +
                             v["write_bytes_per_launch"])
                     cnt += 1
@@ -199,8 +212,9 @@ def main():
             "unit": "GB/s",
             "frac": round(achieved / PEAK_HBM_GBS, 4),
             "traffic": traffic,
-            "kernel": "pair_scatter (radix pass over u64+u32 pairs)",
-            "avg_launch_ms": round(scat_ms / scat_n, 3),
+            "kernel": "gather (payload permute: out[i] = rec[idx[i]], "
+                      "dominant kernel of the step)",
+            "avg_launch_ms": round(g_ms / g_n, 3),
         }
 
     # timed region
@@ -215,17 +229,19 @@ def main():
     torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
     if dist:
-        t = torch.tensor([elapsed], device="cuda")
+        ctl = "cpu" if dist.get_backend() == "gloo" else "cuda"
+        t = torch.tensor([elapsed], device=ctl)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
     # validation
     keysum, mono, first, last = validate(ts, out, n_out)
     if dist:
-        ksum = torch.tensor([keysum], dtype=torch.int64, device="cuda")
+        ctl = "cpu" if dist.get_backend() == "gloo" else "cuda"
+        ksum = torch.tensor([keysum], dtype=torch.int64, device=ctl)
         dist.all_reduce(ksum)
         keysum = int(ksum.item())
-        isum = torch.tensor([in_keysum], dtype=torch.int64, device="cuda")
+        isum = torch.tensor([in_keysum], dtype=torch.int64, device=ctl)
         dist.all_reduce(isum)
         in_keysum = int(isum.item())
         # rank boundaries: my first key must be >= previous rank's last

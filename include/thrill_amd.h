@@ -35,6 +35,18 @@ int t9_create(t9_context** out, int device, int rank, int world, void* comm);
 int t9_destroy(t9_context* ctx);
 const char* t9_version(void);
 
+/* RCCL communicator bootstrap (world > 1): rank 0 generates the unique id
+ * (t9_comm_id_size() bytes, = sizeof(ncclUniqueId)), the caller moves it
+ * to every rank over any host channel (the reference distributes its TCP
+ * endpoints the same way, thrill/api/context.cpp:604-614), then every
+ * rank calls t9_comm_init collectively. The context owns the resulting
+ * communicator (t9_destroy frees it). Passing an external ncclComm_t via
+ * t9_create(comm) remains supported; t9_comm_init on such a context is an
+ * error. */
+int t9_comm_id_size(void);
+int t9_comm_id(void* out_id);
+int t9_comm_init(t9_context* ctx, const void* id);
+
 /* ------------------------------------------------------------------ *
  * Synthetic input generation (device-side, seeded; bit-identical to the
  * oracle's t9o_gen_* so CPU/GPU parity runs on identical bytes).
@@ -207,10 +219,13 @@ int t9_reduce_by_index(t9_context* ctx, const uint64_t* d_keys,
                        void* stream);
 
 /* by-index partition: bucket = (key-begin)*p/size
- * (core/reduce_functional.hpp:113-128) */
+ * (core/reduce_functional.hpp:113-128). Out-of-range keys clamp to the
+ * last partition and set d_error (same error model as
+ * t9_reduce_by_index). */
 int t9_index_bucket(t9_context* ctx, const uint64_t* d_keys, uint64_t n,
                     uint64_t begin, uint64_t size, uint32_t p,
-                    uint32_t* d_bucket, uint64_t* d_counts, void* stream);
+                    uint32_t* d_bucket, uint64_t* d_counts,
+                    uint32_t* d_error, void* stream);
 
 /* GroupByKey support (SURVEY.md §8f item 3 — thrill/api/group_by_key.hpp
  * is sort-based): the group index of a key-sorted array. d_unique[g] /

@@ -107,6 +107,47 @@ def test_distributed_terasort_protocol(n_total):
     assert np.array_equal(got, expect)
 
 
+def _counts_worker(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from thrill_amd.pipeline import exchange_counts, displs_of
+        # deterministic matrix: rank r sends (r*10 + dest) items to dest
+        send = np.array([rank * 10 + d for d in range(world)],
+                        dtype=np.int64)
+        recv = exchange_counts(dist, send, rank, world)
+        expect = np.array([s * 10 + rank for s in range(world)],
+                          dtype=np.int64)
+        assert np.array_equal(recv, expect), (recv, expect)
+        d = displs_of(recv.astype(np.uint64))
+        assert d[0] == 0 and np.all(np.diff(d.astype(np.int64))
+                                    == recv[:-1])
+        q.put((rank, "ok"))
+        dist.barrier()
+        dist.destroy_process_group()
+    except Exception as e:
+        q.put((rank, f"ERROR: {e!r}"))
+        raise
+
+
+def test_exchange_counts_gloo_world3():
+    """the control-plane count exchange (all-gather matrix, my column)
+    that replaces all_to_all_single(counts) — gloo, world 3."""
+    ctxm = mp.get_context("spawn")
+    q = ctxm.Queue()
+    procs = [ctxm.Process(target=_counts_worker, args=(r, 3, 29713, q))
+             for r in range(3)]
+    for p in procs:
+        p.start()
+    for _ in range(3):
+        rank, msg = q.get(timeout=120)
+        assert msg == "ok", msg
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+
 def test_distributed_terasort_protocol_world3():
     """world_size=3 with ragged shard sizes (n % 3 != 0)."""
     global WORLD

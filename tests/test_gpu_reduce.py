@@ -183,11 +183,23 @@ def test_index_bucket_mapping(nat):
     dk = G.dev(keys)
     db = G.empty(n, np.uint32)
     dc = G.empty(p, np.uint64)
+    de = G.empty(1, np.uint32)
     nat.index_bucket(G.ptr(dk), n, begin, size, p, G.ptr(db), G.ptr(dc),
-                     G.stream())
+                     G.ptr(de), G.stream())
     got = G.host(db, np.uint32)
     expect = ((keys - begin) * p // size).astype(np.uint32)
     assert np.array_equal(got, expect)
+    assert int(G.host(de, np.uint32)[0]) == 0
+
+    # out-of-range key: clamps to the last partition AND flags the error
+    keys2 = keys.copy()
+    keys2[123] = begin + size + 7
+    dk2 = G.dev(keys2)
+    nat.index_bucket(G.ptr(dk2), n, begin, size, p, G.ptr(db), G.ptr(dc),
+                     G.ptr(de), G.stream())
+    got2 = G.host(db, np.uint32)
+    assert got2[123] == p - 1
+    assert int(G.host(de, np.uint32)[0]) == 1
 
 
 def test_reduce_overflow_sets_error(nat):

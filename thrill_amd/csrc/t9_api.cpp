@@ -13,6 +13,7 @@
 #include <rccl/rccl.h>
 
 #include <cstdlib>
+#include <cstring>
 #include <new>
 
 extern "C" {
@@ -29,12 +30,53 @@ int t9_create(t9_context** out, int device, int rank, int world,
     ctx->rank = rank;
     ctx->world = world;
     ctx->comm = comm;
+    ctx->owns_comm = 0;
     *out = ctx;
     return T9_OK;
 }
 
 int t9_destroy(t9_context* ctx) {
+    if (ctx && ctx->comm && ctx->owns_comm)
+        ncclCommDestroy((ncclComm_t)ctx->comm);
     delete ctx;
+    return T9_OK;
+}
+
+/* ------------------------------------------------------------------ *
+ * RCCL communicator bootstrap — so the C ABI owns the xGMI data plane
+ * end to end (the reference seam it replaces builds its own TCP mesh in
+ * net::tcp::Construct, thrill/api/context.cpp:604-614). Rank 0 calls
+ * t9_comm_id, the 128-byte id is distributed out of band (any host
+ * channel — the reference distributes its connection endpoints the same
+ * way), then every rank calls t9_comm_init.
+ * ------------------------------------------------------------------ */
+
+int t9_comm_id_size(void) { return (int)sizeof(ncclUniqueId); }
+
+int t9_comm_id(void* out_id) {
+    if (!out_id) return T9_EINVAL;
+    ncclResult_t rc = ncclGetUniqueId((ncclUniqueId*)out_id);
+    if (rc != ncclSuccess) {
+        fprintf(stderr, "t9_comm_id: %s\n", ncclGetErrorString(rc));
+        return T9_EIO;
+    }
+    return T9_OK;
+}
+
+int t9_comm_init(t9_context* ctx, const void* id) {
+    if (!ctx || !id) return T9_EINVAL;
+    if (ctx->comm) return T9_EINVAL;   /* already connected */
+    HIP_TRY(hipSetDevice(ctx->device));
+    ncclComm_t comm = nullptr;
+    ncclResult_t rc = ncclCommInitRank(&comm, ctx->world,
+                                       *(const ncclUniqueId*)id, ctx->rank);
+    if (rc != ncclSuccess) {
+        fprintf(stderr, "t9_comm_init: rank %d/%d: %s\n", ctx->rank,
+                ctx->world, ncclGetErrorString(rc));
+        return T9_EIO;
+    }
+    ctx->comm = comm;
+    ctx->owns_comm = 1;
     return T9_OK;
 }
 
@@ -57,8 +99,28 @@ int t9_alltoall(t9_context* ctx, const void* d_send, const u64* send_counts,
     if (!ctx->comm) return T9_EINVAL;
     ncclComm_t comm = (ncclComm_t)ctx->comm;
     hipStream_t s = (hipStream_t)stream;
+    /* The rank's own share bypasses RCCL: a plain device copy on the same
+     * stream is both faster (no protocol round trip for ~1/p of the data)
+     * and avoids the self-send path entirely — a 10.7 GB RCCL
+     * self-exchange was measured as a hang in round 1 (pipeline.py
+     * history, commit 551d307); T9_A2A_SELF=nccl restores the RCCL
+     * self-send for investigation. */
+    static const bool self_nccl = [] {
+        const char* e = getenv("T9_A2A_SELF");
+        return e && strcmp(e, "nccl") == 0;
+    }();
+    const int me = ctx->rank;
+    if (!self_nccl) {
+        if (send_counts[me] != recv_counts[me]) return T9_EINVAL;
+        if (send_counts[me])
+            HIP_TRY(hipMemcpyAsync(
+                (char*)d_recv + recv_displs[me] * elem_size,
+                (const char*)d_send + send_displs[me] * elem_size,
+                send_counts[me] * elem_size, hipMemcpyDeviceToDevice, s));
+    }
     ncclResult_t rc = ncclGroupStart();
     for (int r = 0; r < ctx->world && rc == ncclSuccess; ++r) {
+        if (r == me && !self_nccl) continue;
         if (send_counts[r])
             rc = ncclSend((const char*)d_send + send_displs[r] * elem_size,
                           send_counts[r] * elem_size, ncclUint8, r, comm, s);

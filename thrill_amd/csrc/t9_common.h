@@ -45,7 +45,8 @@ struct t9_context {
     int device;
     int rank;
     int world;
-    void* comm;   /* ncclComm_t or nullptr */
+    void* comm;     /* ncclComm_t or nullptr */
+    int owns_comm;  /* 1 if t9_comm_init created it (t9_destroy frees) */
 };
 
 /* splitmix64 random access — must match oracle/t9_oracle.cpp splitmix64_at */

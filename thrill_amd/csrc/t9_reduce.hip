@@ -266,10 +266,14 @@ __global__ __launch_bounds__(256) void k_reduce_by_index(
 }
 
 /* partition for the by-index mapping: bucket = (k-begin)*p / size
- * (reduce_functional.hpp:113-128 with num_buckets = p) */
+ * (reduce_functional.hpp:113-128 with num_buckets = p). Out-of-range keys
+ * clamp to the last partition AND set the error flag, mirroring
+ * k_reduce_by_index's error model (a silently misrouted key
+ * would otherwise only surface on the destination rank). */
 __global__ __launch_bounds__(256) void k_index_bucket(
     const u64* __restrict__ keys, u64 n, u64 begin, u64 size, u32 p,
-    u32* __restrict__ bucket, u64* __restrict__ counts) {
+    u32* __restrict__ bucket, u64* __restrict__ counts,
+    u32* __restrict__ err) {
     __shared__ u32 scnt[256];
     const u32 tid = threadIdx.x;
     scnt[tid] = 0;
@@ -277,7 +281,10 @@ __global__ __launch_bounds__(256) void k_index_bucket(
     const u64 stride = (u64)gridDim.x * 256;
     for (u64 i = (u64)blockIdx.x * 256 + tid; i < n; i += stride) {
         u64 g = keys[i] - begin;
-        if (g >= size) g = size - 1;
+        if (g >= size) {
+            g = size - 1;
+            atomicExch(err, 1u);
+        }
         const u32 b = (u32)(g * p / size);
         bucket[i] = b;
         atomicAdd(&scnt[b], 1u);
@@ -398,15 +405,18 @@ int t9_reduce_by_index(t9_context* ctx, const u64* d_keys,
 
 int t9_index_bucket(t9_context* ctx, const u64* d_keys, u64 n, u64 begin,
                     u64 size, u32 p, u32* d_bucket, u64* d_counts,
-                    void* stream) {
+                    u32* d_error, void* stream) {
     (void)ctx;
-    if (!d_counts || p < 1 || p > 256 || size == 0) return T9_EINVAL;
+    if (!d_counts || !d_error || p < 1 || p > 256 || size == 0)
+        return T9_EINVAL;
     hipStream_t s = (hipStream_t)stream;
     HIP_TRY(hipMemsetAsync(d_counts, 0, p * 8, s));
+    HIP_TRY(hipMemsetAsync(d_error, 0, 4, s));
     if (n == 0) return T9_OK;
     if (!d_keys || !d_bucket) return T9_EINVAL;
     hipLaunchKernelGGL(k_index_bucket, dim3(grid_for(n)), dim3(256), 0, s,
-                       d_keys, n, begin, size, p, d_bucket, d_counts);
+                       d_keys, n, begin, size, p, d_bucket, d_counts,
+                       d_error);
     T9_LAUNCH_CHECK();
     return T9_OK;
 }

@@ -143,8 +143,13 @@ __global__ __launch_bounds__(256) void k_hist_seg(
     if (tid == 0) s_abase[256] = abase[256];
     s_cnt[tid] = 0;
     __syncthreads();
-    const u32 tbase = (u32)(blockIdx.x * T9_MSB_TILE);
-    if (tbase < s_abase[256]) {
+    /* u64 tile base: the pass-2 grid is sized B2max (which includes the
+     * 3-level row allowance), so blockIdx.x * TILE can exceed 2^32 for
+     * large n — a wrapped u32 would pass the < abase[256] guard and
+     * re-scatter a stale tile. */
+    const u64 tbase64 = (u64)blockIdx.x * T9_MSB_TILE;
+    if (tbase64 < s_abase[256]) {
+        const u32 tbase = (u32)tbase64;
         const u32 b = bucket_of(s_abase, tbase);
         const u32 off = tbase - s_abase[b];
         const u32 tn = (bucket_n[b] > off)
@@ -221,8 +226,10 @@ __global__ __launch_bounds__(1024, 4) void k_scatter_seg(
     const u32 tid = threadIdx.x, wave = tid >> 6, lane = tid & 63;
     if (tid < 257) s_abase[tid] = abase[tid];
     __syncthreads();
-    const u32 tbase = (u32)(blockIdx.x * TILE);
-    if (tbase >= s_abase[256]) return;
+    /* u64 guard before narrowing — see k_hist_seg */
+    const u64 tbase64 = (u64)blockIdx.x * TILE;
+    if (tbase64 >= s_abase[256]) return;
+    const u32 tbase = (u32)tbase64;
     const u32 b = bucket_of(s_abase, tbase);
     const u32 off_in_bucket = tbase - s_abase[b];
     const u32 bn = bucket_n[b];

@@ -1,5 +1,7 @@
-/* t9_sort_msb9.hip — experimental 9-bit level-2 digit for the two-level
- * MSB pipeline (T9_PASS2_BITS=9, default off): 256 x 512 = 131072
+/* t9_sort_msb9.hip — 9-bit level-2 digit for the two-level MSB pipeline.
+ * This is the PRODUCTION DEFAULT for 2^22 <= n < T9_L3_MIN
+ * (pass2_bits() in t9_sort_msb.hip defaults to 9; T9_PASS2_BITS=8 is the
+ * byte-digit fallback): 256 x 512 = 131072
  * (b7, 9-bit) sub-buckets of ~n/131072 elements, so the level-3 LDS sort
  * runs the 1024-element / 256-thread variant at ~5 blocks per CU — the
  * measured limiter of the 8-bit flow was the LDS sort's barrier parking
@@ -42,8 +44,11 @@ __global__ __launch_bounds__(256) void k_hist_seg9(
     s_cnt[tid] = 0;
     s_cnt[tid + 256] = 0;
     __syncthreads();
-    const u32 tbase = (u32)(blockIdx.x * T9_MSB_TILE);
-    if (tbase < s_abase[256]) {
+    /* u64 tile base before narrowing (grid is B2max-sized; see the same
+     * guard in t9_sort_msb.hip k_hist_seg — ADVICE r01, medium) */
+    const u64 tbase64 = (u64)blockIdx.x * T9_MSB_TILE;
+    if (tbase64 < s_abase[256]) {
+        const u32 tbase = (u32)tbase64;
         const u32 b = bucket_of9(s_abase, tbase);
         const u32 off = tbase - s_abase[b];
         const u32 tn = (bucket_n[b] > off)
@@ -121,8 +126,10 @@ __global__ __launch_bounds__(1024, 4) void k_scatter_seg9(
     const u32 tid = threadIdx.x, wave = tid >> 6, lane = tid & 63;
     if (tid < 257) s_abase[tid] = abase[tid];
     __syncthreads();
-    const u32 tbase = (u32)(blockIdx.x * TILE);
-    if (tbase >= s_abase[256]) return;
+    /* u64 guard before narrowing — see k_hist_seg9 */
+    const u64 tbase64 = (u64)blockIdx.x * TILE;
+    if (tbase64 >= s_abase[256]) return;
+    const u32 tbase = (u32)tbase64;
     const u32 b = bucket_of9(s_abase, tbase);
     const u32 off_in_bucket = tbase - s_abase[b];
     const u32 bn = bucket_n[b];

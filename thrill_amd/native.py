@@ -38,6 +38,9 @@ _SIGS = {
     "t9_version": (ctypes.c_char_p, []),
     "t9_create": (i32, [ctypes.POINTER(vp), i32, i32, i32, vp]),
     "t9_destroy": (i32, [vp]),
+    "t9_comm_id_size": (i32, []),
+    "t9_comm_id": (i32, [vp]),
+    "t9_comm_init": (i32, [vp, vp]),
     "t9_gen_u64": (i32, [vp, vp, u64, u64, u64, vp]),
     "t9_gen_records": (i32, [vp, vp, u64, u64, u64, vp]),
     "t9_sort_u64_workspace": (u64, [u64]),
@@ -58,7 +61,7 @@ _SIGS = {
     "t9_alltoall": (i32, [vp, vp, vp, vp, vp, vp, vp, u64, vp]),
     "t9_hash_bucket": (i32, [vp, vp, u64, u64, u32, vp, vp, vp]),
     "t9_reduce_by_index": (i32, [vp, vp, vp, u64, u64, u64, vp, vp, vp]),
-    "t9_index_bucket": (i32, [vp, vp, u64, u64, u64, u32, vp, vp, vp]),
+    "t9_index_bucket": (i32, [vp, vp, u64, u64, u64, u32, vp, vp, vp, vp]),
     "t9_reduce_init": (i32, [vp, vp, u64, vp]),
     "t9_reduce_build": (i32, [vp, vp, vp, u64, vp, u64, u64, vp, vp]),
     "t9_reduce_drain": (i32, [vp, vp, u64, vp, vp, vp, vp]),
@@ -100,6 +103,23 @@ class Native:
                 raise T9Error(f"t9_{name} failed rc={rc}")
             return rc
         return call
+
+    # RCCL comm bootstrap (t9_comm_id takes no context argument)
+    def comm_id(self):
+        """rank 0: generate the ncclUniqueId as bytes."""
+        size = self._lib.t9_comm_id_size()
+        buf = (ctypes.c_uint8 * size)()
+        rc = self._lib.t9_comm_id(buf)
+        if rc != 0:
+            raise T9Error(f"t9_comm_id failed rc={rc}")
+        return bytes(buf)
+
+    def comm_init(self, id_bytes):
+        """every rank, collectively: connect the context's communicator."""
+        buf = (ctypes.c_uint8 * len(id_bytes)).from_buffer_copy(id_bytes)
+        rc = self._lib.t9_comm_init(self.ctx, buf)
+        if rc != 0:
+            raise T9Error(f"t9_comm_init failed rc={rc}")
 
     # perf registry functions take no context argument
     def perf_enable(self, on):

@@ -2,9 +2,13 @@
 the role of SortNode::MainOp (thrill/api/sort.hpp:537-663) — sample,
 splitter selection on rank 0, classification, partition, all-to-all
 exchange, local sort — with every bulk step a libt9 HIP kernel and the
-exchange torch.distributed (RCCL) / t9_alltoall. Host python here is
-control plane only (splitters are ~KBs, as in the reference where
-FindAndSendSplitters runs on worker 0's CPU).
+BULK exchange through the C-ABI t9_alltoall (RCCL grouped send/recv over
+xGMI, the product data plane; T9_EXCHANGE=torch restores the
+torch.distributed all_to_all_single fallback). Host python here is
+control plane only (splitters and counts are ~KBs, as in the reference
+where FindAndSendSplitters runs on worker 0's CPU); scalar collectives go
+through torch.distributed on whatever backend the process group uses —
+gloo (CPU) by default in bench.py, so the C ABI owns ALL device traffic.
 
 No CPU fallback: everything data-sized runs through the C ABI on the GPU.
 """
@@ -24,8 +28,62 @@ def _ptr(t):
     return ctypes.c_void_p(t.data_ptr())
 
 
+def _nptr(a):
+    """host pointer of a (kept-alive) numpy array."""
+    return ctypes.c_void_p(a.ctypes.data)
+
+
 def _stream():
     return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+def _ctl_device(dist):
+    """device for small control-plane collectives: CPU under gloo (the
+    bench default — keeps ALL device traffic inside the C ABI), CUDA
+    under nccl."""
+    return "cpu" if dist.get_backend() == "gloo" else "cuda"
+
+
+def bootstrap_comm(nat, dist, rank):
+    """Connect the context's RCCL communicator: rank 0 generates the
+    ncclUniqueId (t9_comm_id), the id travels over the torch control
+    plane (one broadcast at init), every rank joins via t9_comm_init —
+    the same out-of-band endpoint exchange the reference does for its
+    TCP mesh (api/context.cpp:604-614)."""
+    obj = [nat.comm_id() if rank == 0 else None]
+    dist.broadcast_object_list(obj, src=0)
+    nat.comm_init(obj[0])
+
+
+def exchange_counts(dist, send_counts, rank, world):
+    """recv_counts[r] = what rank r sends me: all-gather the p x p count
+    matrix over the control plane (tiny) and read my column. Replaces
+    all_to_all_single(counts) which gloo does not support."""
+    dev = _ctl_device(dist)
+    mine = torch.from_numpy(np.ascontiguousarray(send_counts)).to(dev)
+    rows = [torch.empty_like(mine) for _ in range(world)]
+    dist.all_gather(rows, mine)
+    return np.array([int(rows[r][rank].item()) for r in range(world)],
+                    dtype=np.int64)
+
+
+def displs_of(counts):
+    d = np.zeros(len(counts), dtype=np.uint64)
+    if len(counts) > 1:
+        d[1:] = np.cumsum(counts[:-1].astype(np.uint64))
+    return d
+
+
+def a2a(nat, d_send, send_counts, d_recv, recv_counts, elem_size):
+    """one all-to-all-v through the C ABI (t9_alltoall: RCCL grouped
+    send/recv over xGMI; self-share via direct device copy; world==1
+    shortcuts to a device memcpy). counts are element counts."""
+    sc = np.ascontiguousarray(send_counts, dtype=np.uint64)
+    rc = np.ascontiguousarray(recv_counts, dtype=np.uint64)
+    sd = displs_of(sc)
+    rd = displs_of(rc)
+    nat.alltoall(_ptr(d_send), _nptr(sc), _nptr(sd), _ptr(d_recv),
+                 _nptr(rc), _nptr(rd), elem_size, _stream())
 
 
 def sample_size(n_total):
@@ -78,6 +136,9 @@ class WordCount:
     def __init__(self, n_total, vocab, s, seed, rank=0, world=1, device=0):
         self.nat = Native(device=device, rank=rank, world=world)
         self.rank, self.world = rank, world
+        if world > 1 and os.environ.get("T9_EXCHANGE", "t9") == "t9":
+            import torch.distributed as dist
+            bootstrap_comm(self.nat, dist, rank)
         self.n_total, self.seed = n_total, seed
         base = n_total // world
         rem = n_total % world
@@ -145,16 +206,15 @@ class WordCount:
             nat.gather_records(_ptr(keys), _ptr(d_perm), m, 8, _ptr(ks), s)
             nat.gather_records(_ptr(vals), _ptr(d_perm), m, 8, _ptr(vs), s)
         send_counts = d_counts.cpu().numpy().astype(np.int64)
-        sc_t = torch.from_numpy(send_counts).cuda()
-        rc_t = torch.empty(p, dtype=torch.int64, device="cuda")
-        dist.all_to_all_single(rc_t, sc_t)
-        recv_counts = rc_t.cpu().numpy()
+        recv_counts = exchange_counts(dist, send_counts, self.rank,
+                                      self.world)
         n_recv = int(recv_counts.sum())
         rk = torch.empty(max(n_recv, 1), dtype=torch.int64, device="cuda")
         rv = torch.empty(max(n_recv, 1), dtype=torch.int64, device="cuda")
-        if self.world == 1:
-            rk[:n_recv].copy_(ks[:m])
-            rv[:n_recv].copy_(vs[:m])
+        exchange = os.environ.get("T9_EXCHANGE", "t9")
+        if exchange == "t9" or self.world == 1:
+            a2a(self.nat, ks, send_counts, rk, recv_counts, 8)
+            a2a(self.nat, vs, send_counts, rv, recv_counts, 8)
         else:
             dist.all_to_all_single(rk[:n_recv], ks[:m],
                                    output_split_sizes=recv_counts.tolist(),
@@ -178,6 +238,10 @@ class TeraSort:
         self.nat = Native(device=device, rank=rank, world=world)
         self.rank, self.world = rank, world
         self.n_total, self.seed = n_total, seed
+        self.exchange = os.environ.get("T9_EXCHANGE", "t9")
+        if world > 1 and self.exchange == "t9":
+            import torch.distributed as dist
+            bootstrap_comm(self.nat, dist, rank)
         base = n_total // world
         rem = n_total % world
         self.n_local = base + (1 if rank < rem else 0)
@@ -230,36 +294,49 @@ class TeraSort:
         # tensors are equal-sized and plain all_gather works (no object
         # collectives — they cost milliseconds per step at N=8).
         base = self.n_total // self.world
-        S = max(1, min(base, sample_size(self.n_total) // self.world))
+        S = max(1, min(max(base, 1), sample_size(self.n_total) // self.world))
         stride = max(1, self.n_local // S)
         pos = torch.arange(0, self.n_local, stride, device="cuda")[:S]
         if len(pos) < S:   # only possible for tiny shards
-            pos = torch.cat([pos, pos[-1:].expand(S - len(pos))])
+            # degenerate shard (n_local < S, possibly 0): pad with repeats
+            # of the last sample, or position 0 for an empty shard — the
+            # splitter rule is invariant to duplicate samples
+            pad = pos[-1:] if len(pos) else torch.zeros(
+                1, dtype=pos.dtype, device=pos.device)
+            pos = torch.cat([pos, pad.expand(S - len(pos))])
         dpos = pos.to(torch.int32)
-        d_samp = torch.empty(S * REC, dtype=torch.uint8, device="cuda")
-        self.nat.gather_records(_ptr(self.d_in), _ptr(dpos), S, REC,
-                                _ptr(d_samp), _stream())
+        d_samp = torch.zeros(S * REC, dtype=torch.uint8, device="cuda")
+        if self.n_local:
+            self.nat.gather_records(_ptr(self.d_in), _ptr(dpos), S, REC,
+                                    _ptr(d_samp), _stream())
         gidx = (pos + self.gidx0).to(torch.int64)
-        gs = torch.empty(self.world * S * REC, dtype=torch.uint8,
-                         device="cuda")
-        gi = torch.empty(self.world * S, dtype=torch.int64, device="cuda")
-        dist.all_gather_into_tensor(gs, d_samp)
-        dist.all_gather_into_tensor(gi, gidx)
+        # samples are ~KBs: gather + splitter broadcast run on the control
+        # plane (CPU tensors under gloo, CUDA under nccl)
+        ctl = _ctl_device(dist)
+        samp_c = d_samp.to(ctl)
+        gidx_c = gidx.to(ctl)
+        gs_l = [torch.empty_like(samp_c) for _ in range(self.world)]
+        gi_l = [torch.empty_like(gidx_c) for _ in range(self.world)]
+        dist.all_gather(gs_l, samp_c)
+        dist.all_gather(gi_l, gidx_c)
         p = self.world
-        spl_recs_t = torch.empty(max(p - 1, 1) * REC, dtype=torch.uint8,
-                                 device="cuda")
-        spl_idx_t = torch.empty(max(p - 1, 1), dtype=torch.int64,
-                                device="cuda")
+        spl_recs_c = torch.empty(max(p - 1, 1) * REC, dtype=torch.uint8,
+                                 device=ctl)
+        spl_idx_c = torch.empty(max(p - 1, 1), dtype=torch.int64,
+                                device=ctl)
         if self.rank == 0:
-            all_recs = gs.cpu().numpy().reshape(self.world * S, REC)
-            all_idx = gi.cpu().numpy().astype(np.uint64)
+            all_recs = torch.cat(gs_l).cpu().numpy().reshape(
+                self.world * S, REC)
+            all_idx = torch.cat(gi_l).cpu().numpy().astype(np.uint64)
             spl_recs, spl_idx = select_splitters(all_recs, all_idx, p)
-            spl_recs_t.copy_(torch.from_numpy(
-                spl_recs.reshape(-1).copy()).cuda())
-            spl_idx_t.copy_(torch.from_numpy(
-                spl_idx.view(np.int64).copy()).cuda())
-        dist.broadcast(spl_recs_t, 0)
-        dist.broadcast(spl_idx_t, 0)
+            spl_recs_c.copy_(torch.from_numpy(
+                spl_recs.reshape(-1).copy()).to(ctl))
+            spl_idx_c.copy_(torch.from_numpy(
+                spl_idx.view(np.int64).copy()).to(ctl))
+        dist.broadcast(spl_recs_c, 0)
+        dist.broadcast(spl_idx_c, 0)
+        spl_recs_t = spl_recs_c.cuda()
+        spl_idx_t = spl_idx_c.cuda()
         spl_recs = spl_recs_t.cpu().numpy().reshape(p - 1, REC)
         spl_k64 = torch.from_numpy(
             k64_of_records(spl_recs).view(np.int64)).cuda()
@@ -294,18 +371,16 @@ class TeraSort:
         nat.gather_records(_ptr(self.d_in), _ptr(self.d_perm), self.n_local,
                            REC, _ptr(self.d_send), s)
         send_counts = self.d_counts.cpu().numpy().astype(np.int64)
-        sc_t = torch.from_numpy(send_counts).cuda()
-        rc_t = torch.empty(p, dtype=torch.int64, device="cuda")
-        dist.all_to_all_single(rc_t, sc_t)
-        recv_counts = rc_t.cpu().numpy()
+        recv_counts = exchange_counts(dist, send_counts, self.rank,
+                                      self.world)
         n_recv = int(recv_counts.sum())
         d_recv = torch.empty(max(n_recv, 1) * REC, dtype=torch.uint8,
                              device="cuda")
-        if self.world == 1:
-            # loopback: direct device copy (the same shortcut
-            # t9_alltoall takes at world==1; a 10.7 GB NCCL self-exchange
-            # measured as a hang)
-            d_recv[:n_recv * REC].copy_(self.d_send[:n_recv * REC])
+        if self.exchange == "t9" or self.world == 1:
+            # the product exchange: C-ABI RCCL all-to-all-v over xGMI
+            # (t9_alltoall; self-share via direct device copy, world==1
+            # shortcuts entirely to a device memcpy)
+            a2a(nat, self.d_send, send_counts, d_recv, recv_counts, REC)
         else:
             dist.all_to_all_single(
                 d_recv[:n_recv * REC], self.d_send,
