@@ -102,7 +102,10 @@ uint64_t t9_sort_records_workspace(uint64_t n, uint32_t rec_size);
  * 10-byte key; ties beyond the radix-sorted u64 key prefix are resolved by
  * comparing the remaining bytes). d_in is preserved; d_out receives the
  * sorted sequence. key_len <= rec_size; key bytes start at offset 0.
- * Synchronizes the stream internally (tie fixing may need a host pass). */
+ * Equal-u64-prefix runs are re-ordered ON DEVICE (segmented LSD over the
+ * tail bytes using the stable pair sort as the primitive; identical
+ * chunks are skipped). Synchronizes the stream internally (tie counts
+ * and skip flags round-trip to the host). */
 int t9_sort_records(t9_context* ctx, const uint8_t* d_in, uint8_t* d_out,
                     uint64_t n, uint32_t rec_size, uint32_t key_len,
                     void* d_workspace, void* stream);

@@ -163,6 +163,61 @@ def test_sort_records_prefix_collisions(nat, oracle):
     assert np.array_equal(got, oracle.sort_records(recs))
 
 
+def test_sort_records_many_tie_runs(nat, oracle):
+    # the on-device tie sort's regrouping pass: many distinct equal-prefix
+    # runs interleaved with unique-prefix records; the final stable
+    # prefix pass must restore run grouping after the tail-chunk LSD.
+    n = 200_000
+    rng = np.random.default_rng(7)
+    recs = oracle.gen_records(n, seed=11)
+    # 1000 shared prefixes over half the records, the rest unique
+    share = rng.choice(n, n // 2, replace=False)
+    pref = rng.integers(0, 1000, len(share))
+    for b in range(8):
+        recs[share, b] = ((pref >> (8 * (7 - b))) & 0xFF).astype(np.uint8)
+    recs[share, 10:] = rng.integers(0, 256, (len(share), 90)) \
+        .astype(np.uint8)
+    din = G.dev(recs.reshape(-1))
+    dout = G.empty(n * 100, np.uint8)
+    w = G.ws(nat.ws("sort_records", n, 100))
+    nat.sort_records(G.ptr(din), G.ptr(dout), n, 100, 10, G.ptr(w),
+                     G.stream())
+    got = G.host(dout, np.uint8).reshape(n, 100)
+    assert np.array_equal(got, oracle.sort_records(recs))
+
+
+def test_sort_records_all_identical(nat, oracle):
+    # every byte of every record equal: the tie sort's chunk-skip path
+    # (no chunk differs, so zero pair sorts run; stable order preserved)
+    n = 50_000
+    recs = np.full((n, 100), 0xA7, np.uint8)
+    din = G.dev(recs.reshape(-1))
+    dout = G.empty(n * 100, np.uint8)
+    w = G.ws(nat.ws("sort_records", n, 100))
+    nat.sort_records(G.ptr(din), G.ptr(dout), n, 100, 10, G.ptr(w),
+                     G.stream())
+    got = G.host(dout, np.uint8).reshape(n, 100)
+    assert np.array_equal(got, recs)
+
+
+def test_sort_records_narrow_ties(nat, oracle):
+    # rec_size=12 (< 24 B): the tie scratch cannot fit in d_out, so the
+    # temporary-allocation fallback runs. Ties differ only in the last
+    # 4 tail bytes (a partial 8-byte chunk, zero-padded).
+    n = 30_000
+    rng = np.random.default_rng(13)
+    recs = rng.integers(0, 256, (n, 12)).astype(np.uint8)
+    recs[:, :8] = np.array([1, 2, 3, 4, 5, 6, 7, 8], np.uint8)
+    din = G.dev(recs.reshape(-1))
+    dout = G.empty(n * 12, np.uint8)
+    w = G.ws(nat.ws("sort_records", n, 12))
+    nat.sort_records(G.ptr(din), G.ptr(dout), n, 12, 12, G.ptr(w),
+                     G.stream())
+    got = G.host(dout, np.uint8).reshape(n, 12)
+    order = np.lexsort(tuple(recs[:, c] for c in range(11, -1, -1)))
+    assert np.array_equal(got, recs[order])
+
+
 def test_extract_key64_bigendian(nat, oracle):
     n = 10_000
     recs = oracle.gen_records(n, seed=6)
