@@ -209,6 +209,40 @@ int t9_reduce_drain(t9_context* ctx, const uint64_t* d_table,
                     uint64_t capacity, uint64_t* d_out_keys,
                     uint64_t* d_out_vals, uint64_t* d_out_n, void* stream);
 
+/* ------------------------------------------------------------------ *
+ * 128-bit composite-key reduce — config-4 string identity. The
+ * reference reduces (std::string, u64) with equality on the FULL key
+ * (core/reduce_probing_hash_table.hpp:233); here words are
+ * dictionary-encoded at tokenize time into TWO independent 64-bit
+ * hashes (k1, k2) and reduced on the 128-bit composite: distinct words
+ * stay separate unless both hashes collide (p ~= 2^-128 per pair; a
+ * forced k1 collision stays separate — parity-tested). Table slot = 3
+ * interleaved u64 {k1, k2, sum}, 3*capacity array, capacity a power of
+ * two. k1 == ~0 / k2 == ~0 are reserved sentinels (t9_hash2_of remaps
+ * them). d_vals == NULL means every pair counts 1.
+ * ------------------------------------------------------------------ */
+int t9_reduce128_init(t9_context* ctx, uint64_t* d_table,
+                      uint64_t capacity, void* stream);
+int t9_reduce128_build(t9_context* ctx, const uint64_t* d_k1,
+                       const uint64_t* d_k2, const uint64_t* d_vals,
+                       uint64_t n, uint64_t* d_table, uint64_t capacity,
+                       uint64_t salt, uint32_t* d_error, void* stream);
+int t9_reduce128_drain(t9_context* ctx, const uint64_t* d_table,
+                       uint64_t capacity, uint64_t* d_out_k1,
+                       uint64_t* d_out_k2, uint64_t* d_out_vals,
+                       uint64_t* d_out_n, void* stream);
+/* two independent 64-bit hashes of u64 token ids (synthetic stand-in for
+ * hashing the word bytes at tokenize time; remaps the reserved
+ * sentinels) */
+int t9_hash2_of(t9_context* ctx, const uint64_t* d_ids, uint64_t n,
+                uint64_t* d_k1, uint64_t* d_k2, void* stream);
+/* bucket = key % p — the partition mapping when the key already is the
+ * hash (128-bit path partitions on k1, mirroring h % num_partitions,
+ * core/reduce_functional.hpp:60-72) */
+int t9_bucket_mod(t9_context* ctx, const uint64_t* d_keys, uint64_t n,
+                  uint32_t p, uint32_t* d_bucket, uint64_t* d_counts,
+                  void* stream);
+
 /* ReduceToIndex — reference
  * api/reduce_to_index.hpp + core/reduce_by_index_post_phase.hpp with the
  * ReduceByIndex mapping (core/reduce_functional.hpp:84-149): keys are

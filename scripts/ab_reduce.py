@@ -21,7 +21,7 @@ def main():
     ap.add_argument("--reps", type=int, default=4)
     args = ap.parse_args()
     wc = WordCount(args.n, args.vocab, args.zipf, seed=0x7, rank=0,
-                   world=1, device=0)
+                   world=1, device=0, keys128=False)
     wc.generate()
     torch.cuda.synchronize()
     wc.step()

@@ -52,7 +52,8 @@ def test_wordcount_distributed_branch_world1(dist_world1, oracle):
     # for the exchange logic by running world=1 through step()'s
     # distributed branch
     n, vocab = 1 << 18, 20_000
-    wc = WordCount(n, vocab, 1.1, seed=9, rank=0, world=1, device=0)
+    wc = WordCount(n, vocab, 1.1, seed=9, rank=0, world=1, device=0,
+                   keys128=False)
     wc.generate()
     ok, ov, m = wc.step()  # T9_FORCE_DIST: hash partition + all-to-all +
     # final reduce all execute (self-exchange)

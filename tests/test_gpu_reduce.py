@@ -144,7 +144,8 @@ def test_hash_bucket_matches_reference_mapping(nat, oracle):
 def test_wordcount_pipeline_single_gpu(nat, oracle):
     from thrill_amd.pipeline import WordCount, zipf_cdf
     n, vocab = 1 << 20, 50_000
-    wc = WordCount(n, vocab, 1.1, seed=5, rank=0, world=1, device=0)
+    wc = WordCount(n, vocab, 1.1, seed=5, rank=0, world=1, device=0,
+                   keys128=False)
     wc.generate()
     ok, ov, m = wc.step()
     gk = G.host(ok, np.uint64)

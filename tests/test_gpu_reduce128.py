@@ -1,0 +1,211 @@
+"""128-bit composite-key reduce parity (config-4 string identity).
+
+The reference reduces (std::string, u64) with equality on the FULL key
+(core/reduce_probing_hash_table.hpp:233 probes compare keys, not their
+hashes). The MI355X path dictionary-encodes words into two independent
+64-bit hashes and reduces on the composite; these tests pin:
+  - multiset parity vs a host reduction keyed on the (k1, k2) tuple,
+  - the VERDICT-r01 'done' bar: two words with a FORCED k1 collision
+    (equal k1, different k2) keep separate counts,
+  - the bacon-ipsum KAT (the reference's one in-repo known answer,
+    word_count_test.cpp:36-79) through the 128-bit path with real
+    strings,
+  - the WordCount pipeline on the 128-bit default.
+"""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from tests import _gpu as G
+    from thrill_amd import Native
+
+
+@pytest.fixture(scope="module")
+def nat():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    n = Native(device=0)
+    yield n
+    n.close()
+
+
+def host_reduce128(k1, k2, vals):
+    """expected result: sums keyed on the (k1, k2) tuple (the oracle for
+    this path — equality on the full composite, per the reference's
+    full-key equality; oracle/t9_oracle.cpp pins the u64 probing
+    semantics these kernels share)."""
+    acc = {}
+    for a, b, v in zip(k1.tolist(), k2.tolist(), vals.tolist()):
+        acc[(a, b)] = acc.get((a, b), 0) + v
+    items = sorted(acc.items())
+    return items
+
+
+def run_reduce128(nat, k1, k2, vals, cap, salt=0):
+    n = len(k1)
+    d1, d2 = G.dev(k1), G.dev(k2)
+    dv = G.dev(vals) if vals is not None else None
+    tbl = G.empty(3 * cap, np.uint64)
+    derr = G.empty(1, np.uint32)
+    dn = G.empty(1, np.uint64)
+    o1 = G.empty(cap, np.uint64)
+    o2 = G.empty(cap, np.uint64)
+    ov = G.empty(cap, np.uint64)
+    s = G.stream()
+    nat.reduce128_init(G.ptr(tbl), cap, s)
+    nat.reduce128_build(G.ptr(d1), G.ptr(d2),
+                        G.ptr(dv) if dv is not None else None, n,
+                        G.ptr(tbl), cap, salt, G.ptr(derr), s)
+    nat.reduce128_drain(G.ptr(tbl), cap, G.ptr(o1), G.ptr(o2), G.ptr(ov),
+                        G.ptr(dn), s)
+    assert int(G.host(derr, np.uint32)[0]) == 0, "table overflow"
+    m = int(G.host(dn, np.uint64)[0])
+    g1 = G.host(o1, np.uint64)[:m]
+    g2 = G.host(o2, np.uint64)[:m]
+    gv = G.host(ov, np.uint64)[:m]
+    got = sorted(zip(g1.tolist(), g2.tolist(), gv.tolist()))
+    return [((a, b), v) for a, b, v in got]
+
+
+def test_reduce128_parity_random(nat):
+    rng = np.random.default_rng(21)
+    n = 1 << 19
+    k1 = rng.integers(0, 5000, n).astype(np.uint64)
+    k2 = (k1 * np.uint64(0x9E3779B97F4A7C15)) ^ np.uint64(7)
+    vals = rng.integers(0, 1 << 40, n).astype(np.uint64)
+    got = run_reduce128(nat, k1, k2, vals, cap=1 << 14)
+    assert got == host_reduce128(k1, k2, vals)
+
+
+def test_reduce128_forced_k1_collision(nat):
+    # two distinct 'words' sharing k1 (forced single-hash collision)
+    # MUST keep separate counts — the observable the reference's
+    # full-key equality provides (VERDICT r01 item 5 'done' bar).
+    n = 100_000
+    rng = np.random.default_rng(22)
+    which = rng.integers(0, 2, n).astype(np.uint64)   # word A or B
+    k1 = np.full(n, 0x1234567812345678, np.uint64)    # SAME k1
+    k2 = np.where(which == 0, np.uint64(111), np.uint64(222))
+    got = run_reduce128(nat, k1, k2, None, cap=1 << 10)
+    nA = int((which == 0).sum())
+    assert got == [((0x1234567812345678, 111), nA),
+                   ((0x1234567812345678, 222), n - nA)]
+
+
+def test_reduce128_many_collision_groups(nat):
+    # heavier collision stress: 64 k1 values x 32 k2 values each, all
+    # probing from the same few start slots in a small table.
+    rng = np.random.default_rng(23)
+    n = 1 << 18
+    k1 = rng.integers(0, 64, n).astype(np.uint64)
+    k2 = rng.integers(0, 32, n).astype(np.uint64) + np.uint64(1)
+    vals = rng.integers(0, 1000, n).astype(np.uint64)
+    got = run_reduce128(nat, k1, k2, vals, cap=1 << 13)
+    assert got == host_reduce128(k1, k2, vals)
+
+
+def test_reduce128_vals_none_counts(nat):
+    # d_vals == NULL: each pair counts 1 (the word_count PreOp emits
+    # (word, 1) — word_count.hpp:43-45)
+    k1 = np.array([5, 5, 9, 5], np.uint64)
+    k2 = np.array([1, 1, 2, 1], np.uint64)
+    got = run_reduce128(nat, k1, k2, None, cap=16)
+    assert got == [((5, 1), 3), ((9, 2), 1)]
+
+
+def test_hash2_of_remaps_sentinels(nat):
+    n = 1 << 16
+    ids = np.arange(n, dtype=np.uint64)
+    d = G.dev(ids)
+    o1, o2 = G.empty(n, np.uint64), G.empty(n, np.uint64)
+    nat.hash2_of(G.ptr(d), n, G.ptr(o1), G.ptr(o2), G.stream())
+    h1 = G.host(o1, np.uint64)
+    h2 = G.host(o2, np.uint64)
+    assert not (h1 == np.uint64(2**64 - 1)).any()
+    assert not (h2 == np.uint64(2**64 - 1)).any()
+    # independence sanity: the two hashes never coincide on this range
+    assert not (h1 == h2).any()
+
+
+def test_bucket_mod_partition(nat):
+    n, p = 200_000, 8
+    rng = np.random.default_rng(29)
+    keys = rng.integers(0, 2**63, n).astype(np.uint64)
+    dk = G.dev(keys)
+    db = G.empty(n, np.uint32)
+    dc = G.empty(p, np.uint64)
+    nat.bucket_mod(G.ptr(dk), n, p, G.ptr(db), G.ptr(dc), G.stream())
+    assert np.array_equal(G.host(db, np.uint32),
+                          (keys % p).astype(np.uint32))
+    assert G.host(dc, np.uint64).sum() == n
+
+
+def test_bacon_ipsum_kat_128(nat, oracle):
+    """the reference KAT through the 128-bit path with REAL strings:
+    tokenize tests/golden/wordcount.in, hash each word twice (fnv-1a
+    with two bases, mixed by Hash128to64 — the framework's string
+    hash), reduce on GPU, map hashes back, compare the 71-entry table
+    (word_count_test.cpp:36-79)."""
+    import json
+    import os
+
+    def fnv1a(data, basis):
+        h = basis
+        for b in data:
+            h = ((h ^ b) * 0x100000001B3) % (1 << 64)
+        return h
+
+    here = os.path.dirname(os.path.abspath(__file__))
+    with open(os.path.join(here, "golden",
+                           "bacon_ipsum_correct.json")) as f:
+        table = json.load(f)
+    words = []
+    with open(os.path.join(here, "golden", "wordcount.in")) as f:
+        for line in f:
+            words += [w for w in line.rstrip("\n").split(" ") if w]
+
+    def hashes(w):
+        b = w.encode()
+        h1 = oracle.hash128to64(0x9AE16A3B2F90404F, fnv1a(b,
+                                0xCBF29CE484222325))
+        h2 = oracle.hash128to64(0xC3A5C85C97CB3127, fnv1a(b,
+                                0x84222325CBF29CE4))
+        if h1 == 2**64 - 1:
+            h1 ^= 1
+        if h2 == 2**64 - 1:
+            h2 ^= 1
+        return h1, h2
+
+    wh = {w: hashes(w) for w in set(words)}
+    k1 = np.array([wh[w][0] for w in words], dtype=np.uint64)
+    k2 = np.array([wh[w][1] for w in words], dtype=np.uint64)
+    got = run_reduce128(nat, k1, k2, None, cap=256)
+    back = {h: w for w, h in wh.items()}
+    result = {back[k]: v for k, v in got}
+    assert result == table
+
+
+def test_wordcount_pipeline_128_single_gpu(nat, oracle):
+    from thrill_amd.pipeline import WordCount, zipf_cdf
+    n, vocab = 1 << 20, 50_000
+    wc = WordCount(n, vocab, 1.1, seed=5, rank=0, world=1, device=0)
+    assert wc.keys128
+    wc.generate()
+    k1, k2, v, m = wc.step()
+    g1 = G.host(k1, np.uint64)
+    g2 = G.host(k2, np.uint64)
+    gv = G.host(v, np.uint64)
+    # expected: oracle token stream -> per-id counts -> hashed pairs
+    toks = oracle.zipf_tokens(zipf_cdf(vocab, 1.1), n, seed=5)
+    ids, cnt = np.unique(toks, return_counts=True)
+    def h(salt, x):
+        v_ = oracle.hash128to64(salt, int(x))
+        return v_ ^ 1 if v_ == 2**64 - 1 else v_
+    ek = sorted((h(0x9AE16A3B2F90404F, i), h(0xC3A5C85C97CB3127, i),
+                 int(c)) for i, c in zip(ids, cnt))
+    got = sorted(zip(g1.tolist(), g2.tolist(), gv.tolist()))
+    assert got == ek
+    wc.close()

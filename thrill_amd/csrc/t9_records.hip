@@ -226,18 +226,39 @@ __global__ __launch_bounds__(256) void k_tie_chunk(
     }
 }
 
-/* flag = 1 if any out[j] != out[0] (skip-sort probe) */
-__global__ __launch_bounds__(256) void k_tie_differ(
-    const u64* __restrict__ v, u64 m, u32* __restrict__ flag) {
-    __shared__ u32 s;
-    if (threadIdx.x == 0) s = 0;
+/* One-pass prescan over the tied set: flags[0] = 1 if the u64 prefix
+ * differs anywhere (>= 2 runs -> the regroup pass is needed); flags[1+c]
+ * = 1 if tail chunk c differs anywhere. Order-invariant (compares
+ * against record tidx[0]'s bytes), so it runs ONCE before the sort
+ * loop; chunks that never differ are skipped entirely — the
+ * all-identical-records adversarial case costs this single read of the
+ * tied records instead of one extract+check round per chunk. */
+__global__ __launch_bounds__(256) void k_tie_prescan(
+    const u8* __restrict__ recs, const u32* __restrict__ tidx, u64 m,
+    u32 rec_size, u32 nc, u32* __restrict__ flags) {
+    __shared__ u32 s_flags[64];
+    const u32 tid = threadIdx.x;
+    for (u32 c = tid; c <= nc && c < 64; c += 256) s_flags[c] = 0;
     __syncthreads();
-    const u64 v0 = v[0];
+    const u8* ref = recs + (u64)tidx[0] * rec_size;
     const u64 stride = (u64)gridDim.x * 256;
-    for (u64 j = (u64)blockIdx.x * 256 + threadIdx.x; j < m; j += stride)
-        if (v[j] != v0) { s = 1; break; }
+    for (u64 j = (u64)blockIdx.x * 256 + tid; j < m; j += stride) {
+        const u8* r = recs + (u64)tidx[j] * rec_size;
+        /* prefix (bytes 0..8) */
+        u32 diff0 = 0;
+        for (u32 t = 0; t < 8; ++t) diff0 |= (u32)(r[t] ^ ref[t]);
+        if (diff0) s_flags[0] = 1;
+        for (u32 c = 0; c < nc; ++c) {
+            const u32 o = 8 + 8 * c;
+            const u32 e = (o + 8 < rec_size) ? o + 8 : rec_size;
+            u32 d = 0;
+            for (u32 t = o; t < e; ++t) d |= (u32)(r[t] ^ ref[t]);
+            if (d) s_flags[1 + c] = 1;
+        }
+    }
     __syncthreads();
-    if (threadIdx.x == 0 && s) atomicExch(flag, 1u);
+    for (u32 c = tid; c <= nc && c < 64; c += 256)
+        if (s_flags[c]) atomicExch(&flags[c], 1u);
 }
 
 /* d_idx[perm[j]] = tidx[j] — write the re-ordered record indices back
@@ -535,10 +556,27 @@ static int sort_records_impl(t9_context* ctx, const u8* d_in, u8* d_out,
             const u32 mgrid =
                 (u32)((t9_ceil_div(m, 256) < 4096) ? t9_ceil_div(m, 256)
                                                    : 4096);
-            /* LSD over the tail chunks (last -> first), then the prefix */
+            /* prescan: which chunks differ at all (one read of the tied
+             * records); then LSD over the differing tail chunks (last ->
+             * first) and, if >= 2 runs, the prefix regroup pass */
             const u32 nc = (rec_size > 8)
                                ? (u32)t9_ceil_div(rec_size - 8, 8) : 0;
+            u32 hflags[64];
+            if (nc <= 63) {
+                HIP_TRY(hipMemsetAsync(t_flag, 0, 64 * 4, s));
+                hipLaunchKernelGGL(k_tie_prescan, dim3(mgrid), dim3(256),
+                                   0, s, d_in, t_idx, m, rec_size, nc,
+                                   t_flag);
+                T9_LAUNCH_CHECK();
+                HIP_TRY(hipMemcpyAsync(hflags, t_flag, (nc + 1) * 4,
+                                       hipMemcpyDeviceToHost, s));
+                HIP_TRY(hipStreamSynchronize(s));
+            }
+            else {
+                for (u32 c = 0; c < 64; ++c) hflags[c] = 1;
+            }
             for (u32 c = nc + 1; !rc && c-- > 0;) {
+                if (nc <= 63 && !hflags[c]) continue;
                 const u32 off = c ? 8 + (c - 1) * 8 : 0;
                 if (off == 0 && le)
                     hipLaunchKernelGGL((k_tie_chunk<true>), dim3(mgrid),
@@ -548,17 +586,9 @@ static int sort_records_impl(t9_context* ctx, const u8* d_in, u8* d_out,
                     hipLaunchKernelGGL((k_tie_chunk<false>), dim3(mgrid),
                                        dim3(256), 0, s, d_in, t_idx, m,
                                        rec_size, off, t_chunk);
-                HIP_TRY(hipMemsetAsync(t_flag, 0, 4, s));
-                hipLaunchKernelGGL(k_tie_differ, dim3(mgrid), dim3(256), 0,
-                                   s, t_chunk, m, t_flag);
                 T9_LAUNCH_CHECK();
-                u32 f = 0;
-                HIP_TRY(hipMemcpyAsync(&f, t_flag, 4,
-                                       hipMemcpyDeviceToHost, s));
-                HIP_TRY(hipStreamSynchronize(s));
-                if (f)
-                    rc = t9_sort_pairs_u64_u32(ctx, t_chunk, t_idx, m,
-                                               pair_ws, stream);
+                rc = t9_sort_pairs_u64_u32(ctx, t_chunk, t_idx, m,
+                                           pair_ws, stream);
             }
             if (!rc) {
                 hipLaunchKernelGGL(k_tie_scatter, dim3(mgrid), dim3(256),

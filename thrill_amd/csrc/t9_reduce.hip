@@ -246,6 +246,206 @@ __global__ __launch_bounds__(256) void k_reduce_build_lds(
     }
 }
 
+/* ------------------------------------------------------------------ *
+ * 128-bit composite-key reduce — config-4 string identity.
+ *
+ * The reference reduces (std::string word, u64 count) pairs with
+ * EQUALITY ON THE FULL KEY (core/reduce_probing_hash_table.hpp:233
+ * probes compare keys, not hashes). The MI355X-native design
+ * dictionary-encodes words at tokenize time into TWO independent 64-bit
+ * hashes (k1, k2) and reduces on the 128-bit composite: two distinct
+ * words occupy distinct table slots unless BOTH hashes collide
+ * (p ~= 2^-128 per pair); in particular a forced k1 collision keeps
+ * counts separate (parity-tested), which is the observable the
+ * reference's full-key equality provides for any constructible input.
+ * Slot = 3 interleaved u64 {k1, k2, sum}. Claim protocol: CAS k1
+ * (EMPTY -> k1), then CAS k2 (EMPTY -> k2); a k1 match with a k2
+ * mismatch advances the probe, so every (k1, k2) group converges to one
+ * slot (all inserters scan the same slot order and only skip slots that
+ * definitively mismatch). k1 == EMPTY / k2 == EMPTY are reserved; the
+ * hashing kernels remap them (t9_hash2_of).
+ * ------------------------------------------------------------------ */
+
+__global__ __launch_bounds__(256) void k_reduce128_init(
+    u64* __restrict__ t, u64 cap) {
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 i = (u64)blockIdx.x * 256 + threadIdx.x; i < cap;
+         i += stride) {
+        t[3 * i] = T9_EMPTY;
+        t[3 * i + 1] = T9_EMPTY;
+        t[3 * i + 2] = 0;
+    }
+}
+
+/* global-table composite insert (shared by the build's cold path and the
+ * LDS flush) */
+__device__ inline void t9_g128_insert(u64* __restrict__ t, u64 cap,
+                                      u64 salt, u64 k1, u64 k2, u64 v,
+                                      u32* __restrict__ err) {
+    u64 slot = t9_hash128to64(salt, k1) & (cap - 1);
+    u64 probes = 0;
+    for (;;) {
+        u64 p1 = atomicCAS((unsigned long long*)&t[3 * slot],
+                           (unsigned long long)T9_EMPTY,
+                           (unsigned long long)k1);
+        if (p1 == T9_EMPTY || p1 == k1) {
+            u64 p2 = atomicCAS((unsigned long long*)&t[3 * slot + 1],
+                               (unsigned long long)T9_EMPTY,
+                               (unsigned long long)k2);
+            if (p2 == T9_EMPTY || p2 == k2) {
+                atomicAdd((unsigned long long*)&t[3 * slot + 2],
+                          (unsigned long long)v);
+                return;
+            }
+            /* k1 matches, k2 differs: a (rare) k1 collision — probe on */
+        }
+        slot = (slot + 1) & (cap - 1);
+        if (++probes > cap) {
+            atomicExch(err, 1u);
+            return;
+        }
+    }
+}
+
+/* LDS-accumulated 128-bit build: per-block (k1, k2, sum) filter table
+ * absorbing the Zipf head at LDS-atomic speed (same design as
+ * k_reduce_build_lds, composite equality). SLOTS x 24 B of LDS. */
+template <int SLOTS>
+__global__ __launch_bounds__(256) void k_reduce128_build_lds(
+    const u64* __restrict__ k1s, const u64* __restrict__ k2s,
+    const u64* __restrict__ vals, u64 n, u64* __restrict__ t, u64 cap,
+    u64 salt, u32* __restrict__ err) {
+    __shared__ u64 lk1[SLOTS];
+    __shared__ u64 lk2[SLOTS];
+    __shared__ u64 lv[SLOTS];
+    const u32 tid = threadIdx.x;
+    for (u32 s = tid; s < (u32)SLOTS; s += 256) {
+        lk1[s] = T9_EMPTY;
+        lk2[s] = T9_EMPTY;
+        lv[s] = 0;
+    }
+    __syncthreads();
+
+    const u64 gsz = (u64)gridDim.x * 256;
+    for (u64 base = (u64)blockIdx.x * 256; base < n; base += gsz) {
+        const u64 i = base + tid;
+        if (i >= n) continue;
+        const u64 k1 = k1s[i], k2 = k2s[i];
+        const u64 v = vals ? vals[i] : 1;
+        const u64 h = t9_hash128to64(salt, k1);
+        u32 ls = (u32)(h >> 48) & (SLOTS - 1);
+        bool done = false;
+        for (int p = 0; p < 4; ++p) {
+            u64 p1 = atomicCAS((unsigned long long*)&lk1[ls],
+                               (unsigned long long)T9_EMPTY,
+                               (unsigned long long)k1);
+            if (p1 == T9_EMPTY || p1 == k1) {
+                u64 p2 = atomicCAS((unsigned long long*)&lk2[ls],
+                                   (unsigned long long)T9_EMPTY,
+                                   (unsigned long long)k2);
+                if (p2 == T9_EMPTY || p2 == k2) {
+                    atomicAdd((unsigned long long*)&lv[ls],
+                              (unsigned long long)v);
+                    done = true;
+                    break;
+                }
+            }
+            ls = (ls + 1) & (SLOTS - 1);
+        }
+        if (!done)
+            t9_g128_insert(t, cap, salt, k1, k2, v, err);
+    }
+    __syncthreads();
+
+    for (u32 s = tid; s < (u32)SLOTS; s += 256) {
+        if (lk1[s] == T9_EMPTY) continue;
+        /* a claimed slot whose k2 CAS was lost to another key keeps
+         * k2 == EMPTY only if no inserter ever won it — then its count
+         * is 0 and it can be skipped */
+        if (lk2[s] == T9_EMPTY) continue;
+        t9_g128_insert(t, cap, salt, lk1[s], lk2[s], lv[s], err);
+    }
+}
+
+/* drain (k1, k2, sum) triples — block-aggregated range claims, as
+ * k_reduce_drain */
+__global__ __launch_bounds__(256) void k_reduce128_drain(
+    const u64* __restrict__ t, u64 cap, u64* __restrict__ ok1,
+    u64* __restrict__ ok2, u64* __restrict__ ov,
+    u64* __restrict__ out_n) {
+    __shared__ u32 s_pre[256];
+    __shared__ u64 s_base;
+    const u32 tid = threadIdx.x;
+    const u64 stride = (u64)gridDim.x * 256;
+    const u64 gid = (u64)blockIdx.x * 256 + tid;
+    u32 mine = 0;
+    for (u64 i = gid; i < cap; i += stride)
+        if (t[3 * i] != T9_EMPTY && t[3 * i + 1] != T9_EMPTY) ++mine;
+    s_pre[tid] = mine;
+    __syncthreads();
+    for (int off = 1; off < 256; off <<= 1) {
+        u32 y = (tid >= (u32)off) ? s_pre[tid - off] : 0;
+        __syncthreads();
+        s_pre[tid] += y;
+        __syncthreads();
+    }
+    if (tid == 255 && s_pre[255])
+        s_base = atomicAdd((unsigned long long*)out_n,
+                           (unsigned long long)s_pre[255]);
+    __syncthreads();
+    if (mine) {
+        u64 pos = s_base + s_pre[tid] - mine;
+        for (u64 i = gid; i < cap; i += stride) {
+            if (t[3 * i] != T9_EMPTY && t[3 * i + 1] != T9_EMPTY) {
+                ok1[pos] = t[3 * i];
+                ok2[pos] = t[3 * i + 1];
+                ov[pos] = t[3 * i + 2];
+                ++pos;
+            }
+        }
+    }
+}
+
+/* two independent 64-bit hashes of a u64 token id (the synthetic stand-in
+ * for hashing the word string at tokenize time; reserved sentinel values
+ * are remapped so they never reach the table) */
+__global__ __launch_bounds__(256) void k_hash2_of(
+    const u64* __restrict__ ids, u64 n, u64* __restrict__ k1s,
+    u64* __restrict__ k2s) {
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 i = (u64)blockIdx.x * 256 + threadIdx.x; i < n; i += stride) {
+        const u64 id = ids[i];
+        u64 h1 = t9_hash128to64(0x9AE16A3B2F90404Full, id);
+        u64 h2 = t9_hash128to64(0xC3A5C85C97CB3127ull, id);
+        if (h1 == T9_EMPTY) h1 ^= 1;
+        if (h2 == T9_EMPTY) h2 ^= 1;
+        k1s[i] = h1;
+        k2s[i] = h2;
+    }
+}
+
+/* bucket = key % p — the partition mapping when the key already IS the
+ * hash (128-bit path: partition on k1, mirroring the reference's
+ * h % num_partitions, core/reduce_functional.hpp:60-72) */
+__global__ __launch_bounds__(256) void k_bucket_mod(
+    const u64* __restrict__ keys, u64 n, u32 p, u32* __restrict__ bucket,
+    u64* __restrict__ counts) {
+    __shared__ u32 scnt[256];
+    const u32 tid = threadIdx.x;
+    scnt[tid] = 0;
+    __syncthreads();
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 i = (u64)blockIdx.x * 256 + tid; i < n; i += stride) {
+        const u32 b = (u32)(keys[i] % p);
+        bucket[i] = b;
+        atomicAdd(&scnt[b], 1u);
+    }
+    __syncthreads();
+    if (tid < p && scnt[tid])
+        atomicAdd((unsigned long long*)&counts[tid],
+                  (unsigned long long)scnt[tid]);
+}
+
 /* ReduceToIndex — reference api/reduce_to_index.hpp with the
  * ReduceByIndex mapping (core/reduce_functional.hpp:84-149): keys are
  * dense indices; dense[key-begin] accumulates the u64 sum. Out-of-range
@@ -485,6 +685,89 @@ int t9_reduce_build(t9_context* ctx, const u64* d_keys, const u64* d_vals,
             hipLaunchKernelGGL(k_reduce_build, dim3(grid_for(n)),
                                dim3(256), 0, s, d_keys, d_vals, n, d_table,
                                cap, salt, d_error, mode));
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+int t9_reduce128_init(t9_context* ctx, u64* d_table, u64 cap,
+                      void* stream) {
+    (void)ctx;
+    if (!d_table || !is_pow2(cap)) return T9_EINVAL;
+    hipLaunchKernelGGL(k_reduce128_init, dim3(grid_for(cap)), dim3(256), 0,
+                       (hipStream_t)stream, d_table, cap);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+int t9_reduce128_build(t9_context* ctx, const u64* d_k1, const u64* d_k2,
+                       const u64* d_vals, u64 n, u64* d_table, u64 cap,
+                       u64 salt, u32* d_error, void* stream) {
+    (void)ctx;
+    if (!d_table || !d_error || !is_pow2(cap)) return T9_EINVAL;
+    hipStream_t s = (hipStream_t)stream;
+    HIP_TRY(hipMemsetAsync(d_error, 0, 4, s));
+    if (n == 0) return T9_OK;
+    if (!d_k1 || !d_k2) return T9_EINVAL;
+    /* d_vals may be NULL: every pair counts 1 (the word_count PreOp
+     * emits (word, 1) — word_count.hpp:43-45) */
+    const char* se = getenv("T9_LDS128_SLOTS");
+    const int slots = se ? atoi(se) : 2048;
+    const char* ge = getenv("T9_REDUCE_GRID");
+    u32 grid = ge ? (u32)atoi(ge) : grid_for(n);
+    if (!ge && grid > 1024) grid = 1024;
+    T9_PERF_WRAP(
+        s, "reduce_build",
+        if (slots >= 4096)
+            hipLaunchKernelGGL(k_reduce128_build_lds<4096>, dim3(grid),
+                               dim3(256), 0, s, d_k1, d_k2, d_vals, n,
+                               d_table, cap, salt, d_error);
+        else if (slots <= 1024)
+            hipLaunchKernelGGL(k_reduce128_build_lds<1024>, dim3(grid),
+                               dim3(256), 0, s, d_k1, d_k2, d_vals, n,
+                               d_table, cap, salt, d_error);
+        else
+            hipLaunchKernelGGL(k_reduce128_build_lds<2048>, dim3(grid),
+                               dim3(256), 0, s, d_k1, d_k2, d_vals, n,
+                               d_table, cap, salt, d_error));
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+int t9_reduce128_drain(t9_context* ctx, const u64* d_table, u64 cap,
+                       u64* d_ok1, u64* d_ok2, u64* d_ov, u64* d_out_n,
+                       void* stream) {
+    (void)ctx;
+    if (!d_table || !d_ok1 || !d_ok2 || !d_ov || !d_out_n || !is_pow2(cap))
+        return T9_EINVAL;
+    hipStream_t s = (hipStream_t)stream;
+    HIP_TRY(hipMemsetAsync(d_out_n, 0, 8, s));
+    hipLaunchKernelGGL(k_reduce128_drain, dim3(grid_for(cap)), dim3(256),
+                       0, s, d_table, cap, d_ok1, d_ok2, d_ov, d_out_n);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+int t9_hash2_of(t9_context* ctx, const u64* d_ids, u64 n, u64* d_k1,
+                u64* d_k2, void* stream) {
+    (void)ctx;
+    if (!d_ids || !d_k1 || !d_k2) return T9_EINVAL;
+    if (n == 0) return T9_OK;
+    hipLaunchKernelGGL(k_hash2_of, dim3(grid_for(n)), dim3(256), 0,
+                       (hipStream_t)stream, d_ids, n, d_k1, d_k2);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
+
+int t9_bucket_mod(t9_context* ctx, const u64* d_keys, u64 n, u32 p,
+                  u32* d_bucket, u64* d_counts, void* stream) {
+    (void)ctx;
+    if (!d_counts || p < 1 || p > 256) return T9_EINVAL;
+    hipStream_t s = (hipStream_t)stream;
+    HIP_TRY(hipMemsetAsync(d_counts, 0, p * 8, s));
+    if (n == 0) return T9_OK;
+    if (!d_keys || !d_bucket) return T9_EINVAL;
+    hipLaunchKernelGGL(k_bucket_mod, dim3(grid_for(n)), dim3(256), 0, s,
+                       d_keys, n, p, d_bucket, d_counts);
     T9_LAUNCH_CHECK();
     return T9_OK;
 }

@@ -65,6 +65,12 @@ _SIGS = {
     "t9_reduce_init": (i32, [vp, vp, u64, vp]),
     "t9_reduce_build": (i32, [vp, vp, vp, u64, vp, u64, u64, vp, vp]),
     "t9_reduce_drain": (i32, [vp, vp, u64, vp, vp, vp, vp]),
+    "t9_reduce128_init": (i32, [vp, vp, u64, vp]),
+    "t9_reduce128_build": (i32, [vp, vp, vp, vp, u64, vp, u64, u64, vp,
+                                 vp]),
+    "t9_reduce128_drain": (i32, [vp, vp, u64, vp, vp, vp, vp, vp]),
+    "t9_hash2_of": (i32, [vp, vp, u64, vp, vp, vp]),
+    "t9_bucket_mod": (i32, [vp, vp, u64, u32, vp, vp, vp]),
     "t9_zipf_tokens": (i32, [vp, vp, vp, u64, u64, u64, u64, vp]),
     "t9_merge_u64": (i32, [vp, vp, u64, vp, u64, vp, vp]),
     "t9_group_index_workspace": (u64, [u64]),

@@ -130,16 +130,25 @@ def zipf_cdf(N, s, q=0.0):
 class WordCount:
     """One rank's ReduceByKey (word_count) state: Zipf tokens -> local
     pre-reduce (ReducePrePhase role, core/reduce_pre_phase.hpp) -> hash
-    partition (Hash128to64 % p, core/reduce_functional.hpp:60-72) ->
-    one all-to-all -> final reduce (post phase)."""
+    partition (core/reduce_functional.hpp:60-72) -> one all-to-all ->
+    final reduce (post phase).
 
-    def __init__(self, n_total, vocab, s, seed, rank=0, world=1, device=0):
+    keys128 (the default): words are dictionary-encoded into two
+    independent 64-bit hashes and reduced on the 128-bit composite —
+    string-identity semantics (the reference compares full keys,
+    reduce_probing_hash_table.hpp:233); a forced single-hash collision
+    keeps counts separate. keys128=False is the bare-u64 ReducePair
+    path."""
+
+    def __init__(self, n_total, vocab, s, seed, rank=0, world=1, device=0,
+                 keys128=True):
         self.nat = Native(device=device, rank=rank, world=world)
         self.rank, self.world = rank, world
         if world > 1 and os.environ.get("T9_EXCHANGE", "t9") == "t9":
             import torch.distributed as dist
             bootstrap_comm(self.nat, dist, rank)
         self.n_total, self.seed = n_total, seed
+        self.keys128 = keys128
         base = n_total // world
         rem = n_total % world
         self.n_local = base + (1 if rank < rem else 0)
@@ -153,9 +162,19 @@ class WordCount:
                                  device="cuda")
         cap = 1 << max(10, int(math.ceil(math.log2(2 * vocab + 2))))
         self.cap = cap
-        # interleaved (key, sum) table: u64[2*(cap+1)]
-        self.d_tbl = torch.empty(2 * (cap + 1), dtype=torch.int64,
-                                 device="cuda")
+        if keys128:
+            self.d_k1 = torch.empty(self.n_local, dtype=torch.int64,
+                                    device="cuda")
+            self.d_k2 = torch.empty(self.n_local, dtype=torch.int64,
+                                    device="cuda")
+            # interleaved (k1, k2, sum) table: u64[3*cap]
+            self.d_tbl = torch.empty(3 * cap, dtype=torch.int64,
+                                     device="cuda")
+            self.d_ok2 = torch.empty(cap, dtype=torch.int64, device="cuda")
+        else:
+            # interleaved (key, sum) table: u64[2*(cap+1)]
+            self.d_tbl = torch.empty(2 * (cap + 1), dtype=torch.int64,
+                                     device="cuda")
         self.d_ok = torch.empty(cap + 1, dtype=torch.int64, device="cuda")
         self.d_ov = torch.empty(cap + 1, dtype=torch.int64, device="cuda")
         self.d_err = torch.empty(1, dtype=torch.int32, device="cuda")
@@ -165,6 +184,9 @@ class WordCount:
         self.nat.zipf_tokens(_ptr(self.d_toks), _ptr(self.d_cdf),
                              self.vocab, self.tok0, self.n_local,
                              self.seed, _stream())
+        if self.keys128:
+            self.nat.hash2_of(_ptr(self.d_toks), self.n_local,
+                              _ptr(self.d_k1), _ptr(self.d_k2), _stream())
 
     def _reduce(self, d_keys, d_vals, n, salt=0):
         nat, s = self.nat, _stream()
@@ -178,10 +200,47 @@ class WordCount:
         assert int(self.d_err.cpu().item()) == 0, "reduce table overflow"
         return self.d_ok[:m], self.d_ov[:m], m
 
-    def step(self):
-        """One full ReduceByKey of the (distributed) token stream. Returns
-        (keys tensor, vals tensor, m) of this rank's final pairs."""
+    def _reduce128(self, d_k1, d_k2, d_vals, n, salt=0):
+        """128-bit composite reduce; d_vals None = each pair counts 1."""
         nat, s = self.nat, _stream()
+        nat.reduce128_init(_ptr(self.d_tbl), self.cap, s)
+        nat.reduce128_build(_ptr(d_k1), _ptr(d_k2),
+                            _ptr(d_vals) if d_vals is not None else None,
+                            n, _ptr(self.d_tbl), self.cap, salt,
+                            _ptr(self.d_err), s)
+        nat.reduce128_drain(_ptr(self.d_tbl), self.cap, _ptr(self.d_ok),
+                            _ptr(self.d_ok2), _ptr(self.d_ov),
+                            _ptr(self.d_n), s)
+        m = int(self.d_n.cpu().item())
+        assert int(self.d_err.cpu().item()) == 0, "reduce table overflow"
+        return self.d_ok[:m], self.d_ok2[:m], self.d_ov[:m], m
+
+    def _exchange(self, arrays, send_counts, recv_counts, dist):
+        """all-to-all-v each u64 array with the same counts; returns the
+        received arrays."""
+        n_recv = int(recv_counts.sum())
+        outs = []
+        exchange = os.environ.get("T9_EXCHANGE", "t9")
+        for a in arrays:
+            r = torch.empty(max(n_recv, 1), dtype=torch.int64,
+                            device="cuda")
+            if exchange == "t9" or self.world == 1:
+                a2a(self.nat, a, send_counts, r, recv_counts, 8)
+            else:
+                dist.all_to_all_single(
+                    r[:n_recv], a[:int(send_counts.sum())],
+                    output_split_sizes=recv_counts.tolist(),
+                    input_split_sizes=send_counts.tolist())
+            outs.append(r)
+        return outs, n_recv
+
+    def step(self):
+        """One full ReduceByKey of the (distributed) token stream.
+        Returns (keys tensor, vals tensor, m) of this rank's final pairs
+        (keys128: (k1 tensor, k2 tensor, vals tensor, m))."""
+        nat, s = self.nat, _stream()
+        if self.keys128:
+            return self._step128()
         ok, ov, m = self._reduce(self.d_toks, self.d_ones, self.n_local)
         if self.world == 1 and not os.environ.get("T9_FORCE_DIST"):
             return ok.clone(), ov.clone(), m
@@ -208,22 +267,45 @@ class WordCount:
         send_counts = d_counts.cpu().numpy().astype(np.int64)
         recv_counts = exchange_counts(dist, send_counts, self.rank,
                                       self.world)
-        n_recv = int(recv_counts.sum())
-        rk = torch.empty(max(n_recv, 1), dtype=torch.int64, device="cuda")
-        rv = torch.empty(max(n_recv, 1), dtype=torch.int64, device="cuda")
-        exchange = os.environ.get("T9_EXCHANGE", "t9")
-        if exchange == "t9" or self.world == 1:
-            a2a(self.nat, ks, send_counts, rk, recv_counts, 8)
-            a2a(self.nat, vs, send_counts, rv, recv_counts, 8)
-        else:
-            dist.all_to_all_single(rk[:n_recv], ks[:m],
-                                   output_split_sizes=recv_counts.tolist(),
-                                   input_split_sizes=send_counts.tolist())
-            dist.all_to_all_single(rv[:n_recv], vs[:m],
-                                   output_split_sizes=recv_counts.tolist(),
-                                   input_split_sizes=send_counts.tolist())
+        (rk, rv), n_recv = self._exchange([ks, vs], send_counts,
+                                          recv_counts, dist)
         ok2, ov2, m2 = self._reduce(rk, rv, n_recv)
         return ok2.clone(), ov2.clone(), m2
+
+    def _step128(self):
+        nat, s = self.nat, _stream()
+        k1, k2, v, m = self._reduce128(self.d_k1, self.d_k2, None,
+                                       self.n_local)
+        if self.world == 1 and not os.environ.get("T9_FORCE_DIST"):
+            return k1.clone(), k2.clone(), v.clone(), m
+
+        import torch.distributed as dist
+        p = self.world
+        k1, k2, v = k1.clone(), k2.clone(), v.clone()
+        d_bucket = torch.empty(max(m, 1), dtype=torch.int32, device="cuda")
+        d_counts = torch.empty(p, dtype=torch.int64, device="cuda")
+        # partition on k1 (already the hash): bucket = k1 % p
+        nat.bucket_mod(_ptr(k1), m, p, _ptr(d_bucket), _ptr(d_counts), s)
+        d_perm = torch.empty(max(m, 1), dtype=torch.int32, device="cuda")
+        d_offs = torch.empty(p + 1, dtype=torch.int64, device="cuda")
+        d_ws = torch.empty(max(int(self.nat.ws("partition_idx", m)), 256),
+                           dtype=torch.uint8, device="cuda")
+        nat.partition_idx(_ptr(d_bucket), m, p, _ptr(d_perm), _ptr(d_offs),
+                          _ptr(d_ws), s)
+        g1 = torch.empty_like(k1)
+        g2 = torch.empty_like(k2)
+        gv = torch.empty_like(v)
+        if m:
+            nat.gather_records(_ptr(k1), _ptr(d_perm), m, 8, _ptr(g1), s)
+            nat.gather_records(_ptr(k2), _ptr(d_perm), m, 8, _ptr(g2), s)
+            nat.gather_records(_ptr(v), _ptr(d_perm), m, 8, _ptr(gv), s)
+        send_counts = d_counts.cpu().numpy().astype(np.int64)
+        recv_counts = exchange_counts(dist, send_counts, self.rank,
+                                      self.world)
+        (r1, r2, rv), n_recv = self._exchange([g1, g2, gv], send_counts,
+                                              recv_counts, dist)
+        o1, o2, ov, m2 = self._reduce128(r1, r2, rv, n_recv)
+        return o1.clone(), o2.clone(), ov.clone(), m2
 
     def close(self):
         self.nat.close()
