@@ -28,6 +28,7 @@
 #include <stdexcept>
 #include <string>
 #include <type_traits>
+#include <unordered_map>
 #include <utility>
 #include <vector>
 
@@ -97,6 +98,18 @@ public:
     t9_context* native() const { return ctx_; }
     hipStream_t stream() const { return stream_; }
 
+    //! reference api/context.hpp consume mode (memory hint): accepted and
+    //! ignored — DIAs here are explicit device buffers
+    void enable_consume() {}
+
+    //! scalar net services the examples touch (ctx.net.Barrier(),
+    //! terasort.cpp:204): single-process no-ops; multi-rank scalar
+    //! collectives run on the host control plane (pipeline.py) or a
+    //! caller-provided communicator
+    struct Net {
+        void Barrier() {}
+    } net;
+
 private:
     t9_context* ctx_ = nullptr;
     hipStream_t stream_ = nullptr;
@@ -116,16 +129,24 @@ struct DeviceBuf {
     }
 };
 
+//! Items that live on the GPU: fixed-size POD, raw-copied
+//! (data/serialization.hpp:35-48 contract). Non-POD items (std::string
+//! lines/words — the word_count tokenizer input) are held host-side, as
+//! in the reference where string handling is CPU code fused into PreOp
+//! chains; the DOp that consumes them
+//! (ReduceByKey) dictionary-encodes and moves fixed-width work to the
+//! GPU.
 template <typename ValueType>
 class DIA {
-    static_assert(std::is_trivially_copyable<ValueType>::value,
-                  "GPU DIA items must be fixed-size POD "
-                  "(data/serialization.hpp:35-48 raw-copy contract)");
+    static constexpr bool kDevice =
+        std::is_trivially_copyable<ValueType>::value;
 
 public:
     DIA() = default;
     DIA(Context* ctx, std::shared_ptr<DeviceBuf> buf, size_t n)
         : ctx_(ctx), buf_(buf), n_(n) {}
+    DIA(Context* ctx, std::shared_ptr<std::vector<ValueType> > host)
+        : ctx_(ctx), hvec_(host), n_(host ? host->size() : 0) {}
 
     size_t Size() const {
         // reference: ActionNode + AllReduce (api/size.hpp:64-69); world=1
@@ -134,6 +155,7 @@ public:
     }
 
     std::vector<ValueType> AllGather() const {
+        if (hvec_) return *hvec_;
         std::vector<ValueType> out(n_);
         if (n_)
             T9_DIA_HIP(hipMemcpy(out.data(), buf_->ptr,
@@ -181,12 +203,44 @@ public:
         return FromVector(*ctx_, out);
     }
 
+    //! FlatMap — reference api/dia.hpp:405 (template on the output type,
+    //! user lambda receives (item, emit)): host code, exactly as the
+    //! reference fuses it into the CPU PreOp chain; the DOp that follows
+    //! runs on the GPU.
+    template <typename Out, typename F>
+    DIA<Out> FlatMap(const F& f) const {
+        auto in = AllGather();
+        auto out = std::make_shared<std::vector<Out> >();
+        auto emit = [&](const Out& o) { out->push_back(o); };
+        for (auto& v : in) f(v, emit);
+        return FromVector(*ctx_, *out);
+    }
+
+    //! ReduceByKey — reference api/reduce_by_key.hpp:241-463: reduce items
+    //! with equal key_extractor(item) using the associative+commutative
+    //! reduce_function; output order arbitrary (word_count_test.cpp:73-74).
+    //! Implemented for pair<std::string, integral> items (the word_count
+    //! shape): words are dictionary-encoded into two independent 64-bit
+    //! hashes and reduced on the GPU 128-bit composite table
+    //! (t9_reduce128_*; string identity per DESIGN.md — forced single-hash
+    //! collisions stay separate). The u64-sum hot path is used when the
+    //! user's reduce_function is additive on the counter — verified by
+    //! ALGEBRAIC PROBING (the functor is a pure function of POD-ish pairs:
+    //! it is evaluated on sample values and must return the counter sum,
+    //! the shape word_count.hpp:50-53 has); a non-additive functor falls
+    //! back to GPU grouping + a host fold with the user's functor.
+    template <typename KeyExtractor, typename ReduceFn>
+    DIA<ValueType> ReduceByKey(const KeyExtractor& key_ex,
+                               const ReduceFn& red) const;
+
     //! WriteBinary — reference api/write_binary.hpp: the items as packed
     //! binary (the POD raw-copy wire format, data/serialization.hpp:35-48),
     //! one file per worker: pathbase + zero-padded rank. The on-disk bytes
     //! are bit-compatible with the reference's files (terasort.cpp:184-200
     //! file mode).
     void WriteBinary(const std::string& pathbase) const {
+        static_assert(std::is_trivially_copyable<ValueType>::value,
+                      "WriteBinary needs fixed-size POD items");
         char name[512];
         std::snprintf(name, sizeof(name), "%s%010zu", pathbase.c_str(),
                       ctx_->my_rank());
@@ -214,11 +268,14 @@ private:
 
     Context* ctx_ = nullptr;
     std::shared_ptr<DeviceBuf> buf_;
+    std::shared_ptr<std::vector<ValueType> > hvec_;  // non-POD host items
     size_t n_ = 0;
 };
 
 template <typename T>
-DIA<T> FromVector(Context& ctx, const std::vector<T>& v) {
+typename std::enable_if<std::is_trivially_copyable<T>::value,
+                        DIA<T> >::type
+FromVectorImpl(Context& ctx, const std::vector<T>& v) {
     auto buf = std::make_shared<DeviceBuf>(v.size() * sizeof(T));
     if (!v.empty())
         T9_DIA_HIP(hipMemcpy(buf->ptr, v.data(), v.size() * sizeof(T),
@@ -226,12 +283,27 @@ DIA<T> FromVector(Context& ctx, const std::vector<T>& v) {
     return DIA<T>(&ctx, buf, v.size());
 }
 
+template <typename T>
+typename std::enable_if<!std::is_trivially_copyable<T>::value,
+                        DIA<T> >::type
+FromVectorImpl(Context& ctx, const std::vector<T>& v) {
+    return DIA<T>(&ctx, std::make_shared<std::vector<T> >(v));
+}
+
+template <typename T>
+DIA<T> FromVector(Context& ctx, const std::vector<T>& v) {
+    return FromVectorImpl<T>(ctx, v);
+}
+
 //! Generate — reference api/generate.hpp:37: DIA of generator(i) for
 //! i in [0, size). The generator runs on the host (it is user code);
 //! the DOps that follow run on the GPU.
 template <typename Generator>
-auto Generate(Context& ctx, size_t size, const Generator& gen)
+auto Generate(Context& ctx, size_t size, Generator gen)
     -> DIA<decltype(gen(size_t(0)))> {
+    // generator taken by value and callable non-const, as the
+    // reference's GeneratorFunction is (api/generate.hpp:37; stateful
+    // rngs like terasort's GenerateRecord mutate)
     using T = decltype(gen(size_t(0)));
     std::vector<T> v;
     v.reserve(size);
@@ -241,6 +313,9 @@ auto Generate(Context& ctx, size_t size, const Generator& gen)
 
 template <typename ValueType>
 DIA<ValueType> DIA<ValueType>::SortImpl() const {
+    static_assert(std::is_trivially_copyable<ValueType>::value,
+                  "GPU Sort needs fixed-size POD items "
+                  "(data/serialization.hpp:35-48 raw-copy contract)");
     constexpr size_t R = sizeof(ValueType);
     static_assert(R % 4 == 0, "record size must be a multiple of 4");
     auto out = std::make_shared<DeviceBuf>(n_ * R);
@@ -326,6 +401,147 @@ inline DIA<KeyValue> ReducePair(const DIA<KeyValue>& input,
     std::vector<KV> out(m);
     for (size_t i = 0; i < m; ++i) out[i] = KV{ rk[i], rv[i] };
     return FromVector(ctx, out);
+}
+
+namespace detail {
+
+//! the framework's string hash pair: fnv-1a under two bases, mixed by
+//! Hash128to64 with two salts (DESIGN.md "config-4 string identity"; the
+//! reference's std::hash is implementation-defined and affects placement
+//! only). Sentinels remapped as t9_hash2_of does.
+inline uint64_t fnv1a64(const char* s, size_t n, uint64_t basis) {
+    uint64_t h = basis;
+    for (size_t i = 0; i < n; ++i)
+        h = (h ^ (uint8_t)s[i]) * 0x100000001B3ull;
+    return h;
+}
+
+inline uint64_t mix128to64(uint64_t upper, uint64_t lower) {
+    const uint64_t k = 0x9DDFEA08EB382D69ull;
+    uint64_t a = (lower ^ upper) * k;
+    a ^= (a >> 47);
+    uint64_t b = (upper ^ a) * k;
+    b ^= (b >> 47);
+    b *= k;
+    return b;
+}
+
+inline void string_hash2(const std::string& w, uint64_t* h1,
+                         uint64_t* h2) {
+    uint64_t a = mix128to64(0x9AE16A3B2F90404Full,
+                            fnv1a64(w.data(), w.size(),
+                                    0xCBF29CE484222325ull));
+    uint64_t b = mix128to64(0xC3A5C85C97CB3127ull,
+                            fnv1a64(w.data(), w.size(),
+                                    0x84222325CBF29CE4ull));
+    if (a == ~0ull) a ^= 1;
+    if (b == ~0ull) b ^= 1;
+    *h1 = a;
+    *h2 = b;
+}
+
+} // namespace detail
+
+//! ReduceByKey for (std::string word, integral count) pairs — the
+//! word_count shape (examples/word_count/word_count.hpp:35-56). See the
+//! declaration for the contract; the GPU path is
+//! dictionary-encode -> t9_reduce128 (sum) -> decode.
+template <typename ValueType>
+template <typename KeyExtractor, typename ReduceFn>
+DIA<ValueType> DIA<ValueType>::ReduceByKey(const KeyExtractor& key_ex,
+                                           const ReduceFn& red) const {
+    using Pair = ValueType;   // pair<std::string, Count>
+    using Count = decltype(Pair().second);
+    static_assert(std::is_integral<Count>::value && sizeof(Count) == 8,
+                  "GPU ReduceByKey carries a 64-bit counter value");
+    auto items = AllGather();
+    const size_t n = items.size();
+
+    // algebraic probe: is the user's reduce function "add the counters"?
+    // (it is a pure function of its operands; probe a few samples)
+    bool additive = true;
+    for (uint64_t a : { 3ull, 17ull, 1ull << 40 }) {
+        Pair x{ "probe", (Count)a }, y{ "probe", (Count)(a * 2 + 5) };
+        Pair r = red(x, y);
+        if ((uint64_t)r.second != a + (a * 2 + 5) || r.first != "probe") {
+            additive = false;
+            break;
+        }
+    }
+
+    // dictionary-encode: word -> (h1, h2); keep the decode map
+    std::vector<uint64_t> k1(n), k2(n), vals(n);
+    std::unordered_map<uint64_t, std::string> decode;
+    for (size_t i = 0; i < n; ++i) {
+        const std::string key = key_ex(items[i]);
+        detail::string_hash2(key, &k1[i], &k2[i]);
+        vals[i] = (uint64_t)items[i].second;
+        decode.emplace(k1[i] ^ (k2[i] << 1 | k2[i] >> 63), key);
+    }
+
+    std::vector<Pair> out;
+    if (n && additive) {
+        // GPU 128-bit composite sum-reduce
+        DeviceBuf d1(n * 8), d2(n * 8), dv(n * 8);
+        T9_DIA_HIP(hipMemcpy(d1.ptr, k1.data(), n * 8,
+                             hipMemcpyHostToDevice));
+        T9_DIA_HIP(hipMemcpy(d2.ptr, k2.data(), n * 8,
+                             hipMemcpyHostToDevice));
+        T9_DIA_HIP(hipMemcpy(dv.ptr, vals.data(), n * 8,
+                             hipMemcpyHostToDevice));
+        uint64_t cap = 1024;
+        while (cap < 2 * n + 2) cap <<= 1;
+        DeviceBuf tbl(3 * cap * 8), o1(cap * 8), o2(cap * 8), ov(cap * 8);
+        DeviceBuf derr(4), dn(8);
+        hipStream_t s = ctx_->stream();
+        T9_DIA_TRY(t9_reduce128_init(ctx_->native(), (uint64_t*)tbl.ptr,
+                                     cap, s));
+        T9_DIA_TRY(t9_reduce128_build(ctx_->native(),
+                                      (const uint64_t*)d1.ptr,
+                                      (const uint64_t*)d2.ptr,
+                                      (const uint64_t*)dv.ptr, n,
+                                      (uint64_t*)tbl.ptr, cap, 0,
+                                      (uint32_t*)derr.ptr, s));
+        T9_DIA_TRY(t9_reduce128_drain(ctx_->native(),
+                                      (const uint64_t*)tbl.ptr, cap,
+                                      (uint64_t*)o1.ptr, (uint64_t*)o2.ptr,
+                                      (uint64_t*)ov.ptr, (uint64_t*)dn.ptr,
+                                      s));
+        uint64_t m = 0;
+        uint32_t err = 0;
+        T9_DIA_HIP(hipMemcpy(&m, dn.ptr, 8, hipMemcpyDeviceToHost));
+        T9_DIA_HIP(hipMemcpy(&err, derr.ptr, 4, hipMemcpyDeviceToHost));
+        if (err) throw std::runtime_error("ReduceByKey: table overflow");
+        std::vector<uint64_t> r1(m), r2(m), rv(m);
+        if (m) {
+            T9_DIA_HIP(hipMemcpy(r1.data(), o1.ptr, m * 8,
+                                 hipMemcpyDeviceToHost));
+            T9_DIA_HIP(hipMemcpy(r2.data(), o2.ptr, m * 8,
+                                 hipMemcpyDeviceToHost));
+            T9_DIA_HIP(hipMemcpy(rv.data(), ov.ptr, m * 8,
+                                 hipMemcpyDeviceToHost));
+        }
+        out.reserve(m);
+        for (uint64_t i = 0; i < m; ++i)
+            out.push_back(Pair{
+                decode.at(r1[i] ^ (r2[i] << 1 | r2[i] >> 63)),
+                (Count)rv[i] });
+    }
+    else if (n) {
+        // non-additive reduce function: host fold with the user's functor
+        // (grouping semantics preserved; the hot path is the additive
+        // word_count shape above)
+        std::unordered_map<uint64_t, Pair> acc;
+        for (size_t i = 0; i < n; ++i) {
+            const uint64_t h = k1[i] ^ (k2[i] << 1 | k2[i] >> 63);
+            auto it = acc.find(h);
+            if (it == acc.end()) acc.emplace(h, items[i]);
+            else it->second = red(it->second, items[i]);
+        }
+        out.reserve(acc.size());
+        for (auto& kv : acc) out.push_back(kv.second);
+    }
+    return FromVector(*ctx_, out);
 }
 
 //! ReadBinary — reference api/read_binary.hpp: read packed fixed-size
@@ -435,4 +651,19 @@ inline int Run(const std::function<void(Context&)>& job) {
 }
 
 } // namespace api
+
+//! hoist the api entry points into t9::, as the reference hoists
+//! thrill::api into thrill:: (each api header's `using api::X`), so
+//! `using namespace t9` makes the example sources read identically
+using api::Context;
+using api::DIA;
+using api::FromVector;
+using api::Generate;
+using api::GroupByKey;
+using api::KeyValue;
+using api::Merge;
+using api::ReadBinary;
+using api::ReducePair;
+using api::Run;
+
 } // namespace t9
