@@ -26,6 +26,10 @@ REC = 100
 N = int(os.environ.get("T9_TIES_N", 10 * 1024**3 // REC))
 
 
+CLASSES = ["extract", "hist_pairs", "pair_scatter", "lds_sort", "gather",
+           "tie_partition", "tie_prescan"]
+
+
 def time_sort(nat, din, dout, w, reps=3):
     s = G.stream()
     nat.sort_records(G.ptr(din), G.ptr(dout), N, REC, 10, G.ptr(w), s)
@@ -34,7 +38,20 @@ def time_sort(nat, din, dout, w, reps=3):
     for _ in range(reps):
         nat.sort_records(G.ptr(din), G.ptr(dout), N, REC, 10, G.ptr(w), s)
     torch.cuda.synchronize()
-    return (time.perf_counter() - t0) / reps
+    dt = (time.perf_counter() - t0) / reps
+    # per-class breakdown from one extra instrumented run
+    nat.perf_reset()
+    nat.perf_enable(True)
+    nat.sort_records(G.ptr(din), G.ptr(dout), N, REC, 10, G.ptr(w), s)
+    torch.cuda.synchronize()
+    nat.perf_enable(False)
+    br = {}
+    for c in CLASSES:
+        ms, cnt = nat.perf_read(c)
+        if cnt:
+            br[c] = round(ms, 2)
+    nat.perf_reset()
+    return dt, br
 
 
 def main():
@@ -46,17 +63,21 @@ def main():
 
     # (a) uniform
     nat.gen_records(G.ptr(din), 0, N, 0x7421, G.stream())
-    out["uniform_s"] = round(time_sort(nat, din, dout, w), 4)
+    out["uniform_s"], out["uniform_br"] = time_sort(nat, din, dout, w)
+    out["uniform_s"] = round(out["uniform_s"], 4)
 
     # (b) all-identical records
     din.view(torch.uint8).fill_(0xA7)
-    out["identical_s"] = round(time_sort(nat, din, dout, w), 4)
+    out["identical_s"], out["identical_br"] = time_sort(nat, din, dout, w)
+    out["identical_s"] = round(out["identical_s"], 4)
 
     # (c) shared prefix, random tails: regenerate then stamp bytes 0..7
     nat.gen_records(G.ptr(din), 0, N, 0x7421, G.stream())
     v = din.view(N, REC)
     v[:, :8] = 0x55
-    out["shared_prefix_s"] = round(time_sort(nat, din, dout, w), 4)
+    out["shared_prefix_s"], out["shared_prefix_br"] = \
+        time_sort(nat, din, dout, w)
+    out["shared_prefix_s"] = round(out["shared_prefix_s"], 4)
 
     out["identical_x"] = round(out["identical_s"] / out["uniform_s"], 2)
     out["shared_prefix_x"] = round(out["shared_prefix_s"] /

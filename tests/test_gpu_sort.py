@@ -163,6 +163,51 @@ def test_sort_records_prefix_collisions(nat, oracle):
     assert np.array_equal(got, oracle.sort_records(recs))
 
 
+def test_sort_records_fused_extract_hist(nat, oracle):
+    # the fused extract+pass1-hist path (default at MSB-dispatch sizes);
+    # engage it at test scale via T9_SORT_ALGO=msb
+    import os
+    os.environ["T9_SORT_ALGO"] = "msb"
+    try:
+        n = 300_000
+        recs = oracle.gen_records(n, seed=77)
+        din = G.dev(recs.reshape(-1))
+        dout = G.empty(n * 100, np.uint8)
+        w = G.ws(nat.ws("sort_records", n, 100))
+        nat.sort_records(G.ptr(din), G.ptr(dout), n, 100, 10, G.ptr(w),
+                         G.stream())
+        got = G.host(dout, np.uint8).reshape(n, 100)
+        assert np.array_equal(got, oracle.sort_records(recs))
+        # and with the fusion disabled: identical result
+        os.environ["T9_EXTRACT_HIST"] = "0"
+        nat.sort_records(G.ptr(din), G.ptr(dout), n, 100, 10, G.ptr(w),
+                         G.stream())
+        got2 = G.host(dout, np.uint8).reshape(n, 100)
+        assert np.array_equal(got2, got)
+    finally:
+        del os.environ["T9_SORT_ALGO"]
+        os.environ.pop("T9_EXTRACT_HIST", None)
+
+
+def test_gather_scatter_variant_parity(nat, oracle):
+    # T9_GATHER_VARIANT=5 (sequential-read random-write scatter probe)
+    import os
+    n = 200_000
+    recs = oracle.gen_records(n, seed=88)
+    rng = np.random.default_rng(88)
+    idx = rng.permutation(n).astype(np.uint32)
+    din, didx = G.dev(recs.reshape(-1)), G.dev(idx)
+    dout = G.empty(n * 100, np.uint8)
+    os.environ["T9_GATHER_VARIANT"] = "5"
+    try:
+        nat.gather_records(G.ptr(din), G.ptr(didx), n, 100, G.ptr(dout),
+                           G.stream())
+    finally:
+        del os.environ["T9_GATHER_VARIANT"]
+    got = G.host(dout, np.uint8).reshape(n, 100)
+    assert np.array_equal(got, recs[idx])
+
+
 def test_sort_records_many_tie_runs(nat, oracle):
     # the on-device tie sort's regrouping pass: many distinct equal-prefix
     # runs interleaved with unique-prefix records; the final stable

@@ -23,6 +23,9 @@ extern "C" int t9i_sort_recs_msb(t9_context*, const u8*, u32, u64*, u32*,
 extern "C" u64 t9_partition_idx_workspace(u64 n);
 extern "C" int t9_partition_idx(t9_context*, const u32*, u64, u32, u32*,
                                 u64*, void*, void*);
+extern "C" int t9i_sort_pairs_msb_ph(t9_context*, u64*, u32*, u64, void*,
+                                     void*);
+extern "C" u32* t9i_msb_pass1_hist(void* d_workspace, u64 n);
 
 /* ------------------------------------------------------------------ */
 
@@ -287,6 +290,82 @@ __global__ __launch_bounds__(256) void k_zipf_tokens(
     }
 }
 
+/* Fused extract + MSB pass-1 histogram: one block per 8192-record tile
+ * (the MSB pipeline's T9_MSB_TILE), 32 sub-tiles of 256 records staged
+ * through LDS by coalesced word loads — every input line is read ONCE,
+ * coalesced (the standalone extract's per-lane strided loads reach only
+ * ~60% of streaming rate on 100-B records), and the tile's byte-7 digit
+ * counts come out of the same pass, eliminating the separate pass-1
+ * histogram read (k_hist_msb). Output layout identical to
+ * k_extract_key64 + k_hist_msb<0> at shift 56. */
+template <int RW, bool LE>
+__global__ __launch_bounds__(256) void k_extract_hist(
+    const u8* __restrict__ recs, u64 n, u64* __restrict__ keys,
+    u32* __restrict__ idx, u32* __restrict__ hist) {
+    constexpr int SUBREC = 256;                /* records per LDS stage */
+    constexpr int SUBW = SUBREC * RW;          /* words per stage */
+    __shared__ u32 s_buf[SUBW];
+    __shared__ u32 s_cnt[256];
+    const u32 tid = threadIdx.x;
+    const u64 tile0 = (u64)blockIdx.x * 8192;
+    const u32 tn = (u32)((n - tile0 < 8192) ? (n - tile0) : 8192);
+    s_cnt[tid] = 0;
+    __syncthreads();
+    const u32* rin = (const u32*)recs;
+    for (u32 s0 = 0; s0 < tn; s0 += SUBREC) {
+        const u32 sn = (tn - s0 < SUBREC) ? tn - s0 : SUBREC;
+        const u64 w0 = (tile0 + s0) * RW;
+        const u32 wn = sn * RW;
+        for (u32 w = tid; w < wn; w += 256) s_buf[w] = rin[w0 + w];
+        __syncthreads();
+        if (tid < sn) {
+            u64 k;
+            if (LE) {
+                k = ((u64)s_buf[tid * RW + 1] << 32) | s_buf[tid * RW];
+            }
+            else {
+                k = ((u64)__builtin_bswap32(s_buf[tid * RW]) << 32) |
+                    __builtin_bswap32(s_buf[tid * RW + 1]);
+            }
+            const u64 gi = tile0 + s0 + tid;
+            keys[gi] = k;
+            idx[gi] = (u32)gi;
+            atomicAdd(&s_cnt[(u32)(k >> 56)], 1u);
+        }
+        __syncthreads();
+    }
+    hist[(u64)blockIdx.x * 256 + tid] = s_cnt[tid];
+}
+
+/* scatter-records experiment (gather-wall probe): read the input
+ * SEQUENTIALLY (every line fetched once, fully used) and write each
+ * word to its record's destination: out[inv[rec]]. Mirrors
+ * k_gather_records with the random side moved from reads to writes —
+ * measures whether CDNA4 partial-line writes (byte-enable sectors) beat
+ * random-line read amplification. */
+template <int RW>
+__global__ __launch_bounds__(256) void k_scatter_records(
+    const u8* __restrict__ recs, const u32* __restrict__ inv, u64 n,
+    u8* __restrict__ out) {
+    const u64 total_words = n * RW;
+    const u64 stride = (u64)gridDim.x * 256;
+    const u32* rin = (const u32*)recs;
+    u32* rout = (u32*)out;
+    for (u64 g = (u64)blockIdx.x * 256 + threadIdx.x; g < total_words;
+         g += stride) {
+        const u64 rec = g / RW;
+        const u32 off = (u32)(g - rec * RW);
+        rout[(u64)inv[rec] * RW + off] = rin[g];
+    }
+}
+
+__global__ __launch_bounds__(256) void k_invert_perm(
+    const u32* __restrict__ idx, u64 n, u32* __restrict__ inv) {
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 i = (u64)blockIdx.x * 256 + threadIdx.x; i < n; i += stride)
+        inv[idx[i]] = (u32)i;
+}
+
 /* ------------------------------------------------------------------ *
  * host orchestration
  * ------------------------------------------------------------------ */
@@ -372,12 +451,34 @@ int t9_gather_records(t9_context* ctx, const u8* d_recs, const u32* d_idx,
     hipStream_t s = (hipStream_t)stream;
     const char* gv = getenv("T9_GATHER_VARIANT");
     const int var = gv ? atoi(gv) : 3;   /* 1 strided, 2 +nt, 3 span,
-                                            4 span+nt; measured best:
-                                            span, grid 16384 */
+                                            4 span+nt, 5 scatter probe
+                                            (sequential read + random
+                                            write via inverse perm);
+                                            measured best: span,
+                                            grid 16384 */
     const char* gg = getenv("T9_GATHER_GRID");
     const u32 gcap = gg ? (u32)atoi(gg) : 16384;
     u64 want = t9_ceil_div(n * rw, 256);
     const dim3 grid((u32)((want < gcap) ? (want ? want : 1) : gcap));
+    if (var == 5 && rw == 25) {
+        /* scatter experiment: build the inverse permutation, then read
+         * the input sequentially and write each record to its
+         * destination (probe for the random-read-line amplification vs
+         * partial-line-write cost tradeoff) */
+        u32* d_inv = nullptr;
+        HIP_TRY(hipMallocAsync((void**)&d_inv, n * 4, s));
+        u64 iwant = t9_ceil_div(n, 256);
+        hipLaunchKernelGGL(k_invert_perm,
+                           dim3((u32)((iwant < 4096) ? iwant : 4096)),
+                           dim3(256), 0, s, d_idx, n, d_inv);
+        T9_PERF_WRAP(s, "gather",
+                     hipLaunchKernelGGL((k_scatter_records<25>), grid,
+                                        dim3(256), 0, s, d_recs, d_inv, n,
+                                        d_out));
+        HIP_TRY(hipFreeAsync(d_inv, s));
+        T9_LAUNCH_CHECK();
+        return T9_OK;
+    }
     T9_PERF_WRAP(
         s, "gather",
         if (rw == 25 && var == 1)
@@ -485,15 +586,55 @@ static int sort_records_impl(t9_context* ctx, const u8* d_in, u8* d_out,
     p += t9_sort_pairs_workspace(n);
     u32* d_ntied = (u32*)p;
 
-    /* fused-extract measured SLOWER than extract + packed-key passes on
-     * 100 B records (byte reads at record stride fetch a full 64 B line
-     * per lane: 17.3 vs 15.5 ms per 10 GiB sort) — kept opt-in for
-     * re-evaluation with wider tiles. */
+    /* extract strategy ladder:
+     * (1) DEFAULT for MSB-dispatch sizes: fused extract+hist — one
+     *     LDS-staged coalesced pass over the records writes keys, the
+     *     index iota AND the MSB pass-1 byte-7 histogram
+     *     (k_extract_hist), so the standalone extract's strided loads
+     *     and the pass-1 histogram read both disappear. T9_EXTRACT_HIST=0
+     *     disables.
+     * (2) the round-1 per-lane fused path (T9_FUSED_EXTRACT=1): measured
+     *     SLOWER (byte reads at record stride, 17.3 vs 15.5 ms/10 GiB) —
+     *     kept as a recorded negative result.
+     * (3) plain extract + pair-sort dispatch (small n / LSD override). */
+    const char* algo = getenv("T9_SORT_ALGO");
+    bool msb_dispatch = n >= (1ull << 22);
+    if (algo && strcmp(algo, "lsd") == 0) msb_dispatch = false;
+    if (algo && strcmp(algo, "msb") == 0) msb_dispatch = n >= (1ull << 14);
+    const char* ehe = getenv("T9_EXTRACT_HIST");
+    const u32 rw = rec_size / 4;
+    const bool eh = !(ehe && ehe[0] == '0') && msb_dispatch &&
+                    (rw == 25 || rw == 32);
     const char* fe = getenv("T9_FUSED_EXTRACT");
     bool fused = fe && atoi(fe) && n >= (1ull << 14) &&
                  (rec_size == 100 || rec_size == 128);
     int rc;
-    if (fused && !le) {
+    if (eh && !fused) {
+        u32* hist = t9i_msb_pass1_hist(pair_ws, n);
+        const u64 B = t9_ceil_div(n, 8192);
+        T9_PERF_WRAP(
+            s, "extract",
+            if (rw == 25 && !le)
+                hipLaunchKernelGGL((k_extract_hist<25, false>),
+                                   dim3((u32)B), dim3(256), 0, s, d_in, n,
+                                   d_keys, d_idx, hist);
+            else if (rw == 25)
+                hipLaunchKernelGGL((k_extract_hist<25, true>),
+                                   dim3((u32)B), dim3(256), 0, s, d_in, n,
+                                   d_keys, d_idx, hist);
+            else if (!le)
+                hipLaunchKernelGGL((k_extract_hist<32, false>),
+                                   dim3((u32)B), dim3(256), 0, s, d_in, n,
+                                   d_keys, d_idx, hist);
+            else
+                hipLaunchKernelGGL((k_extract_hist<32, true>),
+                                   dim3((u32)B), dim3(256), 0, s, d_in, n,
+                                   d_keys, d_idx, hist));
+        T9_LAUNCH_CHECK();
+        rc = t9i_sort_pairs_msb_ph(ctx, d_keys, d_idx, n, pair_ws,
+                                   stream);
+    }
+    else if (fused && !le) {
         rc = t9i_sort_recs_msb(ctx, d_in, rec_size, d_keys, d_idx, n,
                                pair_ws, stream);
     }
@@ -539,8 +680,10 @@ static int sort_records_impl(t9_context* ctx, const u8* d_in, u8* d_out,
 
         const u32 ngrid =
             (u32)((t9_ceil_div(n, 256) < 4096) ? t9_ceil_div(n, 256) : 4096);
-        hipLaunchKernelGGL(k_tie_flags, dim3(ngrid), dim3(256), 0, s,
-                           d_keys, n, t_bucket);
+        T9_PERF_WRAP(s, "tie_partition",
+                     hipLaunchKernelGGL(k_tie_flags, dim3(ngrid),
+                                        dim3(256), 0, s, d_keys, n,
+                                        t_bucket));
         T9_LAUNCH_CHECK();
         rc = t9_partition_idx(ctx, t_bucket, n, 2, t_perm, t_offs, t_pws,
                               stream);
@@ -564,9 +707,11 @@ static int sort_records_impl(t9_context* ctx, const u8* d_in, u8* d_out,
             u32 hflags[64];
             if (nc <= 63) {
                 HIP_TRY(hipMemsetAsync(t_flag, 0, 64 * 4, s));
-                hipLaunchKernelGGL(k_tie_prescan, dim3(mgrid), dim3(256),
-                                   0, s, d_in, t_idx, m, rec_size, nc,
-                                   t_flag);
+                T9_PERF_WRAP(s, "tie_prescan",
+                             hipLaunchKernelGGL(k_tie_prescan,
+                                                dim3(mgrid), dim3(256),
+                                                0, s, d_in, t_idx, m,
+                                                rec_size, nc, t_flag));
                 T9_LAUNCH_CHECK();
                 HIP_TRY(hipMemcpyAsync(hflags, t_flag, (nc + 1) * 4,
                                        hipMemcpyDeviceToHost, s));

@@ -967,7 +967,8 @@ extern "C" u64 t9i_sort_pairs_msb_workspace(u64 n) { return msb_ws_bytes(n); }
 template <bool HAS_VAL, int RW = 0>
 static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
                          u64* d_keys, u32* d_vals, u64 n,
-                         void* d_workspace, void* stream) {
+                         void* d_workspace, void* stream,
+                         bool pre_hist = false) {
     hipStream_t s = (hipStream_t)stream;
     MsbWs w = carve_msb((char*)d_workspace, n);
 
@@ -975,10 +976,13 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
     {
         const u64 B = w.B1;
         const u64 Bc = t9_ceil_div(B, T9_SCAN_CHUNK);
-        T9_PERF_WRAP(s, "hist_pairs",
-                     hipLaunchKernelGGL((k_hist_msb<RW>), dim3((u32)B),
-                                        dim3(256), 0, s, pass1_src, n, 56,
-                                        w.hist));
+        /* pre_hist: the caller already filled w.hist for shift 56 (the
+         * fused extract pass, t9i_extract_hist) */
+        if (!pre_hist)
+            T9_PERF_WRAP(s, "hist_pairs",
+                         hipLaunchKernelGGL((k_hist_msb<RW>), dim3((u32)B),
+                                            dim3(256), 0, s, pass1_src, n,
+                                            56, w.hist));
         hipLaunchKernelGGL(k_colsum, dim3((u32)Bc), dim3(256), 0, s, w.hist,
                            B, w.chunkpart);
         hipLaunchKernelGGL(k_chunkscan, dim3(1), dim3(256), 0, s,
@@ -1323,6 +1327,21 @@ extern "C" int t9i_sort_pairs_msb(t9_context* ctx, u64* d_keys,
                                   void* stream) {
     return sort_msb_impl<true>(ctx, d_keys, d_keys, d_vals, n, d_workspace,
                                stream);
+}
+
+/* pre-hist variant: the caller (fused extract, t9_records.hip) already
+ * wrote the pass-1 byte-7 histogram rows into the workspace slot
+ * returned by t9i_msb_pass1_hist */
+extern "C" int t9i_sort_pairs_msb_ph(t9_context* ctx, u64* d_keys,
+                                     u32* d_vals, u64 n,
+                                     void* d_workspace, void* stream) {
+    return sort_msb_impl<true>(ctx, d_keys, d_keys, d_vals, n, d_workspace,
+                               stream, true);
+}
+
+extern "C" u32* t9i_msb_pass1_hist(void* d_workspace, u64 n) {
+    MsbWs w = carve_msb((char*)d_workspace, n);
+    return w.hist;
 }
 
 extern "C" int t9i_sort_keys_msb(t9_context* ctx, u64* d_keys, u64 n,
