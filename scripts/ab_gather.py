@@ -37,7 +37,7 @@ for var in [3, 4, 5]:
     torch.cuda.synchronize()
     ms = (time.perf_counter() - t0) / 3 * 1e3
     res[f"var{var}_ms"] = round(ms, 3)
-    res[f"var{var}_algo_TBps"] = round(204e-3 * N / ms / 1e9, 2)
+    res[f"var{var}_algo_TBps"] = round(204.0 * N / (ms * 1e-3) / 1e12, 2)
 del os.environ["T9_GATHER_VARIANT"]
 print(json.dumps(res), flush=True)
 nat.close()

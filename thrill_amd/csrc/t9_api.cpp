@@ -86,7 +86,18 @@ int t9_alltoall(t9_context* ctx, const void* d_send, const u64* send_counts,
     if (!ctx || !d_send || !d_recv || !send_counts || !send_displs ||
         !recv_counts || !recv_displs || elem_size == 0)
         return T9_EINVAL;
-    if (ctx->world == 1) {
+    /* The rank's own share bypasses RCCL: a plain device copy on the same
+     * stream is both faster (no protocol round trip for ~1/p of the data)
+     * and avoids the self-send path entirely — a 10.7 GB RCCL
+     * self-exchange was measured as a hang in round 1 (pipeline.py
+     * history, commit 551d307); T9_A2A_SELF=nccl restores the RCCL
+     * self-send for investigation (including at world == 1, where it
+     * overrides the loopback shortcut if a communicator exists). */
+    static const bool self_nccl = [] {
+        const char* e = getenv("T9_A2A_SELF");
+        return e && strcmp(e, "nccl") == 0;
+    }();
+    if (ctx->world == 1 && !(self_nccl && ctx->comm)) {
         /* loopback: single rank exchanges with itself */
         if (send_counts[0] != recv_counts[0]) return T9_EINVAL;
         HIP_TRY(hipMemcpyAsync(
@@ -99,16 +110,6 @@ int t9_alltoall(t9_context* ctx, const void* d_send, const u64* send_counts,
     if (!ctx->comm) return T9_EINVAL;
     ncclComm_t comm = (ncclComm_t)ctx->comm;
     hipStream_t s = (hipStream_t)stream;
-    /* The rank's own share bypasses RCCL: a plain device copy on the same
-     * stream is both faster (no protocol round trip for ~1/p of the data)
-     * and avoids the self-send path entirely — a 10.7 GB RCCL
-     * self-exchange was measured as a hang in round 1 (pipeline.py
-     * history, commit 551d307); T9_A2A_SELF=nccl restores the RCCL
-     * self-send for investigation. */
-    static const bool self_nccl = [] {
-        const char* e = getenv("T9_A2A_SELF");
-        return e && strcmp(e, "nccl") == 0;
-    }();
     const int me = ctx->rank;
     if (!self_nccl) {
         if (send_counts[me] != recv_counts[me]) return T9_EINVAL;

@@ -47,6 +47,25 @@ __device__ inline void t9_scan256_onewave(u32* s_vals, u32 tid) {
     }
 }
 
+/* ballot-combined LDS histogram add: lanes holding the same digit are
+ * grouped by BITS ballots; the group leader issues ONE atomicAdd of the
+ * group size. Same cost as a plain atomicAdd on uniform digits
+ * (memory-bound either way) but immune to the degenerate all-equal-digit
+ * case, where 64-way same-address LDS atomics serialize (measured
+ * 16x on the segmented hist during the all-identical-records tie
+ * bench). */
+template <int BITS>
+__device__ inline void t9_hist_ballot_add(u32* __restrict__ s_cnt, u32 d,
+                                          bool valid, u32 lane) {
+    u64 m = __ballot(valid);
+    for (int b = 0; b < BITS; ++b) {
+        const u64 bb = __ballot((d >> b) & 1u);
+        m &= ((d >> b) & 1u) ? bb : ~bb;
+    }
+    if (valid && (u32)__popcll(m & ((1ull << lane) - 1ull)) == 0)
+        atomicAdd(&s_cnt[d], (u32)__popcll(m));
+}
+
 /* 512-thread (8-wave) wave-autonomous scatter, always global re-read:
  * doubles waves/SIMD to 4 at the same LDS footprint — the PMC-measured
  * limiter of the 256-thread version was 81% SQ_WAIT_ANY at 2 waves/SIMD.

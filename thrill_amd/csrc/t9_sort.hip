@@ -23,6 +23,7 @@
  */
 
 #include "t9_common.h"
+#include "t9_rank_scatter.h"
 
 #include <algorithm>
 #include <cstdlib>
@@ -42,10 +43,15 @@ __global__ __launch_bounds__(256) void k_hist(
     const u32 tn = (u32)((n - base < (u64)TILE) ? (n - base) : (u64)TILE);
     s_cnt[tid] = 0;
     __syncthreads();
-    for (u32 i = tid; i < tn; i += 256) {
-        u32 d = EXT_DIGIT ? ext_digit[base + i]
+    const u32 lane = tid & 63;
+    for (u32 i0 = 0; i0 < (u32)TILE; i0 += 256) {
+        const u32 i = i0 + tid;
+        const bool valid = i < tn;
+        u32 d = 0;
+        if (valid)
+            d = EXT_DIGIT ? ext_digit[base + i]
                           : ((u32)(in_keys[base + i] >> shift) & 255u);
-        atomicAdd(&s_cnt[d], 1u);
+        t9_hist_ballot_add<8>(s_cnt, d, valid, lane);
     }
     __syncthreads();
     hist[(u64)blockIdx.x * T9_RADIX + tid] = s_cnt[tid];

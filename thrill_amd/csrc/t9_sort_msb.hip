@@ -82,13 +82,18 @@ __global__ __launch_bounds__(256) void k_hist_msb(
     s_cnt[tid] = 0;
     __syncthreads();
     const u8* rec8 = (const u8*)keys;
-    for (u32 i = tid; i < tn; i += 256) {
-        u32 d;
-        if (RW)
-            d = rec8[(base + i) * (u64)RW * 4 + (7 - shift / 8)];
-        else
-            d = (u32)(keys[base + i] >> shift) & 255u;
-        atomicAdd(&s_cnt[d], 1u);
+    const u32 lane = tid & 63;
+    for (u32 i0 = 0; i0 < (u32)T9_MSB_TILE; i0 += 256) {
+        const u32 i = i0 + tid;
+        const bool valid = i < tn;
+        u32 d = 0;
+        if (valid) {
+            if (RW)
+                d = rec8[(base + i) * (u64)RW * 4 + (7 - shift / 8)];
+            else
+                d = (u32)(keys[base + i] >> shift) & 255u;
+        }
+        t9_hist_ballot_add<8>(s_cnt, d, valid, lane);
     }
     __syncthreads();
     hist[(u64)blockIdx.x * T9_RADIX + tid] = s_cnt[tid];
@@ -157,8 +162,15 @@ __global__ __launch_bounds__(256) void k_hist_seg(
                                   ? bucket_n[b] - off
                                   : T9_MSB_TILE)
                            : 0;
-        for (u32 i = tid; i < tn; i += 256)
-            atomicAdd(&s_cnt[(u32)(keys[tbase + i] >> shift) & 255u], 1u);
+        const u32 lane = tid & 63;
+        for (u32 i0 = 0; i0 < (u32)T9_MSB_TILE; i0 += 256) {
+            const u32 i = i0 + tid;
+            const bool valid = i < tn;
+            const u32 d = valid
+                              ? (u32)(keys[tbase + i] >> shift) & 255u
+                              : 0;
+            t9_hist_ballot_add<8>(s_cnt, d, valid, lane);
+        }
     }
     __syncthreads();
     hist[(u64)blockIdx.x * T9_RADIX + tid] = s_cnt[tid];
@@ -342,9 +354,14 @@ __global__ __launch_bounds__(256) void k_hist_seg3(
         const u32 base = sub_start[s] + toff;
         const u32 rem = sub_n[s] - toff;
         const u32 tn = (rem < (u32)T9_MSB_TILE) ? rem : (u32)T9_MSB_TILE;
-        for (u32 i = tid; i < tn; i += 256)
-            atomicAdd(&s_cnt[(u32)(keys[(u64)base + i] >> shift) & 255u],
-                      1u);
+        const u32 lane = tid & 63;
+        for (u32 i0 = 0; i0 < (u32)T9_MSB_TILE; i0 += 256) {
+            const u32 i = i0 + tid;
+            const bool valid = i < tn;
+            const u32 d =
+                valid ? (u32)(keys[(u64)base + i] >> shift) & 255u : 0;
+            t9_hist_ballot_add<8>(s_cnt, d, valid, lane);
+        }
     }
     __syncthreads();
     hist[(u64)blockIdx.x * T9_RADIX + tid] = s_cnt[tid];

@@ -56,8 +56,14 @@ __global__ __launch_bounds__(256) void k_hist_seg9(
                                   ? bucket_n[b] - off
                                   : T9_MSB_TILE)
                            : 0;
-        for (u32 i = tid; i < tn; i += 256)
-            atomicAdd(&s_cnt[(u32)(keys[tbase + i] >> 47) & 511u], 1u);
+        const u32 lane = tid & 63;
+        for (u32 i0 = 0; i0 < (u32)T9_MSB_TILE; i0 += 256) {
+            const u32 i = i0 + tid;
+            const bool valid = i < tn;
+            const u32 d =
+                valid ? (u32)(keys[tbase + i] >> 47) & 511u : 0;
+            t9_hist_ballot_add<9>(s_cnt, d, valid, lane);
+        }
     }
     __syncthreads();
     hist[(u64)blockIdx.x * NDIG9 + tid] = s_cnt[tid];
