@@ -119,15 +119,30 @@ int t9_alltoall(t9_context* ctx, const void* d_send, const u64* send_counts,
                 (const char*)d_send + send_displs[me] * elem_size,
                 send_counts[me] * elem_size, hipMemcpyDeviceToDevice, s));
     }
+    /* Chunk every point-to-point transfer at 1 GiB: RCCL's p2p path
+     * SILENTLY TRUNCATES single messages at count mod 2^32 bytes
+     * (measured on MI355X, scripts/probe_self_nccl.py: a 2^32-byte
+     * self-send moves 0 bytes and "succeeds" in 4 ms; 10.7 GB moves the
+     * low-32-bit remainder — the same class of failure round 1 observed
+     * as a hang through torch's path). Chunks to the same peer inside
+     * one group are ordered, so sender and receiver split
+     * deterministically and identically. */
+    const u64 CHUNK = 1ull << 30;
     ncclResult_t rc = ncclGroupStart();
     for (int r = 0; r < ctx->world && rc == ncclSuccess; ++r) {
         if (r == me && !self_nccl) continue;
-        if (send_counts[r])
-            rc = ncclSend((const char*)d_send + send_displs[r] * elem_size,
-                          send_counts[r] * elem_size, ncclUint8, r, comm, s);
-        if (rc == ncclSuccess && recv_counts[r])
-            rc = ncclRecv((char*)d_recv + recv_displs[r] * elem_size,
-                          recv_counts[r] * elem_size, ncclUint8, r, comm, s);
+        const u64 sb = send_counts[r] * elem_size;
+        const char* sp = (const char*)d_send + send_displs[r] * elem_size;
+        for (u64 off = 0; off < sb && rc == ncclSuccess; off += CHUNK)
+            rc = ncclSend(sp + off,
+                          (sb - off < CHUNK) ? (sb - off) : CHUNK,
+                          ncclUint8, r, comm, s);
+        const u64 rb = recv_counts[r] * elem_size;
+        char* rp = (char*)d_recv + recv_displs[r] * elem_size;
+        for (u64 off = 0; off < rb && rc == ncclSuccess; off += CHUNK)
+            rc = ncclRecv(rp + off,
+                          (rb - off < CHUNK) ? (rb - off) : CHUNK,
+                          ncclUint8, r, comm, s);
     }
     ncclResult_t rce = ncclGroupEnd();
     if (rc != ncclSuccess || rce != ncclSuccess) {

@@ -262,6 +262,35 @@ __global__ __launch_bounds__(256) void k_tie_prescan(
         if (s_flags[c]) atomicExch(&flags[c], 1u);
 }
 
+/* sequential prescan for the m == n case (every record tied — the
+ * adversarial all-equal benches): word g of the packed record array is
+ * compared directly against the reference record's word g % rw, fully
+ * coalesced (no index indirection; the set of records scanned is the
+ * same, and the flags are order-invariant). */
+__global__ __launch_bounds__(256) void k_tie_prescan_seq(
+    const u8* __restrict__ recs, const u32* __restrict__ tidx, u64 m,
+    u32 rec_size, u32 nc, u32* __restrict__ flags) {
+    __shared__ u32 s_flags[64];
+    __shared__ u32 s_ref[160];
+    const u32 tid = threadIdx.x;
+    const u32 rw = rec_size / 4;
+    for (u32 c = tid; c <= nc && c < 64; c += 256) s_flags[c] = 0;
+    const u32* rin = (const u32*)recs;
+    if (tid < rw && rw <= 160)
+        s_ref[tid] = rin[(u64)tidx[0] * rw + tid];
+    __syncthreads();
+    const u64 total = m * rw;
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 g = (u64)blockIdx.x * 256 + tid; g < total; g += stride) {
+        const u32 w = (u32)(g % rw);
+        if (rin[g] ^ s_ref[w])
+            s_flags[w < 2 ? 0 : 1 + (w - 2) / 2] = 1;
+    }
+    __syncthreads();
+    for (u32 c = tid; c <= nc && c < 64; c += 256)
+        if (s_flags[c]) atomicExch(&flags[c], 1u);
+}
+
 /* d_idx[perm[j]] = tidx[j] — write the re-ordered record indices back
  * into the (ascending) tied positions */
 __global__ __launch_bounds__(256) void k_tie_scatter(
@@ -707,11 +736,17 @@ static int sort_records_impl(t9_context* ctx, const u8* d_in, u8* d_out,
             u32 hflags[64];
             if (nc <= 63) {
                 HIP_TRY(hipMemsetAsync(t_flag, 0, 64 * 4, s));
-                T9_PERF_WRAP(s, "tie_prescan",
-                             hipLaunchKernelGGL(k_tie_prescan,
-                                                dim3(mgrid), dim3(256),
-                                                0, s, d_in, t_idx, m,
-                                                rec_size, nc, t_flag));
+                T9_PERF_WRAP(
+                    s, "tie_prescan",
+                    if (m == n && rec_size / 4 <= 160)
+                        /* every record tied: fully coalesced word scan */
+                        hipLaunchKernelGGL(k_tie_prescan_seq, dim3(8192),
+                                           dim3(256), 0, s, d_in, t_idx,
+                                           m, rec_size, nc, t_flag);
+                    else
+                        hipLaunchKernelGGL(k_tie_prescan, dim3(mgrid),
+                                           dim3(256), 0, s, d_in, t_idx,
+                                           m, rec_size, nc, t_flag));
                 T9_LAUNCH_CHECK();
                 HIP_TRY(hipMemcpyAsync(hflags, t_flag, (nc + 1) * 4,
                                        hipMemcpyDeviceToHost, s));

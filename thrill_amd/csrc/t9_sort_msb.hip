@@ -559,6 +559,42 @@ __global__ __launch_bounds__(256) void k_span_pack(
     span_len[slot] = cur_len;
 }
 
+/* differ[0] = 1 if any keys[i] != keys[0] over the range — guards the
+ * oversize-sub-bucket LSD fallback: an all-equal oversize sub-bucket
+ * (the degenerate-key adversarial inputs) is already in stable order
+ * and must NOT pay 8 LSD passes (measured ~10 ms on the all-identical
+ * 10 GiB tie bench before this check). */
+__global__ __launch_bounds__(256) void k_range_differ(
+    const u64* __restrict__ keys, u64 n, u32* __restrict__ differ) {
+    __shared__ u32 s;
+    if (threadIdx.x == 0) s = 0;
+    __syncthreads();
+    const u64 v0 = keys[0];
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 i = (u64)blockIdx.x * 256 + threadIdx.x; i < n; i += stride)
+        if (keys[i] != v0) { s = 1; break; }
+    __syncthreads();
+    if (threadIdx.x == 0 && s) atomicExch(differ, 1u);
+}
+
+/* host: does the key range hold >1 distinct value? (scratch = any device
+ * u32) */
+static bool range_differs(const u64* d_keys, u64 cnt, u32* d_scratch,
+                          hipStream_t s) {
+    if (cnt < 2) return false;
+    if (hipMemsetAsync(d_scratch, 0, 4, s) != hipSuccess) return true;
+    u64 want = t9_ceil_div(cnt, 256);
+    hipLaunchKernelGGL(k_range_differ,
+                       dim3((u32)((want < 4096) ? want : 4096)), dim3(256),
+                       0, s, d_keys, cnt, d_scratch);
+    u32 f = 1;
+    if (hipMemcpyAsync(&f, d_scratch, 4, hipMemcpyDeviceToHost, s) !=
+        hipSuccess)
+        return true;
+    if (hipStreamSynchronize(s) != hipSuccess) return true;
+    return f != 0;
+}
+
 /* sub-bucket stats: info[0] = max size (atomicMax), info[1] = count of
  * sub-buckets above hardmax, list = their indices */
 __global__ __launch_bounds__(256) void k_subinfo(
@@ -1088,6 +1124,8 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
                                   hipMemcpyDeviceToHost));
             }
             for (u32 i = 0; i < novr9; ++i) {
+                if (!range_differs(d_keys + st9[i], cn9[i], w.ovr, s))
+                    continue;   /* all-equal: stable order already */
                 int rc = HAS_VAL
                              ? t9i_sort_pairs_lsd(ctx, d_keys + st9[i],
                                                   d_vals + st9[i], cn9[i],
@@ -1221,6 +1259,8 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
                               hipMemcpyDeviceToHost));
         }
         for (u32 i = 0; i < novr3; ++i) {
+            if (!range_differs(d_keys + st3[i], cn3[i], w.ovr, s))
+                continue;   /* all-equal: stable order already */
             int rc = HAS_VAL
                          ? t9i_sort_pairs_lsd(ctx, d_keys + st3[i],
                                               d_vals + st3[i], cn3[i],
@@ -1328,6 +1368,8 @@ static int sort_msb_impl(t9_context* ctx, const u64* pass1_src,
                           hipMemcpyDeviceToHost));
     }
     for (u32 i = 0; i < novr; ++i) {
+        if (!range_differs(d_keys + starts[i], counts[i], w.ovr, s))
+            continue;   /* all-equal: stable order already */
         int rc = HAS_VAL
                      ? t9i_sort_pairs_lsd(ctx, d_keys + starts[i],
                                           d_vals + starts[i], counts[i],
