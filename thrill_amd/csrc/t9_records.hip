@@ -331,36 +331,41 @@ __global__ __launch_bounds__(256) void k_extract_hist(
     u32* __restrict__ idx, u32* __restrict__ hist) {
     constexpr int SUBREC = 256;                /* records per LDS stage */
     constexpr int SUBW = SUBREC * RW;          /* words per stage */
-    __shared__ u32 s_buf[SUBW];
+    __shared__ u32 s_buf[2][SUBW];             /* double-buffered: one
+                                                  barrier per stage */
     __shared__ u32 s_cnt[256];
     const u32 tid = threadIdx.x;
     const u64 tile0 = (u64)blockIdx.x * 8192;
     const u32 tn = (u32)((n - tile0 < 8192) ? (n - tile0) : 8192);
     s_cnt[tid] = 0;
-    __syncthreads();
     const u32* rin = (const u32*)recs;
-    for (u32 s0 = 0; s0 < tn; s0 += SUBREC) {
+
+    auto load_stage = [&](u32 s0, int buf) {
         const u32 sn = (tn - s0 < SUBREC) ? tn - s0 : SUBREC;
         const u64 w0 = (tile0 + s0) * RW;
         const u32 wn = sn * RW;
-        for (u32 w = tid; w < wn; w += 256) s_buf[w] = rin[w0 + w];
-        __syncthreads();
-        {
-            const bool valid = tid < sn;
-            u64 k = 0;
-            if (valid) {
-                if (LE)
-                    k = ((u64)s_buf[tid * RW + 1] << 32) | s_buf[tid * RW];
-                else
-                    k = ((u64)__builtin_bswap32(s_buf[tid * RW]) << 32) |
-                        __builtin_bswap32(s_buf[tid * RW + 1]);
-                const u64 gi = tile0 + s0 + tid;
-                keys[gi] = k;
-                idx[gi] = (u32)gi;
-            }
-            t9_hist_ballot_add<8>(s_cnt, (u32)(k >> 56), valid,
-                                  tid & 63);
+        for (u32 w = tid; w < wn; w += 256) s_buf[buf][w] = rin[w0 + w];
+    };
+    load_stage(0, 0);
+    __syncthreads();
+    int cur = 0;
+    for (u32 s0 = 0; s0 < tn; s0 += SUBREC, cur ^= 1) {
+        if (s0 + SUBREC < tn) load_stage(s0 + SUBREC, cur ^ 1);
+        const u32 sn = (tn - s0 < SUBREC) ? tn - s0 : SUBREC;
+        const bool valid = tid < sn;
+        u64 k = 0;
+        if (valid) {
+            if (LE)
+                k = ((u64)s_buf[cur][tid * RW + 1] << 32) |
+                    s_buf[cur][tid * RW];
+            else
+                k = ((u64)__builtin_bswap32(s_buf[cur][tid * RW]) << 32) |
+                    __builtin_bswap32(s_buf[cur][tid * RW + 1]);
+            const u64 gi = tile0 + s0 + tid;
+            keys[gi] = k;
+            idx[gi] = (u32)gi;
         }
+        t9_hist_ballot_add<8>(s_cnt, (u32)(k >> 56), valid, tid & 63);
         __syncthreads();
     }
     hist[(u64)blockIdx.x * 256 + tid] = s_cnt[tid];
