@@ -189,6 +189,30 @@ def test_sort_records_fused_extract_hist(nat, oracle):
         os.environ.pop("T9_EXTRACT_HIST", None)
 
 
+def test_sort_records_keyle_fused_extract(nat):
+    # the fused LE extract+hist path (config 5's default at scale),
+    # engaged at test size via T9_SORT_ALGO=msb
+    import os
+    os.environ["T9_SORT_ALGO"] = "msb"
+    try:
+        n = 200_000
+        rng = np.random.default_rng(91)
+        recs = rng.integers(0, 256, (n, 128)).astype(np.uint8)
+        recs[: n // 20, :8] = recs[0, :8]   # duplicate numeric keys
+        din = G.dev(recs.reshape(-1))
+        dout = G.empty(n * 128, np.uint8)
+        w = G.ws(nat.ws("sort_records", n, 128))
+        nat.sort_records_keyle(G.ptr(din), G.ptr(dout), n, 128, G.ptr(w),
+                               G.stream())
+        got = G.host(dout, np.uint8).reshape(n, 128)
+        keys = recs[:, :8].copy().view("<u8").reshape(-1)
+        order = np.lexsort(tuple(recs[:, c] for c in range(127, 7, -1))
+                           + (keys,))
+        assert np.array_equal(got, recs[order])
+    finally:
+        del os.environ["T9_SORT_ALGO"]
+
+
 def test_gather_scatter_variant_parity(nat, oracle):
     # T9_GATHER_VARIANT=5 (sequential-read random-write scatter probe)
     import os
@@ -229,6 +253,36 @@ def test_sort_records_many_tie_runs(nat, oracle):
                      G.stream())
     got = G.host(dout, np.uint8).reshape(n, 100)
     assert np.array_equal(got, oracle.sort_records(recs))
+
+
+def test_sort_records_msd_fallback_low_entropy_tails(nat, oracle):
+    # tie pattern that PERSISTS past 2 MSD levels: one shared prefix,
+    # tail chunks drawn from {0,1} — every chunk differs globally but
+    # most pairs still tie after each level, so the bounded-LSD fallback
+    # runs. Also cross-checks T9_TIE_MSD=0 (pure LSD) bit-for-bit.
+    import os
+    n = 50_000
+    rng = np.random.default_rng(17)
+    recs = np.zeros((n, 100), np.uint8)
+    recs[:, :10] = 0x42
+    for b in (15, 23, 31, 47, 80):
+        recs[:, b] = rng.integers(0, 2, n).astype(np.uint8)
+    din = G.dev(recs.reshape(-1))
+    dout = G.empty(n * 100, np.uint8)
+    w = G.ws(nat.ws("sort_records", n, 100))
+    nat.sort_records(G.ptr(din), G.ptr(dout), n, 100, 10, G.ptr(w),
+                     G.stream())
+    got = G.host(dout, np.uint8).reshape(n, 100)
+    expect = oracle.sort_records(recs)
+    assert np.array_equal(got, expect)
+    os.environ["T9_TIE_MSD"] = "0"
+    try:
+        nat.sort_records(G.ptr(din), G.ptr(dout), n, 100, 10, G.ptr(w),
+                         G.stream())
+        got2 = G.host(dout, np.uint8).reshape(n, 100)
+        assert np.array_equal(got2, expect)
+    finally:
+        del os.environ["T9_TIE_MSD"]
 
 
 def test_sort_records_all_identical(nat, oracle):

@@ -219,32 +219,26 @@ def test_three_level_forced_records(nat, oracle):
         del os.environ["T9_MSB_LEVELS"]
 
 
-@pytest.mark.parametrize("algo", [None, "msb"])
-def test_sort_records_keyle_config5_shape(nat, algo):
+def test_sort_records_keyle_config5_shape(nat):
     """config 5 semantics: 128 B records, native u64 key, numeric order,
-    payload-byte tiebreak. algo='msb' engages the fused LE extract+hist
-    path at test scale (the default path at config-5 sizes)."""
-    if algo:
-        os.environ["T9_SORT_ALGO"] = algo
-    try:
-        n = 150_000
-        rng = np.random.default_rng(41)
-        recs = rng.integers(0, 256, (n, 128)).astype(np.uint8)
-        # plant duplicate numeric keys to exercise the payload tiebreak
-        recs[: n // 10, :8] = recs[0, :8]
-        din = G.dev(recs.reshape(-1))
-        dout = G.empty(n * 128, np.uint8)
-        w = G.ws(nat.ws("sort_records", n, 128))
-        nat.sort_records_keyle(G.ptr(din), G.ptr(dout), n, 128, G.ptr(w),
-                               G.stream())
-        got = G.host(dout, np.uint8).reshape(n, 128)
-        keys = recs[:, :8].copy().view("<u8").reshape(-1)
-        order = np.lexsort(tuple(recs[:, c] for c in range(127, 7, -1))
-                           + (keys,))
-        assert np.array_equal(got, recs[order])
-    finally:
-        if algo:
-            del os.environ["T9_SORT_ALGO"]
+    payload-byte tiebreak. The module's msb_env fixture forces the MSB
+    dispatch, so this also exercises the fused LE extract+hist path (the
+    default at config-5 sizes)."""
+    n = 150_000
+    rng = np.random.default_rng(41)
+    recs = rng.integers(0, 256, (n, 128)).astype(np.uint8)
+    # plant duplicate numeric keys to exercise the payload tiebreak
+    recs[: n // 10, :8] = recs[0, :8]
+    din = G.dev(recs.reshape(-1))
+    dout = G.empty(n * 128, np.uint8)
+    w = G.ws(nat.ws("sort_records", n, 128))
+    nat.sort_records_keyle(G.ptr(din), G.ptr(dout), n, 128, G.ptr(w),
+                           G.stream())
+    got = G.host(dout, np.uint8).reshape(n, 128)
+    keys = recs[:, :8].copy().view("<u8").reshape(-1)
+    order = np.lexsort(tuple(recs[:, c] for c in range(127, 7, -1))
+                       + (keys,))
+    assert np.array_equal(got, recs[order])
 
 
 @pytest.mark.parametrize("case", ["uniform", "few_values", "skew_top"])

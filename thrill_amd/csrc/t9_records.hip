@@ -291,6 +291,122 @@ __global__ __launch_bounds__(256) void k_tie_prescan_seq(
         if (s_flags[c]) atomicExch(&flags[c], 1u);
 }
 
+/* ---- MSD tie strategy (group-id sorts) --------------------------- *
+ * LSD over the tail is oblivious: it sorts EVERY differing chunk even
+ * when the first one already separates everything (random tails — 13
+ * full sorts of m = n on the shared-prefix adversarial input). The MSD
+ * strategy processes chunks most-significant first and STOPS as soon as
+ * no two elements still tie: per level, stably sort by the chunk, then
+ * stably sort by the current group id (= run-start position), which
+ * yields exactly the segmented order (group asc, chunk asc, stable);
+ * then recompute groups as runs of equal (group, chunk) and count the
+ * remaining tied elements. Random tails exit after ONE level (two pair
+ * sorts); inputs whose ties persist past 2 levels fall back to the
+ * bounded LSD loop. Group ids sort as (u64)grp << 32 so the MSB radix
+ * pipeline sees their entropy in its top bytes. */
+
+/* v[j] = j if a group boundary starts at j (j==0, group changed, or
+ * chunk value changed), else 0; d_ties += #non-boundary elements.
+ * grp == NULL: boundaries from ck alone (the init pass on the prefix). */
+__global__ __launch_bounds__(256) void k_tie_bound_vals(
+    const u32* __restrict__ grp, const u64* __restrict__ ck, u64 m,
+    u32* __restrict__ v, u32* __restrict__ ties) {
+    __shared__ u32 s_t;
+    if (threadIdx.x == 0) s_t = 0;
+    __syncthreads();
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 j = (u64)blockIdx.x * 256 + threadIdx.x; j < m; j += stride) {
+        bool b = (j == 0) || (ck[j] != ck[j - 1]) ||
+                 (grp && grp[j] != grp[j - 1]);
+        v[j] = b ? (u32)j : 0;
+        if (!b) atomicAdd(&s_t, 1u);
+    }
+    __syncthreads();
+    if (threadIdx.x == 0 && s_t) atomicAdd(ties, s_t);
+}
+
+#define T9_MAXSCAN_TILE 2048
+
+/* per-tile inclusive max-scan of v (in place); partials[b] = tile max */
+__global__ __launch_bounds__(256) void k_maxscan_local(
+    u32* __restrict__ v, u64 m, u32* __restrict__ partials) {
+    __shared__ u32 s[256];
+    const u32 tid = threadIdx.x;
+    const u64 base = (u64)blockIdx.x * T9_MAXSCAN_TILE;
+    u32 carry = 0;
+    for (u32 c = 0; c < T9_MAXSCAN_TILE / 256; ++c) {
+        const u64 j = base + c * 256 + tid;
+        u32 x = (j < m) ? v[j] : 0;
+        s[tid] = x;
+        __syncthreads();
+        for (int off = 1; off < 256; off <<= 1) {
+            u32 y = (tid >= (u32)off) ? s[tid - off] : 0;
+            __syncthreads();
+            if (y > s[tid]) s[tid] = y;
+            __syncthreads();
+        }
+        const u32 val = (s[tid] > carry) ? s[tid] : carry;
+        if (j < m) v[j] = val;
+        const u32 cmax = (s[255] > carry) ? s[255] : carry;
+        __syncthreads();
+        carry = cmax;
+    }
+    if (tid == 0) partials[blockIdx.x] = carry;
+}
+
+/* single block: EXCLUSIVE max-scan over B partials, in place */
+__global__ __launch_bounds__(256) void k_maxscan_part(
+    u32* __restrict__ p, u64 B) {
+    __shared__ u32 s[256];
+    const u32 tid = threadIdx.x;
+    u32 carry = 0;
+    for (u64 c0 = 0; c0 < B; c0 += 256) {
+        const u64 b = c0 + tid;
+        u32 x = (b < B) ? p[b] : 0;
+        s[tid] = x;
+        __syncthreads();
+        for (int off = 1; off < 256; off <<= 1) {
+            u32 y = (tid >= (u32)off) ? s[tid - off] : 0;
+            __syncthreads();
+            if (y > s[tid]) s[tid] = y;
+            __syncthreads();
+        }
+        /* exclusive: shift by one (carry for lane 0) */
+        const u32 excl = (tid == 0) ? carry
+                                    : ((s[tid - 1] > carry) ? s[tid - 1]
+                                                            : carry);
+        if (b < B) p[b] = excl;
+        const u32 cmax = (s[255] > carry) ? s[255] : carry;
+        __syncthreads();
+        carry = cmax;
+    }
+}
+
+/* v[j] = max(v[j], partials[tile]) — v becomes the new group-id array */
+__global__ __launch_bounds__(256) void k_maxscan_fix(
+    u32* __restrict__ v, u64 m, const u32* __restrict__ partials) {
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 j = (u64)blockIdx.x * 256 + threadIdx.x; j < m; j += stride) {
+        const u32 pmax = partials[j / T9_MAXSCAN_TILE];
+        if (pmax > v[j]) v[j] = pmax;
+    }
+}
+
+/* grpkey[x] = (u64)grp[x] << 32 (entropy into the MSB pipeline's bytes) */
+__global__ __launch_bounds__(256) void k_grp_keys(
+    const u32* __restrict__ grp, u64 m, u64* __restrict__ out) {
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 j = (u64)blockIdx.x * 256 + threadIdx.x; j < m; j += stride)
+        out[j] = (u64)grp[j] << 32;
+}
+
+__global__ __launch_bounds__(256) void k_iota32(u32* __restrict__ v,
+                                                u64 m) {
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 j = (u64)blockIdx.x * 256 + threadIdx.x; j < m; j += stride)
+        v[j] = (u32)j;
+}
+
 /* d_idx[perm[j]] = tidx[j] — write the re-ordered record indices back
  * into the (ascending) tied positions */
 __global__ __launch_bounds__(256) void k_tie_scatter(
@@ -692,8 +808,10 @@ static int sort_records_impl(t9_context* ctx, const u8* d_in, u8* d_out,
          * lives in d_out, which is dead until the final gather (~21 B
          * per element vs rec_size B available); records narrower than
          * 24 B fall back to a temporary allocation. */
-        const u64 need = 3 * t9_align256(n * 4) + t9_align256(n * 8) +
+        const u64 need = 6 * t9_align256(n * 4) +
+                         2 * t9_align256(n * 8) +
                          t9_align256(3 * 8) + 256 +
+                         t9_align256((n / T9_MAXSCAN_TILE + 2) * 4) + 256 +
                          t9_partition_idx_workspace(n);
         void* tmp_alloc = nullptr;
         char* q;
@@ -710,6 +828,15 @@ static int sort_records_impl(t9_context* ctx, const u8* d_in, u8* d_out,
         u64* t_chunk = (u64*)q;  q += t9_align256(n * 8);
         u64* t_offs = (u64*)q;   q += t9_align256(3 * 8);
         u32* t_flag = (u32*)q;   q += 256;
+        /* MSD strategy state (t_bucket doubles as the iota/permutation
+         * buffer once the partition is done) */
+        u32* t_grp = (u32*)q;    q += t9_align256(n * 4);
+        u32* t_grp2 = (u32*)q;   q += t9_align256(n * 4);
+        u32* t_tmp4 = (u32*)q;   q += t9_align256(n * 4);
+        u64* t_chunk2 = (u64*)q; q += t9_align256(n * 8);
+        u32* t_part = (u32*)q;
+        q += t9_align256((n / T9_MAXSCAN_TILE + 2) * 4);
+        u32* t_ties = (u32*)q;   q += 256;
         void* t_pws = q;
 
         const u32 ngrid =
@@ -760,20 +887,125 @@ static int sort_records_impl(t9_context* ctx, const u8* d_in, u8* d_out,
             else {
                 for (u32 c = 0; c < 64; ++c) hflags[c] = 1;
             }
-            for (u32 c = nc + 1; !rc && c-- > 0;) {
-                if (nc <= 63 && !hflags[c]) continue;
-                const u32 off = c ? 8 + (c - 1) * 8 : 0;
-                if (off == 0 && le)
-                    hipLaunchKernelGGL((k_tie_chunk<true>), dim3(mgrid),
-                                       dim3(256), 0, s, d_in, t_idx, m,
-                                       rec_size, off, t_chunk);
-                else
+            /* differing tail chunks, most-significant first */
+            std::vector<u32> tailC;
+            if (nc <= 63) {
+                for (u32 c = 1; c <= nc; ++c)
+                    if (hflags[c]) tailC.push_back(c - 1);
+            }
+            else {
+                for (u32 c = 0; c < nc; ++c) tailC.push_back(c);
+            }
+            const char* me_ = getenv("T9_TIE_MSD");
+            const bool use_msd = !(me_ && me_[0] == '0') &&
+                                 tailC.size() >= 2;
+            bool msd_done = false;
+            if (use_msd && !rc) {
+                /* init groups = runs of equal prefix (gather the tied
+                 * prefixes, boundary where they change) */
+                rc = t9_gather_records(ctx, (const u8*)d_keys, t_perm, m,
+                                       8, (u8*)t_chunk, stream);
+                const u32 B2k = (u32)t9_ceil_div(m, T9_MAXSCAN_TILE);
+                auto regroup = [&](const u32* old_grp, const u64* ck,
+                                   u32* out_grp, u32* h_ties) -> int {
+                    if (hipMemsetAsync(t_ties, 0, 4, s) != hipSuccess)
+                        return T9_EIO;
+                    hipLaunchKernelGGL(k_tie_bound_vals, dim3(mgrid),
+                                       dim3(256), 0, s, old_grp, ck, m,
+                                       out_grp, t_ties);
+                    hipLaunchKernelGGL(k_maxscan_local, dim3(B2k),
+                                       dim3(256), 0, s, out_grp, m,
+                                       t_part);
+                    hipLaunchKernelGGL(k_maxscan_part, dim3(1), dim3(256),
+                                       0, s, t_part, B2k);
+                    hipLaunchKernelGGL(k_maxscan_fix, dim3(mgrid),
+                                       dim3(256), 0, s, out_grp, m,
+                                       t_part);
+                    T9_LAUNCH_CHECK();
+                    if (!h_ties) return T9_OK;
+                    if (hipMemcpyAsync(h_ties, t_ties, 4,
+                                       hipMemcpyDeviceToHost, s) !=
+                            hipSuccess ||
+                        hipStreamSynchronize(s) != hipSuccess)
+                        return T9_EIO;
+                    return T9_OK;
+                };
+                if (!rc) rc = regroup(nullptr, t_chunk, t_grp, nullptr);
+                u32* jbuf = t_bucket;   /* partition is done; reuse */
+                size_t ci = 0;
+                for (; ci < tailC.size() && ci < 2 && !rc; ++ci) {
+                    const u32 off = 8 + tailC[ci] * 8;
                     hipLaunchKernelGGL((k_tie_chunk<false>), dim3(mgrid),
                                        dim3(256), 0, s, d_in, t_idx, m,
                                        rec_size, off, t_chunk);
-                T9_LAUNCH_CHECK();
-                rc = t9_sort_pairs_u64_u32(ctx, t_chunk, t_idx, m,
-                                           pair_ws, stream);
+                    hipLaunchKernelGGL(k_iota32, dim3(mgrid), dim3(256),
+                                       0, s, jbuf, m);
+                    T9_LAUNCH_CHECK();
+                    rc = t9_sort_pairs_u64_u32(ctx, t_chunk, jbuf, m,
+                                               pair_ws, stream);
+                    if (rc) break;
+                    rc = t9_gather_records(ctx, (const u8*)t_idx, jbuf, m,
+                                           4, (u8*)t_tmp4, stream);
+                    if (!rc)
+                        rc = t9_gather_records(ctx, (const u8*)t_grp,
+                                               jbuf, m, 4, (u8*)t_grp2,
+                                               stream);
+                    if (rc) break;
+                    hipLaunchKernelGGL(k_grp_keys, dim3(mgrid), dim3(256),
+                                       0, s, t_grp2, m, t_chunk2);
+                    hipLaunchKernelGGL(k_iota32, dim3(mgrid), dim3(256),
+                                       0, s, jbuf, m);
+                    T9_LAUNCH_CHECK();
+                    rc = t9_sort_pairs_u64_u32(ctx, t_chunk2, jbuf, m,
+                                               pair_ws, stream);
+                    if (rc) break;
+                    rc = t9_gather_records(ctx, (const u8*)t_tmp4, jbuf,
+                                           m, 4, (u8*)t_idx, stream);
+                    if (!rc)
+                        rc = t9_gather_records(ctx, (const u8*)t_grp2,
+                                               jbuf, m, 4, (u8*)t_grp,
+                                               stream);
+                    if (!rc)
+                        rc = t9_gather_records(ctx, (const u8*)t_chunk,
+                                               jbuf, m, 8, (u8*)t_chunk2,
+                                               stream);
+                    if (rc) break;
+                    u32 ties = 0;
+                    rc = regroup(t_grp, t_chunk2, t_grp2, &ties);
+                    if (rc) break;
+                    std::swap(t_grp, t_grp2);
+                    if (ties == 0) {
+                        msd_done = true;
+                        ++ci;
+                        break;
+                    }
+                }
+                if (!rc && !msd_done && ci >= tailC.size())
+                    msd_done = true;   /* all chunks processed: remaining
+                                          ties are true duplicates,
+                                          stable order is correct */
+            }
+            if (!msd_done) {
+                /* LSD over the differing chunks (last -> first), then
+                 * the prefix regroup pass — also the MSD fallback for
+                 * tie patterns that persist past 2 levels (bounded) */
+                for (u32 c = nc + 1; !rc && c-- > 0;) {
+                    if (nc <= 63 && !hflags[c]) continue;
+                    const u32 off = c ? 8 + (c - 1) * 8 : 0;
+                    if (off == 0 && le)
+                        hipLaunchKernelGGL((k_tie_chunk<true>),
+                                           dim3(mgrid), dim3(256), 0, s,
+                                           d_in, t_idx, m, rec_size, off,
+                                           t_chunk);
+                    else
+                        hipLaunchKernelGGL((k_tie_chunk<false>),
+                                           dim3(mgrid), dim3(256), 0, s,
+                                           d_in, t_idx, m, rec_size, off,
+                                           t_chunk);
+                    T9_LAUNCH_CHECK();
+                    rc = t9_sort_pairs_u64_u32(ctx, t_chunk, t_idx, m,
+                                               pair_ws, stream);
+                }
             }
             if (!rc) {
                 hipLaunchKernelGGL(k_tie_scatter, dim3(mgrid), dim3(256),
