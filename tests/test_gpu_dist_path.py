@@ -64,3 +64,33 @@ def test_wordcount_distributed_branch_world1(dist_world1, oracle):
     assert np.array_equal(gk[order], ek)
     assert np.array_equal(gv[order], ev)
     wc.close()
+
+
+def test_wordcount128_distributed_branch_world1(dist_world1, oracle):
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from thrill_amd.pipeline import WordCount, zipf_cdf
+
+    # the 128-bit (config-4 string identity) pipeline through the
+    # distributed branch: pre-reduce -> bucket_mod partition -> C-ABI
+    # exchange (loopback) -> final reduce
+    n, vocab = 1 << 18, 20_000
+    wc = WordCount(n, vocab, 1.1, seed=9, rank=0, world=1, device=0)
+    assert wc.keys128
+    wc.generate()
+    k1, k2, v, m = wc.step()
+    g1 = G.host(k1, np.uint64)
+    g2 = G.host(k2, np.uint64)
+    gv = G.host(v, np.uint64)
+    toks = oracle.zipf_tokens(zipf_cdf(vocab, 1.1), n, seed=9)
+    ids, cnt = np.unique(toks, return_counts=True)
+
+    def h(salt, x):
+        val = oracle.hash128to64(salt, int(x))
+        return val ^ 1 if val == 2**64 - 1 else val
+
+    ek = sorted((h(0x9AE16A3B2F90404F, i), h(0xC3A5C85C97CB3127, i),
+                 int(c)) for i, c in zip(ids, cnt))
+    got = sorted(zip(g1.tolist(), g2.tolist(), gv.tolist()))
+    assert got == ek
+    wc.close()
