@@ -56,3 +56,22 @@ def test_sort_pairs_quarter_billion(nat):
     assert bool((signed[1:] >= signed[:-1]).all().item())
     # permutation check: payload indices sum preserved
     assert int(dv.to(torch.int64).sum().item()) == n * (n - 1) // 2
+
+
+def test_sort_pairs_2e9_tile_base_wrap(nat):
+    """n past ~1.88e9: the pass-2 grid's blockIdx.x * TILE product
+    exceeds 2^32, the exact region where the round-1 u32 tile base
+    wrapped and silently re-scattered a stale tile.
+    Validated by key-sum conservation + sortedness + payload-sum
+    preservation at n = 2e9 (~48 GB of buffers)."""
+    n = 2_000_000_000
+    dk = G.empty(n, np.uint64)
+    nat.gen_u64(G.ptr(dk), 0, n, 0xABC, G.stream())
+    dv = torch.arange(n, dtype=torch.int32, device="cuda")
+    insum = int(dk.sum().item())
+    w = G.ws(nat.ws("sort_pairs", n))
+    nat.sort_pairs_u64_u32(G.ptr(dk), G.ptr(dv), n, G.ptr(w), G.stream())
+    assert int(dk.sum().item()) == insum
+    signed = dk ^ (-2 ** 63)
+    assert bool((signed[1:] >= signed[:-1]).all().item())
+    assert int(dv.to(torch.int64).sum().item()) == n * (n - 1) // 2
