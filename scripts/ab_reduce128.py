@@ -27,8 +27,9 @@ cap = 1 << 25
 tbl = G.empty(3 * cap, np.uint64)
 derr = G.empty(1, np.uint32)
 res = {"n": n, "vocab": vocab}
-for slots in [1024, 2048, 4096]:
+for slots, rf in [(2048, 0), (4096, 0), (4096, 1)]:
     os.environ["T9_LDS128_SLOTS"] = str(slots)
+    os.environ["T9_R128_READFIRST"] = str(rf)
     s = G.stream()
 
     def one():
@@ -41,8 +42,9 @@ for slots in [1024, 2048, 4096]:
     for _ in range(3):
         one()
     torch.cuda.synchronize()
-    res[f"slots{slots}_ms"] = round((time.perf_counter() - t0) / 3 * 1e3,
-                                    2)
+    res[f"slots{slots}_rf{rf}_ms"] = round(
+        (time.perf_counter() - t0) / 3 * 1e3, 2)
 del os.environ["T9_LDS128_SLOTS"]
+del os.environ["T9_R128_READFIRST"]
 print(json.dumps(res), flush=True)
 nat.close()

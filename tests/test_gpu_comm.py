@@ -40,6 +40,27 @@ def test_comm_bootstrap_world1():
         nat.close()
 
 
+def test_rccl_selfsend_4gib_chunked():
+    """regression for the silent RCCL p2p truncation at >= 2^32 bytes
+    (profiles/r02_selfnccl_truncation.json): with t9_alltoall's 1 GiB
+    chunking, a forced 4 GiB RCCL self-send must deliver every byte.
+    Separate process: T9_A2A_SELF latches process-globally."""
+    import json
+    import os
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, os.path.join(repo, "scripts",
+                                      "probe_self_nccl.py"),
+         str(1 << 32)],
+        capture_output=True, text=True, timeout=240)
+    assert r.returncode == 0, r.stderr[-800:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["ok"] is True and d["bytes"] == 1 << 32
+
+
 def test_alltoall_world1_shortcut(nat):
     # world==1: t9_alltoall is a device memcpy honoring displacements
     n = 100_000

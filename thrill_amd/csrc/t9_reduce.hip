@@ -278,20 +278,47 @@ __global__ __launch_bounds__(256) void k_reduce128_init(
 }
 
 /* global-table composite insert (shared by the build's cold path and the
- * LDS flush) */
+ * LDS flush). READFIRST: probe with a plain load and only CAS when the
+ * slot reads empty — hot re-inserts then cost loads + one ADD instead
+ * of CAS round trips (correct: a non-empty k1/k2 is immutable, and a
+ * racing claim is re-checked by the CAS). */
+template <bool READFIRST>
 __device__ inline void t9_g128_insert(u64* __restrict__ t, u64 cap,
                                       u64 salt, u64 k1, u64 k2, u64 v,
                                       u32* __restrict__ err) {
     u64 slot = t9_hash128to64(salt, k1) & (cap - 1);
     u64 probes = 0;
     for (;;) {
-        u64 p1 = atomicCAS((unsigned long long*)&t[3 * slot],
+        u64 p1;
+        if (READFIRST) {
+            p1 = __atomic_load_n((unsigned long long*)&t[3 * slot],
+                                 __ATOMIC_RELAXED);
+            if (p1 == T9_EMPTY)
+                p1 = atomicCAS((unsigned long long*)&t[3 * slot],
+                               (unsigned long long)T9_EMPTY,
+                               (unsigned long long)k1);
+        }
+        else {
+            p1 = atomicCAS((unsigned long long*)&t[3 * slot],
                            (unsigned long long)T9_EMPTY,
                            (unsigned long long)k1);
+        }
         if (p1 == T9_EMPTY || p1 == k1) {
-            u64 p2 = atomicCAS((unsigned long long*)&t[3 * slot + 1],
+            u64 p2;
+            if (READFIRST) {
+                p2 = __atomic_load_n(
+                    (unsigned long long*)&t[3 * slot + 1],
+                    __ATOMIC_RELAXED);
+                if (p2 == T9_EMPTY)
+                    p2 = atomicCAS((unsigned long long*)&t[3 * slot + 1],
+                                   (unsigned long long)T9_EMPTY,
+                                   (unsigned long long)k2);
+            }
+            else {
+                p2 = atomicCAS((unsigned long long*)&t[3 * slot + 1],
                                (unsigned long long)T9_EMPTY,
                                (unsigned long long)k2);
+            }
             if (p2 == T9_EMPTY || p2 == k2) {
                 atomicAdd((unsigned long long*)&t[3 * slot + 2],
                           (unsigned long long)v);
@@ -310,7 +337,7 @@ __device__ inline void t9_g128_insert(u64* __restrict__ t, u64 cap,
 /* LDS-accumulated 128-bit build: per-block (k1, k2, sum) filter table
  * absorbing the Zipf head at LDS-atomic speed (same design as
  * k_reduce_build_lds, composite equality). SLOTS x 24 B of LDS. */
-template <int SLOTS>
+template <int SLOTS, bool READFIRST = false>
 __global__ __launch_bounds__(256) void k_reduce128_build_lds(
     const u64* __restrict__ k1s, const u64* __restrict__ k2s,
     const u64* __restrict__ vals, u64 n, u64* __restrict__ t, u64 cap,
@@ -336,13 +363,33 @@ __global__ __launch_bounds__(256) void k_reduce128_build_lds(
         u32 ls = (u32)(h >> 48) & (SLOTS - 1);
         bool done = false;
         for (int p = 0; p < 4; ++p) {
-            u64 p1 = atomicCAS((unsigned long long*)&lk1[ls],
+            u64 p1;
+            if (READFIRST) {
+                p1 = lk1[ls];
+                if (p1 == T9_EMPTY)
+                    p1 = atomicCAS((unsigned long long*)&lk1[ls],
+                                   (unsigned long long)T9_EMPTY,
+                                   (unsigned long long)k1);
+            }
+            else {
+                p1 = atomicCAS((unsigned long long*)&lk1[ls],
                                (unsigned long long)T9_EMPTY,
                                (unsigned long long)k1);
+            }
             if (p1 == T9_EMPTY || p1 == k1) {
-                u64 p2 = atomicCAS((unsigned long long*)&lk2[ls],
+                u64 p2;
+                if (READFIRST) {
+                    p2 = lk2[ls];
+                    if (p2 == T9_EMPTY)
+                        p2 = atomicCAS((unsigned long long*)&lk2[ls],
+                                       (unsigned long long)T9_EMPTY,
+                                       (unsigned long long)k2);
+                }
+                else {
+                    p2 = atomicCAS((unsigned long long*)&lk2[ls],
                                    (unsigned long long)T9_EMPTY,
                                    (unsigned long long)k2);
+                }
                 if (p2 == T9_EMPTY || p2 == k2) {
                     atomicAdd((unsigned long long*)&lv[ls],
                               (unsigned long long)v);
@@ -353,7 +400,7 @@ __global__ __launch_bounds__(256) void k_reduce128_build_lds(
             ls = (ls + 1) & (SLOTS - 1);
         }
         if (!done)
-            t9_g128_insert(t, cap, salt, k1, k2, v, err);
+            t9_g128_insert<READFIRST>(t, cap, salt, k1, k2, v, err);
     }
     __syncthreads();
 
@@ -363,7 +410,7 @@ __global__ __launch_bounds__(256) void k_reduce128_build_lds(
          * k2 == EMPTY only if no inserter ever won it — then its count
          * is 0 and it can be skipped */
         if (lk2[s] == T9_EMPTY) continue;
-        t9_g128_insert(t, cap, salt, lk1[s], lk2[s], lv[s], err);
+        t9_g128_insert<READFIRST>(t, cap, salt, lk1[s], lk2[s], lv[s], err);
     }
 }
 
@@ -718,20 +765,32 @@ int t9_reduce128_build(t9_context* ctx, const u64* d_k1, const u64* d_k2,
     const char* ge = getenv("T9_REDUCE_GRID");
     u32 grid = ge ? (u32)atoi(ge) : grid_for(n);
     if (!ge && grid > 1024) grid = 1024;
+    /* read-before-CAS probing (T9_R128_READFIRST): hot re-inserts pay
+       loads + one ADD instead of CAS round trips */
+    const char* rf = getenv("T9_R128_READFIRST");
+    const bool readfirst = rf && rf[0] == '1';
     T9_PERF_WRAP(
         s, "reduce_build",
-        if (slots >= 4096)
-            hipLaunchKernelGGL(k_reduce128_build_lds<4096>, dim3(grid),
-                               dim3(256), 0, s, d_k1, d_k2, d_vals, n,
-                               d_table, cap, salt, d_error);
+        if (slots >= 4096 && readfirst)
+            hipLaunchKernelGGL((k_reduce128_build_lds<4096, true>),
+                               dim3(grid), dim3(256), 0, s, d_k1, d_k2,
+                               d_vals, n, d_table, cap, salt, d_error);
+        else if (slots >= 4096)
+            hipLaunchKernelGGL((k_reduce128_build_lds<4096, false>),
+                               dim3(grid), dim3(256), 0, s, d_k1, d_k2,
+                               d_vals, n, d_table, cap, salt, d_error);
         else if (slots <= 1024)
-            hipLaunchKernelGGL(k_reduce128_build_lds<1024>, dim3(grid),
-                               dim3(256), 0, s, d_k1, d_k2, d_vals, n,
-                               d_table, cap, salt, d_error);
+            hipLaunchKernelGGL((k_reduce128_build_lds<1024, false>),
+                               dim3(grid), dim3(256), 0, s, d_k1, d_k2,
+                               d_vals, n, d_table, cap, salt, d_error);
+        else if (readfirst)
+            hipLaunchKernelGGL((k_reduce128_build_lds<2048, true>),
+                               dim3(grid), dim3(256), 0, s, d_k1, d_k2,
+                               d_vals, n, d_table, cap, salt, d_error);
         else
-            hipLaunchKernelGGL(k_reduce128_build_lds<2048>, dim3(grid),
-                               dim3(256), 0, s, d_k1, d_k2, d_vals, n,
-                               d_table, cap, salt, d_error));
+            hipLaunchKernelGGL((k_reduce128_build_lds<2048, false>),
+                               dim3(grid), dim3(256), 0, s, d_k1, d_k2,
+                               d_vals, n, d_table, cap, salt, d_error));
     T9_LAUNCH_CHECK();
     return T9_OK;
 }
