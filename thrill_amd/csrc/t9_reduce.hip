@@ -113,7 +113,7 @@ __global__ __launch_bounds__(256) void k_reduce_build(
 /* SLOTS trades hot-key coverage against occupancy (LDS bytes/block):
  * 1024 -> 8 blocks/CU, 2048 -> 4 (default), 4096 -> 2. Selected with
  * T9_LDS_SLOTS. */
-template <int SLOTS>
+template <int SLOTS, bool READFIRST = false>
 __global__ __launch_bounds__(256) void k_reduce_build_lds(
     const u64* __restrict__ keys, const u64* __restrict__ vals, u64 n,
     u64* __restrict__ t, u64 cap, u64 salt,
@@ -186,9 +186,19 @@ __global__ __launch_bounds__(256) void k_reduce_build_lds(
 
         bool done = false;
         for (int p = 0; p < 4; ++p) {
-            u64 prev = atomicCAS((unsigned long long*)&lk[ls],
+            u64 prev;
+            if (READFIRST) {
+                prev = lk[ls];
+                if (prev == T9_EMPTY)
+                    prev = atomicCAS((unsigned long long*)&lk[ls],
+                                     (unsigned long long)T9_EMPTY,
+                                     (unsigned long long)k);
+            }
+            else {
+                prev = atomicCAS((unsigned long long*)&lk[ls],
                                  (unsigned long long)T9_EMPTY,
                                  (unsigned long long)k);
+            }
             if (prev == T9_EMPTY || prev == k) {
                 atomicAdd((unsigned long long*)&lv[ls],
                           (unsigned long long)gsum);
@@ -203,9 +213,22 @@ __global__ __launch_bounds__(256) void k_reduce_build_lds(
             u64 slot = h & (cap - 1);
             u64 probes = 0;
             for (;;) {
-                u64 prev = atomicCAS((unsigned long long*)&t[2 * slot],
+                u64 prev;
+                if (READFIRST) {
+                    prev = __atomic_load_n(
+                        (unsigned long long*)&t[2 * slot],
+                        __ATOMIC_RELAXED);
+                    if (prev == T9_EMPTY)
+                        prev = atomicCAS(
+                            (unsigned long long*)&t[2 * slot],
+                            (unsigned long long)T9_EMPTY,
+                            (unsigned long long)k);
+                }
+                else {
+                    prev = atomicCAS((unsigned long long*)&t[2 * slot],
                                      (unsigned long long)T9_EMPTY,
                                      (unsigned long long)k);
+                }
                 if (prev == T9_EMPTY || prev == k) {
                     atomicAdd((unsigned long long*)&t[2 * slot + 1],
                               (unsigned long long)gsum);
@@ -707,23 +730,32 @@ int t9_reduce_build(t9_context* ctx, const u64* d_keys, const u64* d_vals,
             const char* ge = getenv("T9_REDUCE_GRID");
             u32 grid = ge ? (u32)atoi(ge) : grid_for(n);
             if (!ge && grid > 1024) grid = 1024;
+            /* read-before-CAS probing (A/B via T9_REDUCE_READFIRST;
+               measured on the 128-bit table: -12%) */
+            const char* urf = getenv("T9_REDUCE_READFIRST");
+            const bool rf = urf && urf[0] == '1';
             if (slots >= 8192)
-                hipLaunchKernelGGL(k_reduce_build_lds<8192>,
+                hipLaunchKernelGGL((k_reduce_build_lds<8192, false>),
                                    dim3(grid), dim3(256), 0, s,
                                    d_keys, d_vals, n, d_table, cap, salt,
                                    d_error, wavecomb);
             else if (slots <= 1024)
-                hipLaunchKernelGGL(k_reduce_build_lds<1024>,
+                hipLaunchKernelGGL((k_reduce_build_lds<1024, false>),
+                                   dim3(grid), dim3(256), 0, s,
+                                   d_keys, d_vals, n, d_table, cap, salt,
+                                   d_error, wavecomb);
+            else if (slots >= 4096 && rf)
+                hipLaunchKernelGGL((k_reduce_build_lds<4096, true>),
                                    dim3(grid), dim3(256), 0, s,
                                    d_keys, d_vals, n, d_table, cap, salt,
                                    d_error, wavecomb);
             else if (slots >= 4096)
-                hipLaunchKernelGGL(k_reduce_build_lds<4096>,
+                hipLaunchKernelGGL((k_reduce_build_lds<4096, false>),
                                    dim3(grid), dim3(256), 0, s,
                                    d_keys, d_vals, n, d_table, cap, salt,
                                    d_error, wavecomb);
             else
-                hipLaunchKernelGGL(k_reduce_build_lds<2048>,
+                hipLaunchKernelGGL((k_reduce_build_lds<2048, false>),
                                    dim3(grid), dim3(256), 0, s,
                                    d_keys, d_vals, n, d_table, cap, salt,
                                    d_error, wavecomb);
@@ -767,8 +799,10 @@ int t9_reduce128_build(t9_context* ctx, const u64* d_k1, const u64* d_k2,
     if (!ge && grid > 1024) grid = 1024;
     /* read-before-CAS probing (T9_R128_READFIRST): hot re-inserts pay
        loads + one ADD instead of CAS round trips */
+    /* default ON: measured 30.5 vs 34.6 ms per 2^29 tokens at 10M
+       vocab (scripts/ab_reduce128.py) */
     const char* rf = getenv("T9_R128_READFIRST");
-    const bool readfirst = rf && rf[0] == '1';
+    const bool readfirst = !(rf && rf[0] == '0');
     T9_PERF_WRAP(
         s, "reduce_build",
         if (slots >= 4096 && readfirst)
