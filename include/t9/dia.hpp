@@ -353,22 +353,20 @@ inline DIA<KeyValue> ReducePair(const DIA<KeyValue>& input,
     using KV = KeyValue;
     Context& ctx = input.context();
     size_t n = input.Size();
-    // SoA split on host? No — pairs are (first,second) adjacent u64s on
-    // device; strided access is a marginal cost at test sizes, so split
-    // via a host roundtrip-free device copy: treat as interleaved and
-    // split with two strided memcpys.
-    auto host = input.AllGather();
-    std::vector<uint64_t> hk(n), hv(n);
-    for (size_t i = 0; i < n; ++i) {
-        hk[i] = host[i].key;
-        hv[i] = host[i].value;
-    }
-    DeviceBuf dk(n * 8), dv(n * 8);
+    // SoA split ON DEVICE: the pairs are interleaved u64s, i.e. 16-byte
+    // records with a u64 field at offset 0 (the key) and offset 8 (the
+    // value) — two t9_extract_key64_le passes split them without any
+    // host round trip (round-1 split staged through host vectors).
+    DeviceBuf dk(n * 8), dv(n * 8), didx(n * 4);
     if (n) {
-        T9_DIA_HIP(hipMemcpy(dk.ptr, hk.data(), n * 8,
-                             hipMemcpyHostToDevice));
-        T9_DIA_HIP(hipMemcpy(dv.ptr, hv.data(), n * 8,
-                             hipMemcpyHostToDevice));
+        T9_DIA_TRY(t9_extract_key64_le(ctx.native(),
+                                       (const uint8_t*)input.device_ptr(),
+                                       n, 16, 0, (uint64_t*)dk.ptr,
+                                       (uint32_t*)didx.ptr, ctx.stream()));
+        T9_DIA_TRY(t9_extract_key64_le(ctx.native(),
+                                       (const uint8_t*)input.device_ptr(),
+                                       n, 16, 8, (uint64_t*)dv.ptr,
+                                       (uint32_t*)didx.ptr, ctx.stream()));
     }
     uint64_t cap = 1024;
     while (cap < 2 * n + 2) cap <<= 1;   // no grow/spill: size for 2x
