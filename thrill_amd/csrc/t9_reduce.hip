@@ -730,10 +730,13 @@ int t9_reduce_build(t9_context* ctx, const u64* d_keys, const u64* d_vals,
             const char* ge = getenv("T9_REDUCE_GRID");
             u32 grid = ge ? (u32)atoi(ge) : grid_for(n);
             if (!ge && grid > 1024) grid = 1024;
-            /* read-before-CAS probing (A/B via T9_REDUCE_READFIRST;
-               measured on the 128-bit table: -12%) */
+            /* read-before-CAS probing, default ON: measured 14.65 vs
+               18.44 ms per 2^29 tokens at 10M vocab (-21%,
+               scripts/ab_reduce_u64_rf.py) — occupied-slot probes and
+               hot re-inserts pay plain loads instead of CAS round
+               trips. T9_REDUCE_READFIRST=0 restores always-CAS. */
             const char* urf = getenv("T9_REDUCE_READFIRST");
-            const bool rf = urf && urf[0] == '1';
+            const bool rf = !(urf && urf[0] == '0');
             if (slots >= 8192)
                 hipLaunchKernelGGL((k_reduce_build_lds<8192, false>),
                                    dim3(grid), dim3(256), 0, s,
