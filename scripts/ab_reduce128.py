@@ -24,12 +24,13 @@ nat.zipf_tokens(G.ptr(toks), G.ptr(cdf), vocab, 0, n, 0x44, G.stream())
 k1, k2 = G.empty(n, np.uint64), G.empty(n, np.uint64)
 nat.hash2_of(G.ptr(toks), n, G.ptr(k1), G.ptr(k2), G.stream())
 cap = 1 << 25
-tbl = G.empty(3 * cap, np.uint64)
+tbl = G.empty(4 * cap, np.uint64)   # sized for the stride-4 variant too
 derr = G.empty(1, np.uint32)
 res = {"n": n, "vocab": vocab}
-for slots, rf in [(2048, 0), (4096, 0), (4096, 1)]:
+for slots, rf, sstr in [(4096, 0, 3), (4096, 1, 3), (4096, 1, 4)]:
     os.environ["T9_LDS128_SLOTS"] = str(slots)
     os.environ["T9_R128_READFIRST"] = str(rf)
+    os.environ["T9_R128_STRIDE"] = str(sstr)
     s = G.stream()
 
     def one():
@@ -42,9 +43,10 @@ for slots, rf in [(2048, 0), (4096, 0), (4096, 1)]:
     for _ in range(3):
         one()
     torch.cuda.synchronize()
-    res[f"slots{slots}_rf{rf}_ms"] = round(
+    res[f"slots{slots}_rf{rf}_s{sstr}_ms"] = round(
         (time.perf_counter() - t0) / 3 * 1e3, 2)
 del os.environ["T9_LDS128_SLOTS"]
 del os.environ["T9_R128_READFIRST"]
+del os.environ["T9_R128_STRIDE"]
 print(json.dumps(res), flush=True)
 nat.close()

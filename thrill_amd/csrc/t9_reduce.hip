@@ -290,13 +290,13 @@ __global__ __launch_bounds__(256) void k_reduce_build_lds(
  * ------------------------------------------------------------------ */
 
 __global__ __launch_bounds__(256) void k_reduce128_init(
-    u64* __restrict__ t, u64 cap) {
+    u64* __restrict__ t, u64 cap, u32 sstr) {
     const u64 stride = (u64)gridDim.x * 256;
     for (u64 i = (u64)blockIdx.x * 256 + threadIdx.x; i < cap;
          i += stride) {
-        t[3 * i] = T9_EMPTY;
-        t[3 * i + 1] = T9_EMPTY;
-        t[3 * i + 2] = 0;
+        t[sstr * i] = T9_EMPTY;
+        t[sstr * i + 1] = T9_EMPTY;
+        t[sstr * i + 2] = 0;
     }
 }
 
@@ -305,13 +305,14 @@ __global__ __launch_bounds__(256) void k_reduce128_init(
  * slot reads empty — hot re-inserts then cost loads + one ADD instead
  * of CAS round trips (correct: a non-empty k1/k2 is immutable, and a
  * racing claim is re-checked by the CAS). */
-template <bool READFIRST>
-__device__ inline void t9_g128_insert(u64* __restrict__ t, u64 cap,
+template <bool READFIRST, int SSTR = 3>
+__device__ inline void t9_g128_insert(u64* __restrict__ tb, u64 cap,
                                       u64 salt, u64 k1, u64 k2, u64 v,
                                       u32* __restrict__ err) {
     u64 slot = t9_hash128to64(salt, k1) & (cap - 1);
     u64 probes = 0;
     for (;;) {
+        u64* t = tb + (SSTR - 3) * slot;  /* SSTR=4: slot base 4*slot */
         u64 p1;
         if (READFIRST) {
             p1 = __atomic_load_n((unsigned long long*)&t[3 * slot],
@@ -360,7 +361,7 @@ __device__ inline void t9_g128_insert(u64* __restrict__ t, u64 cap,
 /* LDS-accumulated 128-bit build: per-block (k1, k2, sum) filter table
  * absorbing the Zipf head at LDS-atomic speed (same design as
  * k_reduce_build_lds, composite equality). SLOTS x 24 B of LDS. */
-template <int SLOTS, bool READFIRST = false>
+template <int SLOTS, bool READFIRST = false, int SSTR = 3>
 __global__ __launch_bounds__(256) void k_reduce128_build_lds(
     const u64* __restrict__ k1s, const u64* __restrict__ k2s,
     const u64* __restrict__ vals, u64 n, u64* __restrict__ t, u64 cap,
@@ -423,7 +424,7 @@ __global__ __launch_bounds__(256) void k_reduce128_build_lds(
             ls = (ls + 1) & (SLOTS - 1);
         }
         if (!done)
-            t9_g128_insert<READFIRST>(t, cap, salt, k1, k2, v, err);
+            t9_g128_insert<READFIRST, SSTR>(t, cap, salt, k1, k2, v, err);
     }
     __syncthreads();
 
@@ -433,7 +434,7 @@ __global__ __launch_bounds__(256) void k_reduce128_build_lds(
          * k2 == EMPTY only if no inserter ever won it — then its count
          * is 0 and it can be skipped */
         if (lk2[s] == T9_EMPTY) continue;
-        t9_g128_insert<READFIRST>(t, cap, salt, lk1[s], lk2[s], lv[s], err);
+        t9_g128_insert<READFIRST, SSTR>(t, cap, salt, lk1[s], lk2[s], lv[s], err);
     }
 }
 
@@ -442,7 +443,7 @@ __global__ __launch_bounds__(256) void k_reduce128_build_lds(
 __global__ __launch_bounds__(256) void k_reduce128_drain(
     const u64* __restrict__ t, u64 cap, u64* __restrict__ ok1,
     u64* __restrict__ ok2, u64* __restrict__ ov,
-    u64* __restrict__ out_n) {
+    u64* __restrict__ out_n, u32 sstr) {
     __shared__ u32 s_pre[256];
     __shared__ u64 s_base;
     const u32 tid = threadIdx.x;
@@ -450,7 +451,8 @@ __global__ __launch_bounds__(256) void k_reduce128_drain(
     const u64 gid = (u64)blockIdx.x * 256 + tid;
     u32 mine = 0;
     for (u64 i = gid; i < cap; i += stride)
-        if (t[3 * i] != T9_EMPTY && t[3 * i + 1] != T9_EMPTY) ++mine;
+        if (t[sstr * i] != T9_EMPTY && t[sstr * i + 1] != T9_EMPTY)
+            ++mine;
     s_pre[tid] = mine;
     __syncthreads();
     for (int off = 1; off < 256; off <<= 1) {
@@ -466,10 +468,10 @@ __global__ __launch_bounds__(256) void k_reduce128_drain(
     if (mine) {
         u64 pos = s_base + s_pre[tid] - mine;
         for (u64 i = gid; i < cap; i += stride) {
-            if (t[3 * i] != T9_EMPTY && t[3 * i + 1] != T9_EMPTY) {
-                ok1[pos] = t[3 * i];
-                ok2[pos] = t[3 * i + 1];
-                ov[pos] = t[3 * i + 2];
+            if (t[sstr * i] != T9_EMPTY && t[sstr * i + 1] != T9_EMPTY) {
+                ok1[pos] = t[sstr * i];
+                ok2[pos] = t[sstr * i + 1];
+                ov[pos] = t[sstr * i + 2];
                 ++pos;
             }
         }
@@ -771,12 +773,20 @@ int t9_reduce_build(t9_context* ctx, const u64* d_keys, const u64* d_vals,
     return T9_OK;
 }
 
+/* slot stride: 3 u64s packed (default; table = 3*capacity u64s) or 4
+   (T9_R128_STRIDE=4: 32-B line-aligned slots, table = 4*capacity u64s —
+   the CALLER must allocate accordingly; A/B knob) */
+static u32 r128_stride() {
+    const char* e = getenv("T9_R128_STRIDE");
+    return (e && atoi(e) == 4) ? 4u : 3u;
+}
+
 int t9_reduce128_init(t9_context* ctx, u64* d_table, u64 cap,
                       void* stream) {
     (void)ctx;
     if (!d_table || !is_pow2(cap)) return T9_EINVAL;
     hipLaunchKernelGGL(k_reduce128_init, dim3(grid_for(cap)), dim3(256), 0,
-                       (hipStream_t)stream, d_table, cap);
+                       (hipStream_t)stream, d_table, cap, r128_stride());
     T9_LAUNCH_CHECK();
     return T9_OK;
 }
@@ -806,9 +816,14 @@ int t9_reduce128_build(t9_context* ctx, const u64* d_k1, const u64* d_k2,
        vocab (scripts/ab_reduce128.py) */
     const char* rf = getenv("T9_R128_READFIRST");
     const bool readfirst = !(rf && rf[0] == '0');
+    const bool s4 = r128_stride() == 4;
     T9_PERF_WRAP(
         s, "reduce_build",
-        if (slots >= 4096 && readfirst)
+        if (slots >= 4096 && readfirst && s4)
+            hipLaunchKernelGGL((k_reduce128_build_lds<4096, true, 4>),
+                               dim3(grid), dim3(256), 0, s, d_k1, d_k2,
+                               d_vals, n, d_table, cap, salt, d_error);
+        else if (slots >= 4096 && readfirst)
             hipLaunchKernelGGL((k_reduce128_build_lds<4096, true>),
                                dim3(grid), dim3(256), 0, s, d_k1, d_k2,
                                d_vals, n, d_table, cap, salt, d_error);
@@ -841,7 +856,8 @@ int t9_reduce128_drain(t9_context* ctx, const u64* d_table, u64 cap,
     hipStream_t s = (hipStream_t)stream;
     HIP_TRY(hipMemsetAsync(d_out_n, 0, 8, s));
     hipLaunchKernelGGL(k_reduce128_drain, dim3(grid_for(cap)), dim3(256),
-                       0, s, d_table, cap, d_ok1, d_ok2, d_ov, d_out_n);
+                       0, s, d_table, cap, d_ok1, d_ok2, d_ov, d_out_n,
+                       r128_stride());
     T9_LAUNCH_CHECK();
     return T9_OK;
 }
