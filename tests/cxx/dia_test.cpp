@@ -127,6 +127,40 @@ static void test_word_count(api::Context& ctx) {
     CHECK(got == expect);
 }
 
+/* the generic string ReduceByKey surface (reference
+ * api/reduce_by_key.hpp:241-463 restricted to (string, u64) pairs):
+ * additive reduce_fn -> GPU 128-bit path; non-additive reduce_fn ->
+ * detected by the algebraic probe, host group-fold with the user's
+ * functor. Both must match a std::map oracle. */
+static void test_reduce_by_key_string(api::Context& ctx) {
+    using P = std::pair<std::string, size_t>;
+    std::vector<P> items;
+    std::map<std::string, size_t> sums, maxs;
+    for (int i = 0; i < 5000; ++i) {
+        std::string w = "w" + std::to_string(i % 97);
+        size_t v = (size_t)(i % 13) + 1;
+        items.push_back(P{ w, v });
+        sums[w] += v;
+        maxs[w] = std::max(maxs[w], v);
+    }
+    auto dia = api::FromVector(ctx, items);
+    auto keyx = [](const P& p) -> std::string { return p.first; };
+
+    auto add = dia.ReduceByKey(keyx, [](const P& a, const P& b) -> P {
+        return P(a.first, a.second + b.second);
+    });
+    std::map<std::string, size_t> got_add;
+    for (auto& p : add.AllGather()) got_add[p.first] = p.second;
+    CHECK(got_add == sums);
+
+    auto mx = dia.ReduceByKey(keyx, [](const P& a, const P& b) -> P {
+        return P(a.first, std::max(a.second, b.second));
+    });
+    std::map<std::string, size_t> got_max;
+    for (auto& p : mx.AllGather()) got_max[p.first] = p.second;
+    CHECK(got_max == maxs);
+}
+
 /* WriteBinary/ReadBinary round trip (terasort.cpp:184-200 file mode;
  * on-disk bytes = packed records, data/serialization.hpp:35-48) */
 static void test_binary_io(api::Context& ctx) {
@@ -180,6 +214,7 @@ int main() {
         test_sort_degenerate(ctx);
         test_sort_records(ctx);
         test_word_count(ctx);
+        test_reduce_by_key_string(ctx);
         test_binary_io(ctx);
         test_group_by_key(ctx);
         if (failures == 0)
