@@ -22,8 +22,9 @@ is the PMC byte count per launch from the committed rocprofv3 calibration
 under profiles/ (not measured live — PMC needs its own rocprofv3 pass).
 
 cpu_baseline: the CPU oracle (reference-semantics restatement,
-oracle/t9_oracle.cpp — kind "port") timed on a bounded sample of the same
-workload on this box's host cores (rank 0, N=1 only).
+oracle/t9_oracle.cpp — kind "port") timed on the FULL workload on this
+box's host cores (OpenMP; ~8 s for 10 GiB — within the bounded-sample
+budget, so the whole thing is the sample; rank 0, N=1 only).
 """
 import argparse
 import ctypes

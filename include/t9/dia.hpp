@@ -9,12 +9,18 @@
  * Context construction fails (the CPU restatement lives in oracle/ and is
  * test infrastructure).
  *
- * Round-1 execution model: eager per-op evaluation (each DOp runs when
+ * Execution model: eager per-op evaluation (each DOp runs when
  * constructed) on a single rank; the reference's lazy Stage/Execute
  * machinery (api/dia_base.cpp:381-443) is unnecessary for the two target
  * pipelines, whose DAGs are straight lines — see DESIGN.md. Multi-rank
- * execution goes through the python pipeline (thrill_amd/pipeline.py) or a
- * caller-provided RCCL communicator on the C ABI.
+ * execution goes through the python pipeline (thrill_amd/pipeline.py) or
+ * the C ABI's own RCCL bootstrap (t9_comm_id/t9_comm_init).
+ *
+ * Round 2 additions, driven by the verbatim example ports (examples/):
+ * host-side DIAs for non-POD items (std::string lines), FlatMap, and a
+ * generic ReduceByKey for (string, u64-counter) pairs that
+ * dictionary-encodes words and reduces on the GPU 128-bit composite
+ * table (string identity — DESIGN.md "Config-4 string identity").
  */
 #pragma once
 

@@ -285,6 +285,33 @@ def test_sort_records_msd_fallback_low_entropy_tails(nat, oracle):
         del os.environ["T9_TIE_MSD"]
 
 
+@pytest.mark.parametrize("seed", [101, 202, 303, 404])
+def test_sort_records_tie_fuzz(nat, oracle, seed):
+    # randomized duplicate structure: random prefix pool sizes, random
+    # low-entropy tail bytes — sweeps MSD levels 0/1/2 and the LSD
+    # fallback across seeds; oracle-exact.
+    rng = np.random.default_rng(seed)
+    n = int(rng.integers(5_000, 60_000))
+    recs = oracle.gen_records(n, seed=seed)
+    pool = int(rng.integers(1, 50))
+    prefixes = rng.integers(0, pool, n)
+    for b in range(8):
+        recs[:, b] = ((prefixes >> max(0, 8 * (3 - b))) & 0xFF) \
+            .astype(np.uint8)
+    # tails: a random subset of byte columns collapsed to tiny alphabets
+    for b in rng.choice(np.arange(10, 100), rng.integers(3, 20),
+                        replace=False):
+        recs[:, b] = rng.integers(0, int(rng.integers(2, 5)), n) \
+            .astype(np.uint8)
+    din = G.dev(recs.reshape(-1))
+    dout = G.empty(n * 100, np.uint8)
+    w = G.ws(nat.ws("sort_records", n, 100))
+    nat.sort_records(G.ptr(din), G.ptr(dout), n, 100, 10, G.ptr(w),
+                     G.stream())
+    got = G.host(dout, np.uint8).reshape(n, 100)
+    assert np.array_equal(got, oracle.sort_records(recs))
+
+
 def test_sort_records_all_identical(nat, oracle):
     # every byte of every record equal: the tie sort's chunk-skip path
     # (no chunk differs, so zero pair sorts run; stable order preserved)
