@@ -48,6 +48,7 @@
 #include <cmath>
 #include <cstdint>
 #include <cstring>
+#include <tuple>
 #include <vector>
 
 #ifdef _OPENMP
@@ -397,6 +398,81 @@ uint64_t t9o_reduce_u64(const uint64_t* keys, const uint64_t* vals,
     for (size_t s = 0; s < out.size(); ++s) {
         out_keys[s] = out[s].first;
         out_vals[s] = out[s].second;
+    }
+    return out.size();
+}
+
+/* ------------------------------------------------------------------ */
+/* 128-bit composite-key reduce — the reference reduces (string, u64)
+ * with EQUALITY ON THE FULL KEY (core/reduce_probing_hash_table.hpp:233
+ * probes compare keys); the MI355X path dictionary-encodes words into
+ * two independent 64-bit hashes and reduces on the composite. This
+ * restatement keeps the same probing-table semantics as t9o_reduce_u64
+ * above (probe start from Hash128to64(salt, k1), equality on the
+ * (k1, k2) pair — a k1 match with a k2 mismatch probes on, exactly the
+ * device protocol, t9_reduce.hip t9_g128_insert). Output sorted by
+ * (k1, k2). Returns the number of pairs, or UINT64_MAX on cap overflow. */
+extern "C" uint64_t t9o_reduce128(const uint64_t* k1s, const uint64_t* k2s,
+                                  const uint64_t* vals, uint64_t n,
+                                  uint64_t salt, uint64_t* out_k1,
+                                  uint64_t* out_k2, uint64_t* out_vals,
+                                  uint64_t cap) {
+    size_t size = 512;
+    std::vector<uint64_t> t1(size, 0), t2(size, 0), tv(size, 0);
+    std::vector<uint8_t> used(size, 0);
+    uint64_t items = 0;
+    const double fill = 0.5;
+
+    auto grow = [&]() {
+        size_t nsize = size * 2;
+        std::vector<uint64_t> n1(nsize, 0), n2(nsize, 0), nv(nsize, 0);
+        std::vector<uint8_t> nu(nsize, 0);
+        for (size_t s2 = 0; s2 < size; ++s2) {
+            if (!used[s2]) continue;
+            size_t idx = (size_t)(t9o_hash128to64(salt, t1[s2]) % nsize);
+            while (nu[idx]) idx = (idx + 1) % nsize;
+            nu[idx] = 1; n1[idx] = t1[s2]; n2[idx] = t2[s2];
+            nv[idx] = tv[s2];
+        }
+        t1.swap(n1); t2.swap(n2); tv.swap(nv); used.swap(nu);
+        size = nsize;
+    };
+
+    for (uint64_t i = 0; i < n; ++i) {
+        const uint64_t k1 = k1s[i], k2 = k2s[i];
+        const uint64_t v = vals ? vals[i] : 1;
+        for (;;) {
+            size_t begin = (size_t)(t9o_hash128to64(salt, k1) % size);
+            size_t idx = begin;
+            bool done = false;
+            while (used[idx]) {
+                if (t1[idx] == k1 && t2[idx] == k2) {
+                    tv[idx] += v; done = true; break;
+                }
+                idx = (idx + 1) % size;
+                if (idx == begin) break;
+            }
+            if (done) break;
+            if (!used[idx]) {
+                used[idx] = 1; t1[idx] = k1; t2[idx] = k2; tv[idx] = v;
+                ++items;
+                while ((double)items >= fill * (double)size) grow();
+                break;
+            }
+            grow();
+        }
+    }
+    std::vector<std::tuple<uint64_t, uint64_t, uint64_t> > out;
+    out.reserve(items);
+    for (size_t s2 = 0; s2 < size; ++s2)
+        if (used[s2])
+            out.push_back(std::make_tuple(t1[s2], t2[s2], tv[s2]));
+    std::sort(out.begin(), out.end());
+    if (out.size() > cap) return UINT64_MAX;
+    for (size_t s2 = 0; s2 < out.size(); ++s2) {
+        out_k1[s2] = std::get<0>(out[s2]);
+        out_k2[s2] = std::get<1>(out[s2]);
+        out_vals[s2] = std::get<2>(out[s2]);
     }
     return out.size();
 }

@@ -115,6 +115,36 @@ class Oracle:
             splitters.reshape(-1), spl_idx, p, out)
         return out
 
+    def reduce128(self, k1, k2, vals=None, salt=0, cap=None):
+        """128-bit composite-key reduce restatement (oracle/t9_oracle.cpp
+        t9o_reduce128): probing-table semantics with equality on the
+        (k1, k2) pair; output sorted by (k1, k2)."""
+        import ctypes
+        import numpy as np
+        k1 = np.ascontiguousarray(k1, dtype=np.uint64)
+        k2 = np.ascontiguousarray(k2, dtype=np.uint64)
+        n = len(k1)
+        if vals is not None:
+            vals = np.ascontiguousarray(vals, dtype=np.uint64)
+        cap = cap or max(16, 2 * n)
+        o1 = np.empty(cap, np.uint64)
+        o2 = np.empty(cap, np.uint64)
+        ov = np.empty(cap, np.uint64)
+        fn = self._lib.t9o_reduce128
+        fn.restype = ctypes.c_uint64
+        m = fn(k1.ctypes.data_as(ctypes.c_void_p),
+               k2.ctypes.data_as(ctypes.c_void_p),
+               vals.ctypes.data_as(ctypes.c_void_p)
+               if vals is not None else None,
+               ctypes.c_uint64(n), ctypes.c_uint64(salt),
+               o1.ctypes.data_as(ctypes.c_void_p),
+               o2.ctypes.data_as(ctypes.c_void_p),
+               ov.ctypes.data_as(ctypes.c_void_p),
+               ctypes.c_uint64(cap))
+        assert m != 2**64 - 1, "oracle reduce128 cap overflow"
+        m = int(m)
+        return o1[:m], o2[:m], ov[:m]
+
     def reduce_u64(self, keys, vals, salt=0, num_partitions=1, cap=None):
         keys = np.ascontiguousarray(keys, dtype=np.uint64)
         vals = np.ascontiguousarray(vals, dtype=np.uint64)
