@@ -80,6 +80,20 @@ def test_reduce128_parity_random(nat):
     assert got == host_reduce128(k1, k2, vals)
 
 
+def test_reduce128_parity_vs_c_oracle(nat, oracle):
+    # the C oracle restatement (oracle/t9_oracle.cpp t9o_reduce128:
+    # probing-table semantics, composite equality) — bit-exact sums
+    rng = np.random.default_rng(31)
+    n = 1 << 18
+    k1 = rng.integers(0, 3000, n).astype(np.uint64)
+    k2 = rng.integers(0, 4, n).astype(np.uint64) + np.uint64(1)
+    vals = rng.integers(0, 1 << 40, n).astype(np.uint64)
+    got = run_reduce128(nat, k1, k2, vals, cap=1 << 15)
+    e1, e2, ev = oracle.reduce128(k1, k2, vals)
+    expect = [((int(a), int(b)), int(v)) for a, b, v in zip(e1, e2, ev)]
+    assert got == expect
+
+
 def test_reduce128_forced_k1_collision(nat):
     # two distinct 'words' sharing k1 (forced single-hash collision)
     # MUST keep separate counts — the observable the reference's
