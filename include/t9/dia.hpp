@@ -473,7 +473,11 @@ DIA<ValueType> DIA<ValueType>::ReduceByKey(const KeyExtractor& key_ex,
         }
     }
 
-    // dictionary-encode: word -> (h1, h2); keep the decode map
+    // dictionary-encode: word -> (h1, h2); keep the decode map. The
+    // decode key folds (h1, h2) into one u64 — a fold collision
+    // (p ~= 2^-64 per pair) would mis-map a word at DECODE time only;
+    // the reduce itself separates on the full 128-bit composite
+    // (DESIGN.md "Config-4 string identity").
     std::vector<uint64_t> k1(n), k2(n), vals(n);
     std::unordered_map<uint64_t, std::string> decode;
     for (size_t i = 0; i < n; ++i) {
