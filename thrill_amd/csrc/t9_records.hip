@@ -407,6 +407,36 @@ __global__ __launch_bounds__(256) void k_iota32(u32* __restrict__ v,
         v[j] = (u32)j;
 }
 
+/* fused permutation applies for the MSD level (one read of the
+ * permutation drives all arrays, replacing per-array gather launches):
+ * after the chunk sort:  a_out[x] = a_in[p[x]], b_out[x] = b_in[p[x]]
+ * after the group sort:  + the chunk values c as a third array */
+__global__ __launch_bounds__(256) void k_msd_apply2(
+    const u32* __restrict__ p, u64 m, const u32* __restrict__ a_in,
+    u32* __restrict__ a_out, const u32* __restrict__ b_in,
+    u32* __restrict__ b_out) {
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 x = (u64)blockIdx.x * 256 + threadIdx.x; x < m; x += stride) {
+        const u32 j = p[x];
+        a_out[x] = a_in[j];
+        b_out[x] = b_in[j];
+    }
+}
+
+__global__ __launch_bounds__(256) void k_msd_apply3(
+    const u32* __restrict__ p, u64 m, const u32* __restrict__ a_in,
+    u32* __restrict__ a_out, const u32* __restrict__ b_in,
+    u32* __restrict__ b_out, const u64* __restrict__ c_in,
+    u64* __restrict__ c_out) {
+    const u64 stride = (u64)gridDim.x * 256;
+    for (u64 x = (u64)blockIdx.x * 256 + threadIdx.x; x < m; x += stride) {
+        const u32 j = p[x];
+        a_out[x] = a_in[j];
+        b_out[x] = b_in[j];
+        c_out[x] = c_in[j];
+    }
+}
+
 /* d_idx[perm[j]] = tidx[j] — write the re-ordered record indices back
  * into the (ascending) tied positions */
 __global__ __launch_bounds__(256) void k_tie_scatter(
@@ -944,13 +974,9 @@ static int sort_records_impl(t9_context* ctx, const u8* d_in, u8* d_out,
                     rc = t9_sort_pairs_u64_u32(ctx, t_chunk, jbuf, m,
                                                pair_ws, stream);
                     if (rc) break;
-                    rc = t9_gather_records(ctx, (const u8*)t_idx, jbuf, m,
-                                           4, (u8*)t_tmp4, stream);
-                    if (!rc)
-                        rc = t9_gather_records(ctx, (const u8*)t_grp,
-                                               jbuf, m, 4, (u8*)t_grp2,
-                                               stream);
-                    if (rc) break;
+                    hipLaunchKernelGGL(k_msd_apply2, dim3(mgrid),
+                                       dim3(256), 0, s, jbuf, m, t_idx,
+                                       t_tmp4, t_grp, t_grp2);
                     hipLaunchKernelGGL(k_grp_keys, dim3(mgrid), dim3(256),
                                        0, s, t_grp2, m, t_chunk2);
                     hipLaunchKernelGGL(k_iota32, dim3(mgrid), dim3(256),
@@ -959,17 +985,11 @@ static int sort_records_impl(t9_context* ctx, const u8* d_in, u8* d_out,
                     rc = t9_sort_pairs_u64_u32(ctx, t_chunk2, jbuf, m,
                                                pair_ws, stream);
                     if (rc) break;
-                    rc = t9_gather_records(ctx, (const u8*)t_tmp4, jbuf,
-                                           m, 4, (u8*)t_idx, stream);
-                    if (!rc)
-                        rc = t9_gather_records(ctx, (const u8*)t_grp2,
-                                               jbuf, m, 4, (u8*)t_grp,
-                                               stream);
-                    if (!rc)
-                        rc = t9_gather_records(ctx, (const u8*)t_chunk,
-                                               jbuf, m, 8, (u8*)t_chunk2,
-                                               stream);
-                    if (rc) break;
+                    hipLaunchKernelGGL(k_msd_apply3, dim3(mgrid),
+                                       dim3(256), 0, s, jbuf, m, t_tmp4,
+                                       t_idx, t_grp2, t_grp, t_chunk,
+                                       t_chunk2);
+                    T9_LAUNCH_CHECK();
                     u32 ties = 0;
                     rc = regroup(t_grp, t_chunk2, t_grp2, &ties);
                     if (rc) break;
