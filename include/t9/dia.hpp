@@ -120,6 +120,117 @@ inline bool read_file_direct(const char* name, void* data, size_t bytes) {
     return ok;
 }
 
+//! device-buffer -> file, O_DIRECT, PINNED ping-pong staging: the
+//! pageable AllGather D2H measured ~1.5-2 GB/s and throttled the whole
+//! write; pinned transfers overlap the previous chunk's disk write.
+//! Returns false to signal the caller's fallback.
+inline bool write_device_direct(const char* name, const void* d_ptr,
+                                size_t bytes, hipStream_t s) {
+    const size_t CH = 16u << 20;
+    int fd = ::open(name, O_WRONLY | O_CREAT | O_TRUNC | O_DIRECT, 0644);
+    if (fd < 0) return false;
+    void* pin[2] = { nullptr, nullptr };
+    if (hipHostMalloc(&pin[0], CH, 0) != hipSuccess ||
+        hipHostMalloc(&pin[1], CH, 0) != hipSuccess) {
+        if (pin[0]) (void)hipHostFree(pin[0]);
+        ::close(fd);
+        return false;
+    }
+    const size_t ALIGN = 4096;
+    const size_t aligned = bytes / ALIGN * ALIGN;
+    bool ok = true;
+    size_t off = 0;
+    int cur = 0;
+    size_t len0 = std::min(CH, aligned);
+    if (len0)
+        ok = hipMemcpyAsync(pin[0], (const char*)d_ptr, len0,
+                            hipMemcpyDeviceToHost, s) == hipSuccess;
+    while (ok && off < aligned) {
+        const size_t len = std::min(CH, aligned - off);
+        ok = hipStreamSynchronize(s) == hipSuccess;
+        const size_t noff = off + len;
+        if (ok && noff < aligned) {
+            const size_t nlen = std::min(CH, aligned - noff);
+            ok = hipMemcpyAsync(pin[cur ^ 1], (const char*)d_ptr + noff,
+                                nlen, hipMemcpyDeviceToHost,
+                                s) == hipSuccess;
+        }
+        ok = ok && ::write(fd, pin[cur], len) == (ssize_t)len;
+        off = noff;
+        cur ^= 1;
+    }
+    ::fsync(fd);
+    ::close(fd);
+    if (ok && aligned < bytes) {
+        ok = hipMemcpy(pin[0], (const char*)d_ptr + aligned,
+                       bytes - aligned, hipMemcpyDeviceToHost) ==
+             hipSuccess;
+        int fd2 = ::open(name, O_WRONLY);
+        ok = ok && fd2 >= 0 &&
+             ::pwrite(fd2, pin[0], bytes - aligned, (off_t)aligned) ==
+                 (ssize_t)(bytes - aligned);
+        if (fd2 >= 0) {
+            ::fsync(fd2);
+            ::close(fd2);
+        }
+    }
+    (void)hipHostFree(pin[0]);
+    (void)hipHostFree(pin[1]);
+    return ok;
+}
+
+//! file -> device buffer, O_DIRECT, pinned ping-pong (mirror of the
+//! writer: disk read of chunk i overlaps the H2D copy of chunk i-1)
+inline bool read_device_direct(const char* name, void* d_ptr,
+                               size_t bytes, hipStream_t s) {
+    const size_t CH = 16u << 20;
+    int fd = ::open(name, O_RDONLY | O_DIRECT);
+    if (fd < 0) return false;
+    void* pin[2] = { nullptr, nullptr };
+    if (hipHostMalloc(&pin[0], CH, 0) != hipSuccess ||
+        hipHostMalloc(&pin[1], CH, 0) != hipSuccess) {
+        if (pin[0]) (void)hipHostFree(pin[0]);
+        ::close(fd);
+        return false;
+    }
+    const size_t ALIGN = 4096;
+    const size_t aligned = bytes / ALIGN * ALIGN;
+    bool ok = true;
+    size_t off = 0;
+    int cur = 0;
+    while (ok && off < aligned) {
+        const size_t len = std::min(CH, aligned - off);
+        ok = ::read(fd, pin[cur], len) == (ssize_t)len;
+        if (ok)
+            ok = hipMemcpyAsync((char*)d_ptr + off, pin[cur], len,
+                                hipMemcpyHostToDevice, s) == hipSuccess;
+        /* the NEXT disk read fills the other pinned buffer while this
+         * H2D drains; block only before REUSING a buffer */
+        cur ^= 1;
+        off += len;
+        if (ok && off < aligned)
+            ok = hipStreamSynchronize(s) == hipSuccess;
+        /* note: full overlap would need per-buffer events; one sync per
+         * chunk already keeps both engines mostly busy at 16 MiB */
+    }
+    ::close(fd);
+    if (ok && aligned < bytes) {
+        int fd2 = ::open(name, O_RDONLY);
+        ok = fd2 >= 0 &&
+             ::pread(fd2, pin[0], bytes - aligned, (off_t)aligned) ==
+                 (ssize_t)(bytes - aligned);
+        if (fd2 >= 0) ::close(fd2);
+        if (ok)
+            ok = hipMemcpyAsync((char*)d_ptr + aligned, pin[0],
+                                bytes - aligned, hipMemcpyHostToDevice,
+                                s) == hipSuccess;
+    }
+    ok = ok && hipStreamSynchronize(s) == hipSuccess;
+    (void)hipHostFree(pin[0]);
+    (void)hipHostFree(pin[1]);
+    return ok;
+}
+
 } // namespace detail
 
 #define T9_DIA_TRY(expr)                                                  \
@@ -329,6 +440,14 @@ public:
         char name[512];
         std::snprintf(name, sizeof(name), "%s%010zu", pathbase.c_str(),
                       ctx_->my_rank());
+        if (buf_ && n_) {
+            // device-resident items: O_DIRECT + pinned ping-pong
+            // (HBM -> disk without the pageable-copy throttle)
+            if (detail::write_device_direct(name, buf_->ptr,
+                                            n_ * sizeof(ValueType),
+                                            ctx_->stream()))
+                return;
+        }
         auto host = AllGather();
         const size_t bytes = host.size() * sizeof(ValueType);
         if (detail::write_file_direct(name, host.data(), bytes))
@@ -639,6 +758,41 @@ DIA<ValueType> DIA<ValueType>::ReduceByKey(const KeyExtractor& key_ex,
 //! items from files (the worker's share; world=1 here reads all).
 template <typename T>
 DIA<T> ReadBinary(Context& ctx, const std::vector<std::string>& files) {
+    // device fast path (POD T): size the files, read each straight into
+    // the device buffer via O_DIRECT + pinned staging (the pageable H2D
+    // of the host-vector path measured ~2 GB/s)
+    if (std::is_trivially_copyable<T>::value && !files.empty()) {
+        size_t total = 0;
+        bool sized = true;
+        std::vector<size_t> sizes;
+        for (const auto& path : files) {
+            std::FILE* f = std::fopen(path.c_str(), "rb");
+            if (!f) throw std::runtime_error("ReadBinary: cannot open " +
+                                             path);
+            std::fseek(f, 0, SEEK_END);
+            long b = std::ftell(f);
+            std::fclose(f);
+            if (b < 0 || b % (long)sizeof(T)) {
+                sized = false;
+                break;
+            }
+            sizes.push_back((size_t)b);
+            total += (size_t)b;
+        }
+        if (sized && total) {
+            auto buf = std::make_shared<DeviceBuf>(total);
+            size_t off = 0;
+            bool ok = true;
+            for (size_t i = 0; ok && i < files.size(); ++i) {
+                ok = detail::read_device_direct(
+                    files[i].c_str(), (char*)buf->ptr + off, sizes[i],
+                    ctx.stream());
+                off += sizes[i];
+            }
+            if (ok) return DIA<T>(&ctx, buf, total / sizeof(T));
+            // else fall through to the host path
+        }
+    }
     std::vector<T> items;
     for (const auto& path : files) {
         std::FILE* f = std::fopen(path.c_str(), "rb");
