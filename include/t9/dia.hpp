@@ -126,7 +126,7 @@ inline bool read_file_direct(const char* name, void* data, size_t bytes) {
 //! Returns false to signal the caller's fallback.
 inline bool write_device_direct(const char* name, const void* d_ptr,
                                 size_t bytes, hipStream_t s) {
-    const size_t CH = 16u << 20;
+    const size_t CH = 64u << 20;   /* 64 MiB: fewer sync/chunk overheads */
     int fd = ::open(name, O_WRONLY | O_CREAT | O_TRUNC | O_DIRECT, 0644);
     if (fd < 0) return false;
     void* pin[2] = { nullptr, nullptr };
@@ -183,7 +183,7 @@ inline bool write_device_direct(const char* name, const void* d_ptr,
 //! writer: disk read of chunk i overlaps the H2D copy of chunk i-1)
 inline bool read_device_direct(const char* name, void* d_ptr,
                                size_t bytes, hipStream_t s) {
-    const size_t CH = 16u << 20;
+    const size_t CH = 64u << 20;
     int fd = ::open(name, O_RDONLY | O_DIRECT);
     if (fd < 0) return false;
     void* pin[2] = { nullptr, nullptr };
