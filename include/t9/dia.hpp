@@ -153,7 +153,6 @@ inline bool write_device_direct(const char* name, const void* d_ptr,
     while (ok && off < aligned) {
         const size_t len = std::min(CH, aligned - off);
         ok = hipStreamSynchronize(s) == hipSuccess;
-        if (ok) std::memcpy(dbuf, pin[cur], len);
         const size_t noff = off + len;
         if (ok && noff < aligned) {
             const size_t nlen = std::min(CH, aligned - noff);
@@ -161,11 +160,15 @@ inline bool write_device_direct(const char* name, const void* d_ptr,
                                 nlen, hipMemcpyDeviceToHost,
                                 s) == hipSuccess;
         }
-        ok = ok && ::write(fd, dbuf, len) == (ssize_t)len;
+        /* write straight from the pinned buffer: an A/B with a plain
+         * aligned bounce measured WORSE on the write side (box-variable
+         * storage; profiles/r02_io_bench.json notes) */
+        ok = ok && ::write(fd, pin[cur], len) == (ssize_t)len;
         off = noff;
         cur ^= 1;
     }
     std::free(dbuf);
+    (void)dbuf;
     ::fsync(fd);
     ::close(fd);
     if (ok && aligned < bytes) {
