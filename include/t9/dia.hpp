@@ -917,6 +917,25 @@ inline DIA<uint64_t> Merge(const DIA<uint64_t>& a, const DIA<uint64_t>& b) {
     return DIA<uint64_t>(&ctx, out, a.Size() + b.Size());
 }
 
+//! Merge for fixed-size POD records under the byte-lexicographic order
+//! (the acceptance comparator; reference api/merge.hpp merges pre-sorted
+//! DIAs with source order among equals — a's before b's)
+template <typename T>
+DIA<T> Merge(const DIA<T>& a, const DIA<T>& b) {
+    static_assert(std::is_trivially_copyable<T>::value &&
+                      sizeof(T) % 4 == 0,
+                  "GPU Merge needs POD records, size % 4 == 0");
+    Context& ctx = a.context();
+    auto out = std::make_shared<DeviceBuf>((a.Size() + b.Size()) *
+                                           sizeof(T));
+    T9_DIA_TRY(t9_merge_records(
+        ctx.native(), (const uint8_t*)a.device_ptr(), a.Size(),
+        (const uint8_t*)b.device_ptr(), b.Size(), sizeof(T),
+        (uint8_t*)out->ptr, ctx.stream()));
+    T9_DIA_HIP(hipStreamSynchronize(ctx.stream()));
+    return DIA<T>(&ctx, out, a.Size() + b.Size());
+}
+
 //! Run — reference api/context.cpp:947: construct the context(s) and run
 //! the job. Round 1: one process, one GPU, rank 0.
 inline int Run(const std::function<void(Context&)>& job) {

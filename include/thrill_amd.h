@@ -273,6 +273,14 @@ int t9_group_index(t9_context* ctx, const uint64_t* d_sorted_keys,
                    uint64_t n, uint64_t* d_unique, uint64_t* d_offsets,
                    uint64_t* d_count, void* d_workspace, void* stream);
 
+/* Merge two byte-lexicographically sorted fixed-size record sequences
+ * (the reference Merge's comparator restricted to the GPU-executable
+ * byte order, api/merge.hpp:368-520); equal records from d_a precede
+ * those from d_b. rec_size % 4 == 0. */
+int t9_merge_records(t9_context* ctx, const uint8_t* d_a, uint64_t na,
+                     const uint8_t* d_b, uint64_t nb, uint32_t rec_size,
+                     uint8_t* d_out, void* stream);
+
 /* Merge (SURVEY.md §8f item 4 — thrill/api/merge.hpp merges pre-sorted
  * DIAs): merge two sorted u64 sequences into d_out (na+nb); equal keys
  * from d_a precede those from d_b. One merge-path pass. */

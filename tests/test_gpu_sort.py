@@ -495,3 +495,42 @@ def test_sort_records_parity_seed_sweep(nat, oracle, seed):
                      G.stream())
     got = G.host(dout, np.uint8).reshape(n, 100)
     assert np.array_equal(got, oracle.sort_records(recs))
+
+
+@pytest.mark.parametrize("rec", [100, 128, 12])
+def test_merge_records_parity(nat, oracle, rec):
+    # t9_merge_records: byte-lex merge of two sorted record sequences,
+    # A-wins ties (reference api/merge.hpp source-order rule)
+    rng = np.random.default_rng(rec)
+    na, nb = 40_000, 55_000
+    A = rng.integers(0, 4, (na, rec)).astype(np.uint8)  # tiny alphabet:
+    B = rng.integers(0, 4, (nb, rec)).astype(np.uint8)  # many equals
+    A = A[np.lexsort(tuple(A[:, c] for c in range(rec - 1, -1, -1)))]
+    B = B[np.lexsort(tuple(B[:, c] for c in range(rec - 1, -1, -1)))]
+    da, db = G.dev(A.reshape(-1)), G.dev(B.reshape(-1))
+    dout = G.empty((na + nb) * rec, np.uint8)
+    nat.merge_records(G.ptr(da), na, G.ptr(db), nb, rec, G.ptr(dout),
+                      G.stream())
+    got = G.host(dout, np.uint8).reshape(na + nb, rec)
+    # expected: stable merge with A-before-B on equals == merging
+    # (record, source) pairs sorted by (record, source)
+    tagged = np.concatenate([A, B])
+    src = np.concatenate([np.zeros(na, np.uint8), np.ones(nb, np.uint8)])
+    order = np.lexsort((src,) + tuple(tagged[:, c]
+                                      for c in range(rec - 1, -1, -1)))
+    # within equal (record, source), keep original order: lexsort stable
+    assert np.array_equal(got, tagged[order])
+
+
+def test_merge_records_empty_sides(nat, oracle):
+    rec = 100
+    recs = oracle.gen_records(1000, seed=3)
+    s = recs[np.lexsort(tuple(recs[:, c] for c in range(99, -1, -1)))]
+    da = G.dev(s.reshape(-1))
+    dout = G.empty(1000 * rec, np.uint8)
+    nat.merge_records(G.ptr(da), 1000, G.ptr(da), 0, rec, G.ptr(dout),
+                      G.stream())
+    assert np.array_equal(G.host(dout, np.uint8).reshape(1000, rec), s)
+    nat.merge_records(G.ptr(da), 0, G.ptr(da), 1000, rec, G.ptr(dout),
+                      G.stream())
+    assert np.array_equal(G.host(dout, np.uint8).reshape(1000, rec), s)

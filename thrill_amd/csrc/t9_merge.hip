@@ -80,7 +80,131 @@ __global__ __launch_bounds__(256) void k_merge_u64(
     }
 }
 
+/* ---- record merge (byte-lexicographic comparator) ----------------- *
+ * The reference's Merge handles arbitrary comparators over whole items
+ * (api/merge.hpp:368-520 pivot search); the GPU surface's executable
+ * order is byte-lexicographic over fixed-size records (the acceptance
+ * total order — TeraSort's Record::operator< is its key-prefix
+ * restriction). Same merge-path structure as k_merge_u64 with
+ * whole-record compares; records are too wide for LDS tiles, so each
+ * thread merges its output span straight from global (L2 catches the
+ * sequential A/B walks). A-wins ties (stable source order). */
+
+/* byte-lex compare of two records, word-wise (bswap makes u32 numeric
+ * order == byte order); returns a <= b */
+template <int RW>
+__device__ inline bool rec_le(const u32* __restrict__ a,
+                              const u32* __restrict__ b, u32 rw_rt) {
+    const u32 rw = RW ? (u32)RW : rw_rt;
+    for (u32 w = 0; w < rw; ++w) {
+        const u32 ua = __builtin_bswap32(a[w]);
+        const u32 ub = __builtin_bswap32(b[w]);
+        if (ua != ub) return ua < ub;
+    }
+    return true;
+}
+
+template <int RW>
+__device__ inline u64 merge_split_rec(const u32* __restrict__ a, u64 na,
+                                      const u32* __restrict__ b, u64 nb,
+                                      u64 d, u32 rw) {
+    u64 lo = (d > nb) ? d - nb : 0;
+    u64 hi = (d < na) ? d : na;
+    const u32 rwe = RW ? (u32)RW : rw;
+    while (lo < hi) {
+        u64 i = (lo + hi) / 2;
+        if (rec_le<RW>(a + i * rwe, b + (d - i - 1) * rwe, rw))
+            lo = i + 1;
+        else
+            hi = i;
+    }
+    return lo;
+}
+
+#define MRG_REC_TILE 1024
+
+template <int RW>
+__global__ __launch_bounds__(256) void k_merge_records(
+    const u32* __restrict__ a, u64 na, const u32* __restrict__ b, u64 nb,
+    u32 rw_rt, u32* __restrict__ out) {
+    const u32 rw = RW ? (u32)RW : rw_rt;
+    const u64 n = na + nb;
+    const u64 d0 = (u64)blockIdx.x * MRG_REC_TILE;
+    if (d0 >= n) return;
+    const u64 d1 = (d0 + MRG_REC_TILE < n) ? d0 + MRG_REC_TILE : n;
+    __shared__ u64 s_meta[2];
+    const u32 tid = threadIdx.x;
+    if (tid == 0) {
+        s_meta[0] = merge_split_rec<RW>(a, na, b, nb, d0, rw);
+        s_meta[1] = merge_split_rec<RW>(a, na, b, nb, d1, rw);
+    }
+    __syncthreads();
+    const u64 i0 = s_meta[0], i1 = s_meta[1];
+    const u64 j0 = d0 - i0;
+    const u64 la = i1 - i0, lb = (d1 - i1) - j0;
+    const u32 span = MRG_REC_TILE / 256;   /* 4 records per thread */
+    const u64 od = (u64)tid * span;
+    if (d0 + od >= d1) return;
+    /* local split within (a[i0..i0+la), b[j0..j0+lb)) */
+    u64 lo = (od > lb) ? od - lb : 0;
+    u64 hi = (od < la) ? od : la;
+    while (lo < hi) {
+        u64 i = (lo + hi) / 2;
+        if (rec_le<RW>(a + (i0 + i) * rw, b + (j0 + od - i - 1) * rw, rw))
+            lo = i + 1;
+        else
+            hi = i;
+    }
+    u64 ia = i0 + lo, ib = j0 + (od - lo);
+    const u64 ea = i0 + la, eb = j0 + lb;
+    const u64 end = ((d1 - d0 - od) < span) ? (d1 - d0 - od) : span;
+    for (u64 k = 0; k < end; ++k) {
+        const bool take_a =
+            ib >= eb || (ia < ea && rec_le<RW>(a + ia * rw, b + ib * rw,
+                                              rw));
+        const u32* src = take_a ? a + (ia++) * rw : b + (ib++) * rw;
+        u32* dst = out + (d0 + od + k) * rw;
+        for (u32 w = 0; w < rw; ++w) dst[w] = src[w];
+    }
+}
+
 extern "C" {
+
+/* Merge two byte-lexicographically sorted fixed-size record sequences;
+ * equal records from d_a precede those from d_b. rec_size % 4 == 0. */
+int t9_merge_records(t9_context* ctx, const uint8_t* d_a, u64 na,
+                     const uint8_t* d_b, u64 nb, u32 rec_size,
+                     uint8_t* d_out, void* stream) {
+    (void)ctx;
+    if (rec_size == 0 || rec_size % 4) return T9_EINVAL;
+    hipStream_t s = (hipStream_t)stream;
+    const u64 n = na + nb;
+    if (n == 0) return T9_OK;
+    if (!d_out || (na && !d_a) || (nb && !d_b)) return T9_EINVAL;
+    if (na == 0 || nb == 0) {
+        const void* src = na ? (const void*)d_a : (const void*)d_b;
+        HIP_TRY(hipMemcpyAsync(d_out, src, n * rec_size,
+                               hipMemcpyDeviceToDevice, s));
+        return T9_OK;
+    }
+    const u64 B = t9_ceil_div(n, MRG_REC_TILE);
+    if (B >= (1ull << 31)) return T9_EINVAL;
+    const u32 rw = rec_size / 4;
+    if (rw == 25)
+        hipLaunchKernelGGL((k_merge_records<25>), dim3((u32)B), dim3(256),
+                           0, s, (const u32*)d_a, na, (const u32*)d_b,
+                           nb, rw, (u32*)d_out);
+    else if (rw == 32)
+        hipLaunchKernelGGL((k_merge_records<32>), dim3((u32)B), dim3(256),
+                           0, s, (const u32*)d_a, na, (const u32*)d_b,
+                           nb, rw, (u32*)d_out);
+    else
+        hipLaunchKernelGGL((k_merge_records<0>), dim3((u32)B), dim3(256),
+                           0, s, (const u32*)d_a, na, (const u32*)d_b,
+                           nb, rw, (u32*)d_out);
+    T9_LAUNCH_CHECK();
+    return T9_OK;
+}
 
 /* Merge two sorted u64 sequences into out (size na+nb); equal keys from
  * d_a precede those from d_b (stable source order). */

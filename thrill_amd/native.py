@@ -73,6 +73,7 @@ _SIGS = {
     "t9_bucket_mod": (i32, [vp, vp, u64, u32, vp, vp, vp]),
     "t9_zipf_tokens": (i32, [vp, vp, vp, u64, u64, u64, u64, vp]),
     "t9_merge_u64": (i32, [vp, vp, u64, vp, u64, vp, vp]),
+    "t9_merge_records": (i32, [vp, vp, u64, vp, u64, u32, vp, vp]),
     "t9_group_index_workspace": (u64, [u64]),
     "t9_group_index": (i32, [vp, vp, u64, vp, vp, vp, vp, vp]),
     "t9_perf_enable": (i32, [i32]),
