@@ -10,6 +10,7 @@
 #include <algorithm>
 #include <cstdio>
 #include <fstream>
+#include <iterator>
 #include <map>
 #include <random>
 #include <sstream>
@@ -185,6 +186,39 @@ static void test_binary_io(api::Context& ctx) {
     std::remove("/tmp/t9_dia_io-0000000000");
 }
 
+/* Merge (api/merge.hpp): u64 and whole-record byte-lex variants, A-wins
+ * ties (source order among equals) */
+static void test_merge(api::Context& ctx) {
+    std::vector<uint64_t> a, b;
+    for (uint64_t i = 0; i < 20000; ++i) a.push_back(2 * i);
+    for (uint64_t i = 0; i < 15000; ++i) b.push_back(3 * i);
+    auto m = api::Merge(api::FromVector(ctx, a), api::FromVector(ctx, b));
+    auto out = m.AllGather();
+    std::vector<uint64_t> expect = a;
+    expect.insert(expect.end(), b.begin(), b.end());
+    std::sort(expect.begin(), expect.end());
+    CHECK(out == expect);
+
+    std::mt19937_64 rng(11);
+    std::vector<Record> ra(5000), rb(7000);
+    for (auto* v : { &ra, &rb })
+        for (auto& r : *v)
+            for (size_t j = 0; j < sizeof(Record); ++j)
+                ((uint8_t*)&r)[j] = (uint8_t)(rng() & 3);  /* many ties */
+    std::sort(ra.begin(), ra.end());
+    std::sort(rb.begin(), rb.end());
+    auto rm = api::Merge(api::FromVector(ctx, ra),
+                         api::FromVector(ctx, rb));
+    auto rout = rm.AllGather();
+    std::vector<Record> rexpect;
+    std::merge(ra.begin(), ra.end(), rb.begin(), rb.end(),
+               std::back_inserter(rexpect));   /* std::merge: a-wins */
+    CHECK(rout.size() == rexpect.size());
+    bool ok = true;
+    for (size_t i = 0; i < rout.size(); ++i) ok &= rout[i] == rexpect[i];
+    CHECK(ok);
+}
+
 /* GroupByKey (api/group_by_key.hpp): per-key value collections */
 static void test_group_by_key(api::Context& ctx) {
     const size_t n = 50000;
@@ -216,6 +250,7 @@ int main() {
         test_word_count(ctx);
         test_reduce_by_key_string(ctx);
         test_binary_io(ctx);
+        test_merge(ctx);
         test_group_by_key(ctx);
         if (failures == 0)
             std::printf("dia_test: all checks passed\n");
