@@ -752,6 +752,9 @@ static int sort_records_impl(t9_context* ctx, const u8* d_in, u8* d_out,
     hipStream_t s = (hipStream_t)stream;
     if (n == 0) return T9_OK;
     if (!d_in || !d_out || !d_workspace) return T9_EINVAL;
+    /* d_in must be preserved (API contract) and d_out doubles as tie
+     * scratch — aliasing would corrupt the input */
+    if (d_in == d_out) return T9_EINVAL;
     if (n == 1) {
         HIP_TRY(hipMemcpyAsync(d_out, d_in, rec_size, hipMemcpyDeviceToDevice,
                                s));
