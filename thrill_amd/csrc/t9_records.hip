@@ -471,14 +471,18 @@ __global__ __launch_bounds__(256) void k_zipf_tokens(
  * counts come out of the same pass, eliminating the separate pass-1
  * histogram read (k_hist_msb). Output layout identical to
  * k_extract_key64 + k_hist_msb<0> at shift 56. */
-template <int RW, bool LE>
+template <int RW, bool LE, bool DB = true>
 __global__ __launch_bounds__(256) void k_extract_hist(
     const u8* __restrict__ recs, u64 n, u64* __restrict__ keys,
     u32* __restrict__ idx, u32* __restrict__ hist) {
     constexpr int SUBREC = 256;                /* records per LDS stage */
     constexpr int SUBW = SUBREC * RW;          /* words per stage */
-    __shared__ u32 s_buf[2][SUBW];             /* double-buffered: one
-                                                  barrier per stage */
+    /* DB: double-buffered (one barrier per stage, 3 blocks/CU at 100-B
+       records) vs single-buffered (two barriers, 6 blocks/CU) — the
+       wave-cycle decomposition shows this kernel parked on the stage
+       barrier, so occupancy vs barrier count is the tradeoff to measure
+       (T9_EXTRACT_SB=1 selects single). */
+    __shared__ u32 s_buf[DB ? 2 : 1][SUBW];
     __shared__ u32 s_cnt[256];
     const u32 tid = threadIdx.x;
     const u64 tile0 = (u64)blockIdx.x * 8192;
@@ -492,11 +496,19 @@ __global__ __launch_bounds__(256) void k_extract_hist(
         const u32 wn = sn * RW;
         for (u32 w = tid; w < wn; w += 256) s_buf[buf][w] = rin[w0 + w];
     };
-    load_stage(0, 0);
-    __syncthreads();
+    if (DB) {
+        load_stage(0, 0);
+        __syncthreads();
+    }
     int cur = 0;
-    for (u32 s0 = 0; s0 < tn; s0 += SUBREC, cur ^= 1) {
-        if (s0 + SUBREC < tn) load_stage(s0 + SUBREC, cur ^ 1);
+    for (u32 s0 = 0; s0 < tn; s0 += SUBREC, cur ^= DB ? 1 : 0) {
+        if (!DB) {
+            load_stage(s0, 0);
+            __syncthreads();
+        }
+        else if (s0 + SUBREC < tn) {
+            load_stage(s0 + SUBREC, cur ^ 1);
+        }
         const u32 sn = (tn - s0 < SUBREC) ? tn - s0 : SUBREC;
         const bool valid = tid < sn;
         u64 k = 0;
@@ -795,9 +807,15 @@ static int sort_records_impl(t9_context* ctx, const u8* d_in, u8* d_out,
     if (eh && !fused) {
         u32* hist = t9i_msb_pass1_hist(pair_ws, n);
         const u64 B = t9_ceil_div(n, 8192);
+        const char* sbe = getenv("T9_EXTRACT_SB");
+        const bool sb = sbe && sbe[0] == '1';
         T9_PERF_WRAP(
             s, "extract",
-            if (rw == 25 && !le)
+            if (rw == 25 && !le && sb)
+                hipLaunchKernelGGL((k_extract_hist<25, false, false>),
+                                   dim3((u32)B), dim3(256), 0, s, d_in, n,
+                                   d_keys, d_idx, hist);
+            else if (rw == 25 && !le)
                 hipLaunchKernelGGL((k_extract_hist<25, false>),
                                    dim3((u32)B), dim3(256), 0, s, d_in, n,
                                    d_keys, d_idx, hist);
