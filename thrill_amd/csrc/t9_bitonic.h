@@ -171,7 +171,6 @@ __global__ __launch_bounds__(BLOCK, 2) void k_wave16_sort_sub(
     static_assert(E == 16, "wave16 mapping: 16 elements per lane");
     constexpr int ROUNDS = (NSORT == 1024) ? 6 : (NSORT == 2048) ? 7 : 8;
     __shared__ u64 lc[NSORT];
-    __shared__ u32 lv[HAS_VAL ? NSORT : 1];
     __shared__ u32 s_differ;
 
     const u32 sb = blockIdx.x;
@@ -188,8 +187,12 @@ __global__ __launch_bounds__(BLOCK, 2) void k_wave16_sort_sub(
     const u64 high16 = keys[gbase] & ~mask48;
     const u64 ref48 = keys[gbase] & mask48;
 
+    /* the composite embeds the ORIGINAL position (low 12 bits), so the
+     * value array never travels through the sort at all — it is
+     * re-gathered from global at writeback (r2: removing the u32 lv LDS
+     * array and the v[] registers raises occupancy, the measured
+     * limiter of this barrier-parked kernel) */
     u64 c[E];
-    u32 v[E];
     u32 differ = 0;
 #pragma unroll
     for (int r = 0; r < E; ++r) {
@@ -197,11 +200,9 @@ __global__ __launch_bounds__(BLOCK, 2) void k_wave16_sort_sub(
         if (i < ns) {
             const u64 k = keys[gbase + i];
             c[r] = ((k & mask48) << 12) | (u64)i;
-            if (HAS_VAL) v[r] = vals[gbase + i];
             differ |= ((k & mask48) != ref48);
         } else {
             c[r] = ~0ull;
-            if (HAS_VAL) v[r] = 0;
         }
     }
     if (differ) s_differ = 1;
@@ -222,7 +223,6 @@ __global__ __launch_bounds__(BLOCK, 2) void k_wave16_sort_sub(
                     const bool asc = (r & k) == 0;
                     if ((c[r] > c[q]) == asc) {
                         u64 tc = c[r]; c[r] = c[q]; c[q] = tc;
-                        if (HAS_VAL) { u32 tv = v[r]; v[r] = v[q]; v[q] = tv; }
                     }
                 }
             }
@@ -237,10 +237,7 @@ __global__ __launch_bounds__(BLOCK, 2) void k_wave16_sort_sub(
         const u32 L = (u32)E << lm;
         __syncthreads();
 #pragma unroll
-        for (int r = 0; r < E; ++r) {
-            lc[e0 + r] = c[r];
-            if (HAS_VAL) lv[e0 + r] = v[r];
-        }
+        for (int r = 0; r < E; ++r) lc[e0 + r] = c[r];
         __syncthreads();
         const u32 pairbase = e0 & ~(2 * L - 1);
         const u32 d = e0 - pairbase;          /* my first output diagonal */
@@ -257,13 +254,23 @@ __global__ __launch_bounds__(BLOCK, 2) void k_wave16_sort_sub(
 #pragma unroll
         for (int r = 0; r < E; ++r) {
             const bool ta = (j >= L) || (i < L && A[i] <= B[j]);
-            const u32 src = ta ? i : L + j;
             c[r] = (ta ? A[i] : B[j]);
-            if (HAS_VAL) v[r] = lv[pairbase + src];
             if (ta) ++i; else ++j;
         }
     }
 
+    /* writeback: keys from the composite; values re-gathered from their
+     * embedded original positions. All reads complete before any lane
+     * writes (barrier), since the sort is in place. */
+    u32 v[HAS_VAL ? E : 1];
+    if (HAS_VAL) {
+#pragma unroll
+        for (int r = 0; r < E; ++r) {
+            const u32 i = e0 + r;
+            if (i < ns) v[r] = vals[gbase + (u32)(c[r] & 0xFFFu)];
+        }
+    }
+    __syncthreads();
 #pragma unroll
     for (int r = 0; r < E; ++r) {
         const u32 i = e0 + r;
@@ -290,7 +297,6 @@ __global__ __launch_bounds__(BLOCK, 2) void k_wave16_sort_span(
     static_assert(E == 16, "wave16 mapping: 16 elements per lane");
     constexpr int ROUNDS = (NSORT == 1024) ? 6 : (NSORT == 2048) ? 7 : 8;
     __shared__ u64 lc[NSORT];
-    __shared__ u32 lv[HAS_VAL ? NSORT : 1];
     __shared__ u32 s_differ;
 
     const u32 sb = blockIdx.x;
@@ -316,8 +322,9 @@ __global__ __launch_bounds__(BLOCK, 2) void k_wave16_sort_span(
     const u64 high16 = keys_in[gbase] & ~mask48;
     const u64 ref48 = keys_in[gbase] & mask48;
 
+    /* values never travel through the sort (positions embedded in the
+     * composite; re-gathered at writeback — see k_wave16_sort_sub) */
     u64 c[E];
-    u32 v[E];
     u32 differ = 0;
 #pragma unroll
     for (int r = 0; r < E; ++r) {
@@ -325,11 +332,9 @@ __global__ __launch_bounds__(BLOCK, 2) void k_wave16_sort_span(
         if (i < ns) {
             const u64 k = keys_in[gbase + i];
             c[r] = ((k & mask48) << 12) | (u64)i;
-            if (HAS_VAL) v[r] = vals_in[gbase + i];
             differ |= ((k & mask48) != ref48);
         } else {
             c[r] = ~0ull;
-            if (HAS_VAL) v[r] = 0;
         }
     }
     if (differ) s_differ = 1;
@@ -341,7 +346,7 @@ __global__ __launch_bounds__(BLOCK, 2) void k_wave16_sort_span(
                 const u32 i = e0 + r;
                 if (i < ns) {
                     keys_out[gbase + i] = high16 | ((c[r] >> 12) & mask48);
-                    if (HAS_VAL) vals_out[gbase + i] = v[r];
+                    if (HAS_VAL) vals_out[gbase + i] = vals_in[gbase + i];
                 }
             }
         }
@@ -361,7 +366,6 @@ __global__ __launch_bounds__(BLOCK, 2) void k_wave16_sort_span(
                     const bool asc = (r & k) == 0;
                     if ((c[r] > c[q]) == asc) {
                         u64 tc = c[r]; c[r] = c[q]; c[q] = tc;
-                        if (HAS_VAL) { u32 tv = v[r]; v[r] = v[q]; v[q] = tv; }
                     }
                 }
             }
@@ -373,10 +377,7 @@ __global__ __launch_bounds__(BLOCK, 2) void k_wave16_sort_span(
         const u32 L = (u32)E << lm;
         __syncthreads();
 #pragma unroll
-        for (int r = 0; r < E; ++r) {
-            lc[e0 + r] = c[r];
-            if (HAS_VAL) lv[e0 + r] = v[r];
-        }
+        for (int r = 0; r < E; ++r) lc[e0 + r] = c[r];
         __syncthreads();
         const u32 pairbase = e0 & ~(2 * L - 1);
         const u32 d = e0 - pairbase;
@@ -393,13 +394,22 @@ __global__ __launch_bounds__(BLOCK, 2) void k_wave16_sort_span(
 #pragma unroll
         for (int r = 0; r < E; ++r) {
             const bool ta = (j >= L) || (i < L && A[i] <= B[j]);
-            const u32 src = ta ? i : L + j;
             c[r] = (ta ? A[i] : B[j]);
-            if (HAS_VAL) v[r] = lv[pairbase + src];
             if (ta) ++i; else ++j;
         }
     }
 
+    /* gather values from their embedded original positions, then write
+     * (barrier keeps all reads ahead of the in-place writes) */
+    u32 v[HAS_VAL ? E : 1];
+    if (HAS_VAL) {
+#pragma unroll
+        for (int r = 0; r < E; ++r) {
+            const u32 i = e0 + r;
+            if (i < ns) v[r] = vals_in[gbase + (u32)(c[r] & 0xFFFu)];
+        }
+    }
+    __syncthreads();
 #pragma unroll
     for (int r = 0; r < E; ++r) {
         const u32 i = e0 + r;
