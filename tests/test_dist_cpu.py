@@ -172,3 +172,30 @@ def test_distributed_terasort_protocol_world3():
     expect = o.sort_records(o.gen_records(n_total, seed=seed))
     got = np.concatenate([results[r] for r in range(3)])
     assert np.array_equal(got, expect)
+
+
+def test_distributed_terasort_protocol_world8():
+    """world_size=8 — the driver's scaling shape — with a small ragged
+    input; full sample->splitter->classify->exchange->sort protocol on
+    the oracle, bit-exact concatenation."""
+    seed, n_total = 0x88, 4003
+    ctxm = mp.get_context("spawn")
+    q = ctxm.Queue()
+    procs = [ctxm.Process(target=_worker,
+                          args=(r, 8, 29817, n_total, seed, q))
+             for r in range(8)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(8):
+        rank, data, cnt = q.get(timeout=300)
+        assert cnt >= 0, data
+        results[rank] = np.frombuffer(data, np.uint8).reshape(cnt, 100)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    from tests._oracle import Oracle
+    o = Oracle()
+    expect = o.sort_records(o.gen_records(n_total, seed=seed))
+    got = np.concatenate([results[r] for r in range(8)])
+    assert np.array_equal(got, expect)
