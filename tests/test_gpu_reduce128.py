@@ -223,3 +223,21 @@ def test_wordcount_pipeline_128_single_gpu(nat, oracle):
     got = sorted(zip(g1.tolist(), g2.tolist(), gv.tolist()))
     assert got == ek
     wc.close()
+
+
+def test_reduce128_overflow_sets_error(nat):
+    # more distinct composites than the table holds: the error flag must
+    # be set and the probe loop must terminate (bounded), not hang —
+    # mirrors the u64 table's overflow contract.
+    n = 1 << 12
+    k1 = np.arange(n, dtype=np.uint64)
+    k2 = np.ones(n, dtype=np.uint64)
+    cap = 256   # < distinct keys
+    d1, d2 = G.dev(k1), G.dev(k2)
+    tbl = G.empty(3 * cap, np.uint64)
+    derr = G.empty(1, np.uint32)
+    s = G.stream()
+    nat.reduce128_init(G.ptr(tbl), cap, s)
+    nat.reduce128_build(G.ptr(d1), G.ptr(d2), None, n, G.ptr(tbl), cap,
+                        0, G.ptr(derr), s)
+    assert int(G.host(derr, np.uint32)[0]) == 1
