@@ -181,7 +181,8 @@ def main():
         algo_bytes = (2.0 * REC + 4.0) * ts.n_local
         achieved = algo_bytes / per_launch_s / 1e9
         # traffic: PMC bytes per launch from the committed calibration run
-        # (profiles/pmc_traffic_terasort_r01.json — separate rocprofv3
+        # (profiles/pmc_traffic_terasort_r02.json, byte conversion
+        # anchored on the gather's exact write size — separate rocprofv3
         # --pmc FETCH_SIZE / WRITE_SIZE passes; FETCH doubled per the
         # gfx950 half-reporting of wide coalesced reads, absolute values
         # carry the guide's per-pattern calibration caveat). Scaled to
@@ -192,7 +193,10 @@ def main():
         # kernel at 82% of its own streaming rate).
         traffic = None
         cal_path = os.path.join(REPO, "profiles",
-                                "pmc_traffic_terasort_r01.json")
+                                "pmc_traffic_terasort_r02.json")
+        if not os.path.exists(cal_path):
+            cal_path = os.path.join(REPO, "profiles",
+                                    "pmc_traffic_terasort_r01.json")
         if os.path.exists(cal_path):
             import json as _json
             with open(cal_path) as f:
