@@ -228,7 +228,12 @@ inline bool read_device_direct(const char* name, void* d_ptr,
         }
         cur ^= 1;
         off += len;
-        if (ok && off < aligned)
+        /* sync unconditionally: the tail path below reuses pin[0], and
+         * the next iteration reuses the flipped buffer — either way the
+         * in-flight H2D must drain first (an intermittent corruption of
+         * the LAST chunk escaped when this sync was skipped on the
+         * final iteration) */
+        if (ok)
             ok = hipStreamSynchronize(s) == hipSuccess;
     }
     std::free(dbuf2);
